@@ -1,0 +1,595 @@
+// dolomite_hip — varlen causal flash attention for gfx950 (MI355X).
+//
+// Replaces flash_attn_varlen_func at the reference call site
+// attention/padding_free.py:51-62 (and flash.py:112-123): causal attention
+// over packed (total_q, heads, head_dim) with cu_seqlens, MHA/GQA/MQA head
+// layouts addressed in place on the fused c_attn projection output
+// (attention/base.py:72-81) via (t_stride, group_stride) addressing.
+//
+// Design (CDNA4-first, correctness-first round 1):
+//   - MFMA v_mfma_f32_16x16x32_bf16 for QK^T and PV; 64-wide waves.
+//   - workgroup = 4 waves = 256 threads; q-tile 64 rows (16 per wave),
+//     k-tile 64 keys streamed through LDS ([key][d] image for QK^T B-frags,
+//     transposed [d][key] image for PV B-frags — both read with 16-byte
+//     ds_read_b128, rows padded +16B so conflict groups hit distinct banks).
+//   - online softmax in fp32 VGPRs, m/l carried per row, row reductions via
+//     16-lane __shfl_xor (the C-fragment's 16 columns of one row).
+//   - P routed through a per-wave LDS tile to re-shape C-layout -> A-layout.
+//   - LSE (H,T) fp32 written for backward; backward recomputes P, computes
+//     dk/dv exclusively per (kv-tile, kv-head) workgroup looping q-tiles and
+//     grouped q-heads, dq via fp32 device atomics then a finalize cast.
+//   - dropout unsupported (hot-path configs run attn_pdrop = 0).
+//
+// Fragment layout assumption for mfma_f32_16x16x32_bf16 (verified on
+// hardware by dolomite_mfma_probe, tests/test_gpu_kernels.py):
+//   A[i][k]: lane l holds i = l&15, k = (l>>4)*8 + e   (e = 0..7)
+//   B[k][j]: lane l holds k = (l>>4)*8 + e, j = l&15
+//   C[r][c]: lane l, reg x holds r = (l>>4)*4 + x, c = l&15
+
+#include "common.h"
+
+typedef __bf16 bf16x8 __attribute__((ext_vector_type(8)));
+typedef float f32x4 __attribute__((ext_vector_type(4)));
+
+#define MFMA16(a, b, c) __builtin_amdgcn_mfma_f32_16x16x32_bf16((a), (b), (c), 0, 0, 0)
+
+// Load 8 consecutive bf16 from global with bounds/padding guard.
+// `base` points at element d0 of a row with `dmax` valid elements from d0's
+// row origin; d0 must be the element offset within the row.
+__device__ __forceinline__ bf16x8 load_bf16x8_guard(const __bf16* p, int d0, int D, bool valid) {
+    bf16x8 r;
+    if (valid && d0 + 8 <= D) {
+        r = *(const bf16x8*)p;
+    } else {
+#pragma unroll
+        for (int e = 0; e < 8; ++e) {
+            float f = (valid && d0 + e < D) ? (float)p[e] : 0.f;
+            r[e] = (__bf16)f;
+        }
+    }
+    return r;
+}
+
+// 16-lane row reduction (the 16 columns of one C-fragment row live in the
+// 16 lanes of one quarter-wave; xor over bits 0..3 stays inside it).
+__device__ __forceinline__ float qwave_reduce_max(float v) {
+#pragma unroll
+    for (int off = 8; off >= 1; off >>= 1) v = fmaxf(v, __shfl_xor(v, off, 64));
+    return v;
+}
+__device__ __forceinline__ float qwave_reduce_sum(float v) {
+#pragma unroll
+    for (int off = 8; off >= 1; off >>= 1) v += __shfl_xor(v, off, 64);
+    return v;
+}
+
+// ===========================================================================
+// Forward
+// ===========================================================================
+
+template <int DPAD>
+__global__ void __launch_bounds__(256) fa_fwd_kernel(
+    const __bf16* __restrict__ q, const __bf16* __restrict__ k, const __bf16* __restrict__ v,
+    __bf16* __restrict__ o, float* __restrict__ lse,
+    const int32_t* __restrict__ cu, int H, int Hkv, int D, int G,
+    int64_t q_ts, int64_t q_gs, int64_t k_ts, int64_t k_hs, int64_t v_ts, int64_t v_hs,
+    int64_t o_ts, int64_t T_total, float scale) {
+    constexpr int KCH = DPAD / 32;   // contraction chunks for QK^T
+    constexpr int DCH = DPAD / 16;   // output col blocks for PV
+    constexpr int SK = DPAD + 8;     // K LDS row stride (elems), +16B pad
+    constexpr int SV = 64 + 8;       // V^T / P LDS row stride
+
+    const int b = blockIdx.y;
+    const int h = blockIdx.z;
+    const int s0 = cu[b];
+    const int L = cu[b + 1] - s0;
+    const int qs = (int)blockIdx.x * 64;
+    if (qs >= L) return;
+
+    const int lane = threadIdx.x & 63;
+    const int wave = threadIdx.x >> 6;
+    const int lr = lane & 15;   // fragment col / A-row index
+    const int lg = lane >> 4;   // fragment k-group / C-row group
+
+    extern __shared__ char smem_raw[];
+    __bf16* Klds = (__bf16*)smem_raw;              // [64][SK]
+    __bf16* Vlds = Klds + 64 * SK;                 // [DPAD][SV] (transposed)
+    __bf16* Plds = Vlds + DPAD * SV;               // [4 waves][16][SV]
+    __bf16* Pw = Plds + wave * 16 * SV;
+
+    const int kvh = h / G;
+    const int64_t q_hoff = (int64_t)(h / G) * q_gs + (int64_t)(h % G) * D;
+
+    // --- load this wave's Q fragments (A-layout: i = lr, k = lg*8+e) ---
+    const int qrow = qs + wave * 16 + lr;
+    const bool qvalid = qrow < L;
+    bf16x8 qf[KCH];
+#pragma unroll
+    for (int kc = 0; kc < KCH; ++kc) {
+        int d0 = kc * 32 + lg * 8;
+        const __bf16* p = q + (int64_t)(s0 + (qvalid ? qrow : 0)) * q_ts + q_hoff + d0;
+        qf[kc] = load_bf16x8_guard(p, d0, D, qvalid);
+    }
+
+    float m_run[4], l_run[4];
+    f32x4 o_acc[DCH];
+#pragma unroll
+    for (int r = 0; r < 4; ++r) { m_run[r] = -INFINITY; l_run[r] = 0.f; }
+#pragma unroll
+    for (int dc = 0; dc < DCH; ++dc) o_acc[dc] = {0.f, 0.f, 0.f, 0.f};
+
+    const int kend = min(L, qs + 64);
+    const int ntiles = (kend + 63) / 64;
+
+    for (int kt = 0; kt < ntiles; ++kt) {
+        const int ks = kt * 64;
+        // --- cooperative staging: K -> [key][d], V -> transposed [d][key] ---
+        {
+            const int pieces = 64 * DPAD / 8;  // 8-elem pieces
+            for (int pidx = threadIdx.x; pidx < pieces; pidx += 256) {
+                int key = pidx / (DPAD / 8);
+                int d0 = (pidx % (DPAD / 8)) * 8;
+                bool kv_valid = (ks + key) < kend;
+                const __bf16* kp = k + (int64_t)(s0 + (kv_valid ? ks + key : 0)) * k_ts + (int64_t)kvh * k_hs + d0;
+                bf16x8 kk = load_bf16x8_guard(kp, d0, D, kv_valid);
+                *(bf16x8*)&Klds[key * SK + d0] = kk;
+                const __bf16* vp = v + (int64_t)(s0 + (kv_valid ? ks + key : 0)) * v_ts + (int64_t)kvh * v_hs + d0;
+                bf16x8 vv = load_bf16x8_guard(vp, d0, D, kv_valid);
+#pragma unroll
+                for (int e = 0; e < 8; ++e) Vlds[(d0 + e) * SV + key] = vv[e];
+            }
+        }
+        __syncthreads();
+
+        // --- QK^T: 4 key-blocks of 16, accumulate over KCH chunks ---
+        f32x4 sc[4];
+#pragma unroll
+        for (int cb = 0; cb < 4; ++cb) {
+            sc[cb] = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+            for (int kc = 0; kc < KCH; ++kc) {
+                bf16x8 kf = *(const bf16x8*)&Klds[(cb * 16 + lr) * SK + kc * 32 + lg * 8];
+                sc[cb] = MFMA16(qf[kc], kf, sc[cb]);
+            }
+        }
+
+        // --- mask + online softmax (fp32, per C-row) ---
+        float p_val[4][4];
+        float alpha[4];
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+            const int qpos = qs + wave * 16 + lg * 4 + r;
+            float rowmax = -INFINITY;
+#pragma unroll
+            for (int cb = 0; cb < 4; ++cb) {
+                const int kpos = ks + cb * 16 + lr;
+                float s = sc[cb][r] * scale;
+                if (kpos > qpos || kpos >= kend || qpos >= L) s = -INFINITY;
+                p_val[cb][r] = s;
+                rowmax = fmaxf(rowmax, s);
+            }
+            rowmax = qwave_reduce_max(rowmax);
+            float mnew = fmaxf(m_run[r], rowmax);
+            if (mnew == -INFINITY) mnew = 0.f;  // fully-masked row guard
+            alpha[r] = (m_run[r] == -INFINITY) ? 0.f : __expf(m_run[r] - mnew);
+            float rsum = 0.f;
+#pragma unroll
+            for (int cb = 0; cb < 4; ++cb) {
+                float e = (p_val[cb][r] == -INFINITY) ? 0.f : __expf(p_val[cb][r] - mnew);
+                p_val[cb][r] = e;
+                rsum += e;
+            }
+            rsum = qwave_reduce_sum(rsum);
+            l_run[r] = l_run[r] * alpha[r] + rsum;
+            m_run[r] = mnew;
+        }
+        // rescale accumulator and stash P (bf16) in this wave's LDS tile
+#pragma unroll
+        for (int dc = 0; dc < DCH; ++dc)
+#pragma unroll
+            for (int r = 0; r < 4; ++r) o_acc[dc][r] *= alpha[r];
+#pragma unroll
+        for (int cb = 0; cb < 4; ++cb)
+#pragma unroll
+            for (int r = 0; r < 4; ++r) Pw[(lg * 4 + r) * SV + cb * 16 + lr] = (__bf16)p_val[cb][r];
+
+        // --- PV: A = P (this wave's rows), B = V^T image ---
+#pragma unroll
+        for (int kc2 = 0; kc2 < 2; ++kc2) {
+            bf16x8 pf = *(const bf16x8*)&Pw[lr * SV + kc2 * 32 + lg * 8];
+#pragma unroll
+            for (int dc = 0; dc < DCH; ++dc) {
+                bf16x8 vf = *(const bf16x8*)&Vlds[(dc * 16 + lr) * SV + kc2 * 32 + lg * 8];
+                o_acc[dc] = MFMA16(pf, vf, o_acc[dc]);
+            }
+        }
+        __syncthreads();  // K/V LDS reused next tile
+    }
+
+    // --- epilogue: normalize, store O and LSE ---
+    float inv_l[4];
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+        inv_l[r] = (l_run[r] > 0.f) ? 1.f / l_run[r] : 0.f;
+        const int qpos = qs + wave * 16 + lg * 4 + r;
+        if (lr == 0 && qpos < L)
+            lse[(int64_t)h * T_total + s0 + qpos] = m_run[r] + __logf(l_run[r]);
+    }
+#pragma unroll
+    for (int dc = 0; dc < DCH; ++dc) {
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+            const int qpos = qs + wave * 16 + lg * 4 + r;
+            const int d = dc * 16 + lr;
+            if (qpos < L && d < D)
+                o[(int64_t)(s0 + qpos) * o_ts + (int64_t)h * D + d] = (__bf16)(o_acc[dc][r] * inv_l[r]);
+        }
+    }
+}
+
+template <int DPAD>
+static int launch_fa_fwd(hipStream_t stream, const __bf16* q, const __bf16* k, const __bf16* v,
+                         __bf16* o, float* lse, const int32_t* cu, int batch, int64_t T,
+                         int H, int Hkv, int D, int G,
+                         int64_t q_ts, int64_t q_gs, int64_t k_ts, int64_t k_hs,
+                         int64_t v_ts, int64_t v_hs, int max_tiles, float scale) {
+    dim3 grid(max_tiles, batch, H), block(256);
+    size_t shmem = (64 * (DPAD + 8) + DPAD * 72 + 4 * 16 * 72) * sizeof(__bf16);
+    hipLaunchKernelGGL((fa_fwd_kernel<DPAD>), grid, block, shmem, stream,
+                       q, k, v, o, lse, cu, H, Hkv, D, G,
+                       q_ts, q_gs, k_ts, k_hs, v_ts, v_hs, (int64_t)H * D, T, scale);
+    return dol_last_error();
+}
+
+extern "C" int dolomite_fa_varlen_fwd(dolomite_stream_t stream,
+                                      const void* q, const void* k, const void* v,
+                                      void* o, float* lse,
+                                      const int32_t* cu_seqlens, int batch, int max_seqlen, int64_t T,
+                                      int H, int Hkv, int D, int G,
+                                      int64_t q_tstride, int64_t q_gstride,
+                                      int64_t k_tstride, int64_t k_hstride,
+                                      int64_t v_tstride, int64_t v_hstride,
+                                      float scale, int dtype) {
+    if (dtype != DOLOMITE_BF16) return 9010;  // bf16 only (north-star dtype)
+    if (D > 128) return 9011;
+    // grid.x = tiles of the longest sequence; shorter sequences' surplus
+    // workgroups exit on the cu_seqlens check.
+    int max_tiles = (max_seqlen + 63) / 64;
+    hipStream_t s = (hipStream_t)stream;
+#define CASE(DP)                                                                                     \
+    return launch_fa_fwd<DP>(s, (const __bf16*)q, (const __bf16*)k, (const __bf16*)v, (__bf16*)o,    \
+                             lse, cu_seqlens, batch, T, H, Hkv, D, G, q_tstride, q_gstride,          \
+                             k_tstride, k_hstride, v_tstride, v_hstride, max_tiles, scale)
+    if (D <= 32) CASE(32);
+    if (D <= 64) CASE(64);
+    if (D <= 96) CASE(96);
+    CASE(128);
+#undef CASE
+}
+
+// ===========================================================================
+// Backward pass 1: delta[h,t] = sum_d dO[t,h,d] * O[t,h,d]   (one wave/row)
+// ===========================================================================
+
+template <typename T>
+__global__ void __launch_bounds__(256) fa_bwd_preprocess_kernel(
+    const T* __restrict__ o, const T* __restrict__ dout, float* __restrict__ delta,
+    int64_t T_total, int H, int D, int64_t o_ts, int64_t do_ts) {
+    int64_t row = ((int64_t)blockIdx.x * 4) + (threadIdx.x >> 6);
+    int lane = threadIdx.x & 63;
+    if (row >= T_total * H) return;
+    int64_t t = row / H;
+    int h = (int)(row % H);
+    const T* op = o + t * o_ts + (int64_t)h * D;
+    const T* dp = dout + t * do_ts + (int64_t)h * D;
+    float acc = 0.f;
+    for (int d = lane; d < D; d += 64) acc += load_as_f32(op + d) * load_as_f32(dp + d);
+    acc = wave_reduce_sum(acc);
+    if (lane == 0) delta[(int64_t)h * T_total + t] = acc;
+}
+
+extern "C" int dolomite_fa_bwd_preprocess(dolomite_stream_t stream,
+                                          const void* o, const void* dout, float* delta,
+                                          int64_t T, int H, int D,
+                                          int64_t o_tstride, int64_t do_tstride, int dtype) {
+    int64_t rows = T * H;
+    dim3 grid((uint32_t)((rows + 3) / 4)), block(256);
+    if (dtype == DOLOMITE_BF16)
+        hipLaunchKernelGGL((fa_bwd_preprocess_kernel<uint16_t>), grid, block, 0, (hipStream_t)stream,
+                           (const uint16_t*)o, (const uint16_t*)dout, delta, T, H, D, o_tstride, do_tstride);
+    else
+        hipLaunchKernelGGL((fa_bwd_preprocess_kernel<float>), grid, block, 0, (hipStream_t)stream,
+                           (const float*)o, (const float*)dout, delta, T, H, D, o_tstride, do_tstride);
+    return dol_last_error();
+}
+
+// ===========================================================================
+// Backward pass 2: per (kv-tile, seq, kv-head) workgroup.
+// ===========================================================================
+
+template <int DPAD>
+__global__ void __launch_bounds__(256) fa_bwd_kernel(
+    const __bf16* __restrict__ q, const __bf16* __restrict__ k, const __bf16* __restrict__ v,
+    const __bf16* __restrict__ dout, const float* __restrict__ lse, const float* __restrict__ delta,
+    float* __restrict__ dq_acc, __bf16* __restrict__ dk, __bf16* __restrict__ dv,
+    const int32_t* __restrict__ cu, int H, int Hkv, int D, int G,
+    int64_t q_ts, int64_t q_gs, int64_t k_ts, int64_t k_hs, int64_t v_ts, int64_t v_hs,
+    int64_t do_ts, int64_t T_total, float scale) {
+    constexpr int KCH = DPAD / 32;
+    constexpr int DCH = DPAD / 16;
+    constexpr int SQ = DPAD + 8;   // [q][d] and [key][d] image stride
+    constexpr int ST = 64 + 8;     // transposed [d][q] / [q][key] image stride
+
+    const int b = blockIdx.y;
+    const int kvh = blockIdx.z;
+    const int s0 = cu[b];
+    const int L = cu[b + 1] - s0;
+    const int ks = (int)blockIdx.x * 64;
+    if (ks >= L) return;
+
+    const int lane = threadIdx.x & 63;
+    const int wave = threadIdx.x >> 6;
+    const int lr = lane & 15;
+    const int lg = lane >> 4;
+
+    extern __shared__ char smem_raw[];
+    __bf16* Qlds = (__bf16*)smem_raw;          // [64 q][SQ]
+    __bf16* QTl = Qlds + 64 * SQ;              // [DPAD][ST]   (Q^T)
+    __bf16* dOl = QTl + DPAD * ST;             // [64 q][SQ]
+    __bf16* dOTl = dOl + 64 * SQ;              // [DPAD][ST]   (dO^T)
+    __bf16* KTl = dOTl + DPAD * ST;            // [DPAD][ST]   (K^T)
+    __bf16* dSl = KTl + DPAD * ST;             // [64 q][ST]   (dS, [q][key])
+    __bf16* dSTl = dSl + 64 * ST;              // [64 key][ST] (dS^T, [key][q])
+    __bf16* PTl = dSTl + 64 * ST;              // [64 key][ST] (P^T, [key][q])
+
+    const int kend = min(L, ks + 64);
+
+    // --- this wave's K and V fragments (A-layout: i = lr -> key) ---
+    const int krow = ks + wave * 16 + lr;
+    const bool kvalid = krow < kend;
+    bf16x8 kfr[KCH], vfr[KCH];
+#pragma unroll
+    for (int kc = 0; kc < KCH; ++kc) {
+        int d0 = kc * 32 + lg * 8;
+        const __bf16* kp = k + (int64_t)(s0 + (kvalid ? krow : 0)) * k_ts + (int64_t)kvh * k_hs + d0;
+        kfr[kc] = load_bf16x8_guard(kp, d0, D, kvalid);
+        const __bf16* vp = v + (int64_t)(s0 + (kvalid ? krow : 0)) * v_ts + (int64_t)kvh * v_hs + d0;
+        vfr[kc] = load_bf16x8_guard(vp, d0, D, kvalid);
+    }
+    // --- stage K^T image cooperatively ---
+    {
+        const int pieces = 64 * DPAD / 8;
+        for (int pidx = threadIdx.x; pidx < pieces; pidx += 256) {
+            int key = pidx / (DPAD / 8);
+            int d0 = (pidx % (DPAD / 8)) * 8;
+            bool valid = (ks + key) < kend;
+            const __bf16* kp = k + (int64_t)(s0 + (valid ? ks + key : 0)) * k_ts + (int64_t)kvh * k_hs + d0;
+            bf16x8 kk = load_bf16x8_guard(kp, d0, D, valid);
+#pragma unroll
+            for (int e = 0; e < 8; ++e) KTl[(d0 + e) * ST + key] = kk[e];
+        }
+    }
+
+    f32x4 dv_acc[DCH], dk_acc[DCH];
+#pragma unroll
+    for (int dc = 0; dc < DCH; ++dc) {
+        dv_acc[dc] = {0.f, 0.f, 0.f, 0.f};
+        dk_acc[dc] = {0.f, 0.f, 0.f, 0.f};
+    }
+
+    const int qt0 = ks / 64;
+    const int nqt = (L + 63) / 64;
+
+    for (int qt = qt0; qt < nqt; ++qt) {
+        const int qs = qt * 64;
+        for (int hg = 0; hg < G; ++hg) {
+            const int h = kvh * G + hg;
+            const int64_t q_hoff = (int64_t)kvh * q_gs + (int64_t)hg * D;
+            // --- stage Q and dO tiles (both images each) ---
+            __syncthreads();  // previous iteration's reads done
+            {
+                const int pieces = 64 * DPAD / 8;
+                for (int pidx = threadIdx.x; pidx < pieces; pidx += 256) {
+                    int qq = pidx / (DPAD / 8);
+                    int d0 = (pidx % (DPAD / 8)) * 8;
+                    bool valid = (qs + qq) < L;
+                    const __bf16* qp = q + (int64_t)(s0 + (valid ? qs + qq : 0)) * q_ts + q_hoff + d0;
+                    bf16x8 qq8 = load_bf16x8_guard(qp, d0, D, valid);
+                    *(bf16x8*)&Qlds[qq * SQ + d0] = qq8;
+#pragma unroll
+                    for (int e = 0; e < 8; ++e) QTl[(d0 + e) * ST + qq] = qq8[e];
+                    const __bf16* dp = dout + (int64_t)(s0 + (valid ? qs + qq : 0)) * do_ts + (int64_t)h * D + d0;
+                    bf16x8 dd8 = load_bf16x8_guard(dp, d0, D, valid);
+                    *(bf16x8*)&dOl[qq * SQ + d0] = dd8;
+#pragma unroll
+                    for (int e = 0; e < 8; ++e) dOTl[(d0 + e) * ST + qq] = dd8[e];
+                }
+            }
+            __syncthreads();
+
+            // --- S^T = K·Q^T, P^T = exp(scale*S^T - lse) ; dP^T = V·dO^T ---
+            // C layout: col = q = lr, row = key = 16*wave + lg*4 + r
+            f32x4 st[4], dpt[4];
+#pragma unroll
+            for (int cb = 0; cb < 4; ++cb) {
+                st[cb] = {0.f, 0.f, 0.f, 0.f};
+                dpt[cb] = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+                for (int kc = 0; kc < KCH; ++kc) {
+                    bf16x8 qtf = *(const bf16x8*)&Qlds[(cb * 16 + lr) * SQ + kc * 32 + lg * 8];
+                    st[cb] = MFMA16(kfr[kc], qtf, st[cb]);
+                    bf16x8 dtf = *(const bf16x8*)&dOl[(cb * 16 + lr) * SQ + kc * 32 + lg * 8];
+                    dpt[cb] = MFMA16(vfr[kc], dtf, dpt[cb]);
+                }
+            }
+
+#pragma unroll
+            for (int cb = 0; cb < 4; ++cb) {
+                const int qpos = qs + cb * 16 + lr;
+                const bool qok = qpos < L;
+                float lsev = qok ? lse[(int64_t)h * T_total + s0 + qpos] : 0.f;
+                float delv = qok ? delta[(int64_t)h * T_total + s0 + qpos] : 0.f;
+#pragma unroll
+                for (int r = 0; r < 4; ++r) {
+                    const int kpos = ks + wave * 16 + lg * 4 + r;
+                    bool ok = qok && kpos < kend && kpos <= qpos;
+                    float pv = ok ? __expf(st[cb][r] * scale - lsev) : 0.f;
+                    float ds = ok ? pv * (dpt[cb][r] - delv) * scale : 0.f;
+                    // stash P^T [key][q] and dS both images
+                    PTl[(wave * 16 + lg * 4 + r) * ST + cb * 16 + lr] = (__bf16)pv;
+                    dSTl[(wave * 16 + lg * 4 + r) * ST + cb * 16 + lr] = (__bf16)ds;
+                    dSl[(cb * 16 + lr) * ST + wave * 16 + lg * 4 + r] = (__bf16)ds;
+                }
+            }
+            __syncthreads();
+
+            // --- dV += P^T·dO ; dK += dS^T·Q  (contraction over q) ---
+#pragma unroll
+            for (int kc2 = 0; kc2 < 2; ++kc2) {
+                bf16x8 ptf = *(const bf16x8*)&PTl[(wave * 16 + lr) * ST + kc2 * 32 + lg * 8];
+                bf16x8 dstf = *(const bf16x8*)&dSTl[(wave * 16 + lr) * ST + kc2 * 32 + lg * 8];
+#pragma unroll
+                for (int dc = 0; dc < DCH; ++dc) {
+                    bf16x8 dotb = *(const bf16x8*)&dOTl[(dc * 16 + lr) * ST + kc2 * 32 + lg * 8];
+                    dv_acc[dc] = MFMA16(ptf, dotb, dv_acc[dc]);
+                    bf16x8 qtb = *(const bf16x8*)&QTl[(dc * 16 + lr) * ST + kc2 * 32 + lg * 8];
+                    dk_acc[dc] = MFMA16(dstf, qtb, dk_acc[dc]);
+                }
+            }
+
+            // --- dQ strip (this wave's 16 q rows): dS·K^T-image, atomics ---
+#pragma unroll
+            for (int dc = 0; dc < DCH; ++dc) {
+                f32x4 dq = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+                for (int kc2 = 0; kc2 < 2; ++kc2) {
+                    bf16x8 dsf = *(const bf16x8*)&dSl[(wave * 16 + lr) * ST + kc2 * 32 + lg * 8];
+                    bf16x8 ktb = *(const bf16x8*)&KTl[(dc * 16 + lr) * ST + kc2 * 32 + lg * 8];
+                    dq = MFMA16(dsf, ktb, dq);
+                }
+#pragma unroll
+                for (int r = 0; r < 4; ++r) {
+                    const int qpos = qs + wave * 16 + lg * 4 + r;
+                    const int d = dc * 16 + lr;
+                    if (qpos < L && d < D)
+                        atomicAdd(&dq_acc[((int64_t)(s0 + qpos) * H + h) * D + d], dq[r]);
+                }
+            }
+        }
+    }
+
+    // --- store dK/dV (exclusive ownership of this kv strip) ---
+#pragma unroll
+    for (int dc = 0; dc < DCH; ++dc) {
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+            const int kpos = ks + wave * 16 + lg * 4 + r;
+            const int d = dc * 16 + lr;
+            if (kpos < kend && d < D) {
+                dk[(int64_t)(s0 + kpos) * k_ts + (int64_t)kvh * k_hs + d] = (__bf16)dk_acc[dc][r];
+                dv[(int64_t)(s0 + kpos) * v_ts + (int64_t)kvh * v_hs + d] = (__bf16)dv_acc[dc][r];
+            }
+        }
+    }
+}
+
+template <int DPAD>
+static int launch_fa_bwd(hipStream_t stream, const __bf16* q, const __bf16* k, const __bf16* v,
+                         const __bf16* dout, const float* lse, const float* delta,
+                         float* dq_acc, __bf16* dk, __bf16* dv,
+                         const int32_t* cu, int batch, int64_t T, int H, int Hkv, int D, int G,
+                         int64_t q_ts, int64_t q_gs, int64_t k_ts, int64_t k_hs,
+                         int64_t v_ts, int64_t v_hs, int64_t do_ts, int max_tiles, float scale) {
+    dim3 grid(max_tiles, batch, Hkv), block(256);
+    constexpr int SQ = DPAD + 8, ST = 64 + 8;
+    size_t shmem = (size_t)(64 * SQ * 2 + DPAD * ST * 3 + 64 * ST * 3) * sizeof(__bf16);
+    hipLaunchKernelGGL((fa_bwd_kernel<DPAD>), grid, block, shmem, stream,
+                       q, k, v, dout, lse, delta, dq_acc, dk, dv, cu, H, Hkv, D, G,
+                       q_ts, q_gs, k_ts, k_hs, v_ts, v_hs, do_ts, T, scale);
+    return dol_last_error();
+}
+
+extern "C" int dolomite_fa_varlen_bwd(dolomite_stream_t stream,
+                                      const void* q, const void* k, const void* v,
+                                      const void* dout, const float* lse,
+                                      const float* delta, float* dq_acc, void* dk, void* dv,
+                                      const int32_t* cu_seqlens, int batch, int max_seqlen, int64_t T,
+                                      int H, int Hkv, int D, int G,
+                                      int64_t q_tstride, int64_t q_gstride,
+                                      int64_t k_tstride, int64_t k_hstride,
+                                      int64_t v_tstride, int64_t v_hstride,
+                                      int64_t do_tstride,
+                                      float scale, int dtype) {
+    if (dtype != DOLOMITE_BF16) return 9010;
+    if (D > 128) return 9011;
+    int max_tiles = (max_seqlen + 63) / 64;
+    hipStream_t s = (hipStream_t)stream;
+#define CASE(DP)                                                                                        \
+    return launch_fa_bwd<DP>(s, (const __bf16*)q, (const __bf16*)k, (const __bf16*)v,                   \
+                             (const __bf16*)dout, lse, delta, dq_acc, (__bf16*)dk, (__bf16*)dv,         \
+                             cu_seqlens, batch, T, H, Hkv, D, G, q_tstride, q_gstride, k_tstride,       \
+                             k_hstride, v_tstride, v_hstride, do_tstride, max_tiles, scale)
+    if (D <= 32) CASE(32);
+    if (D <= 64) CASE(64);
+    if (D <= 96) CASE(96);
+    CASE(128);
+#undef CASE
+}
+
+// ===========================================================================
+// Backward pass 3: cast dq_acc into the packed dqkv q-slots.
+// ===========================================================================
+
+template <typename T>
+__global__ void __launch_bounds__(256) fa_dq_finalize_kernel(
+    const float* __restrict__ dq_acc, T* __restrict__ dqkv_q,
+    int64_t total, int H, int D, int G, int64_t q_ts, int64_t q_gs) {
+    int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (idx >= total) return;
+    int64_t t = idx / ((int64_t)H * D);
+    int rem = (int)(idx % ((int64_t)H * D));
+    int h = rem / D;
+    int d = rem % D;
+    store_from_f32(&dqkv_q[t * q_ts + (int64_t)(h / G) * q_gs + (int64_t)(h % G) * D + d], dq_acc[idx]);
+}
+
+extern "C" int dolomite_fa_dq_finalize(dolomite_stream_t stream,
+                                       const float* dq_acc, void* dqkv_q,
+                                       int64_t T, int H, int D, int G,
+                                       int64_t q_tstride, int64_t q_gstride, int dtype) {
+    int64_t total = T * (int64_t)H * D;
+    if (total == 0) return 0;
+    dim3 grid((uint32_t)((total + 255) / 256)), block(256);
+    if (dtype == DOLOMITE_BF16)
+        hipLaunchKernelGGL((fa_dq_finalize_kernel<uint16_t>), grid, block, 0, (hipStream_t)stream,
+                           dq_acc, (uint16_t*)dqkv_q, total, H, D, G, q_tstride, q_gstride);
+    else
+        hipLaunchKernelGGL((fa_dq_finalize_kernel<float>), grid, block, 0, (hipStream_t)stream,
+                           dq_acc, (float*)dqkv_q, total, H, D, G, q_tstride, q_gstride);
+    return dol_last_error();
+}
+
+// ===========================================================================
+// MFMA fragment-layout self-test (see header comment).
+// ===========================================================================
+
+__global__ void __launch_bounds__(64) mfma_probe_kernel(const __bf16* A, const __bf16* B, float* C) {
+    int lane = threadIdx.x & 63;
+    int lr = lane & 15, lg = lane >> 4;
+    bf16x8 a, b;
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+        a[e] = A[lr * 32 + lg * 8 + e];      // A[i][k], i=lr, k=lg*8+e
+        b[e] = B[(lg * 8 + e) * 16 + lr];    // B[k][j], k=lg*8+e, j=lr
+    }
+    f32x4 c = {0.f, 0.f, 0.f, 0.f};
+    c = MFMA16(a, b, c);
+#pragma unroll
+    for (int r = 0; r < 4; ++r) C[(lg * 4 + r) * 16 + lr] = c[r];
+}
+
+extern "C" int dolomite_mfma_probe(dolomite_stream_t stream, const void* A, const void* B, float* C) {
+    hipLaunchKernelGGL(mfma_probe_kernel, dim3(1), dim3(64), 0, (hipStream_t)stream,
+                       (const __bf16*)A, (const __bf16*)B, C);
+    return dol_last_error();
+}

@@ -1,0 +1,84 @@
+// Shared device helpers for the dolomite_hip kernels (gfx950 / CDNA4).
+#pragma once
+
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+#include <stdint.h>
+
+#include "../../include/dolomite_hip.h"
+
+#define WAVE 64  // CDNA wavefront width (hard-coded per guide §1)
+
+#define HIP_CHECK_RET(expr)                \
+    do {                                   \
+        hipError_t _e = (expr);            \
+        if (_e != hipSuccess) return (int)_e; \
+    } while (0)
+
+static inline int dol_last_error() {
+    hipError_t e = hipGetLastError();
+    return (int)e;
+}
+
+// ---- dtype load/store: fp32 (code 0) or bf16 (code 1) ---------------------
+
+__device__ __forceinline__ float bf16_to_f32(uint16_t u) {
+    union { uint32_t u; float f; } c;
+    c.u = ((uint32_t)u) << 16;
+    return c.f;
+}
+
+__device__ __forceinline__ uint16_t f32_to_bf16(float f) {
+    // round-to-nearest-even, matching torch's float->bfloat16 cast
+    union { float f; uint32_t u; } c;
+    c.f = f;
+    uint32_t u = c.u;
+    if ((u & 0x7fffffffu) > 0x7f800000u) return (uint16_t)((u >> 16) | 0x0040u);  // NaN
+    uint32_t rounding = 0x7fffu + ((u >> 16) & 1u);
+    return (uint16_t)((u + rounding) >> 16);
+}
+
+template <typename T>
+__device__ __forceinline__ float load_as_f32(const T* p);
+template <>
+__device__ __forceinline__ float load_as_f32<float>(const float* p) { return *p; }
+template <>
+__device__ __forceinline__ float load_as_f32<uint16_t>(const uint16_t* p) { return bf16_to_f32(*p); }
+
+template <typename T>
+__device__ __forceinline__ void store_from_f32(T* p, float v);
+template <>
+__device__ __forceinline__ void store_from_f32<float>(float* p, float v) { *p = v; }
+template <>
+__device__ __forceinline__ void store_from_f32<uint16_t>(uint16_t* p, float v) { *p = f32_to_bf16(v); }
+
+// ---- wave / block reductions ---------------------------------------------
+
+__device__ __forceinline__ float wave_reduce_sum(float v) {
+#pragma unroll
+    for (int off = 32; off >= 1; off >>= 1) v += __shfl_xor(v, off, 64);
+    return v;
+}
+
+__device__ __forceinline__ float wave_reduce_max(float v) {
+#pragma unroll
+    for (int off = 32; off >= 1; off >>= 1) v = fmaxf(v, __shfl_xor(v, off, 64));
+    return v;
+}
+
+// Block-level sum over `nthreads` (multiple of 64) using `red` LDS scratch
+// (size nthreads/64 floats). Every thread returns the total.
+__device__ __forceinline__ float block_reduce_sum(float v, float* red, int nthreads) {
+    int lane = threadIdx.x & 63;
+    int wid = threadIdx.x >> 6;
+    int nw = nthreads >> 6;
+    v = wave_reduce_sum(v);
+    if (lane == 0) red[wid] = v;
+    __syncthreads();
+    float total = 0.f;
+#pragma unroll
+    for (int i = 0; i < 8; ++i)
+        if (i < nw) total += red[i];
+    __syncthreads();
+    return total;
+}

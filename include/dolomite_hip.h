@@ -1,0 +1,205 @@
+/* dolomite_hip — C-ABI boundary of the MI355X HIP kernel library.
+ *
+ * These entry points are the gfx950-native replacements for the GPU work the
+ * reference delegates to third-party kernels (SURVEY.md §2 "Delegated
+ * kernel" table). Each declaration cites the reference call site
+ * (paths under /root/reference/dolomite_engine/) it replaces.
+ *
+ * Conventions:
+ *   - caller (the Python host on PyTorch-ROCm) owns all memory; pointers are
+ *     device pointers; strides/sizes are in ELEMENTS (not bytes) as int64;
+ *   - `stream` is the hipStream_t the call is ordered on (passed as void*);
+ *   - dtype codes: 0 = fp32, 1 = bf16;
+ *   - return 0 on success, nonzero hipError_t-style code on failure;
+ *   - thread-safe per stream; no internal allocation on the hot path.
+ */
+
+#ifndef DOLOMITE_HIP_H
+#define DOLOMITE_HIP_H
+
+#include <stdint.h>
+
+#ifdef __cplusplus
+extern "C" {
+#endif
+
+typedef void* dolomite_stream_t; /* hipStream_t */
+
+enum dolomite_dtype { DOLOMITE_F32 = 0, DOLOMITE_BF16 = 1 };
+
+/* Library/version probe (used by the loader to verify ABI). */
+int dolomite_hip_abi_version(void);
+
+/* ------------------------------------------------------------------------
+ * Fused RMSNorm (+ optional residual add), forward / backward.
+ * Replaces: Triton RMSNorm (modeling_utils/normalization/rmsnorm/torchtitan.py)
+ * with the canonical semantics of rmsnorm/base.py:18-25:
+ *   s     = x + res_in            (when res_in != NULL; else s = x)
+ *   rstd  = rsqrt(mean(s_fp32^2) + eps)
+ *   y     = w * cast_to_input_dtype(s_fp32 * rstd)
+ * res_out (when non-NULL) receives s (the pre-norm sum, used as the next
+ * residual). rstd is cached fp32 per row for backward.
+ * x,y,res_*: (T, H) row-major contiguous, dtype `dtype`. w: (H,), dtype.
+ * ---------------------------------------------------------------------- */
+int dolomite_rmsnorm_fwd(dolomite_stream_t stream,
+                         const void* x, const void* res_in, const void* w,
+                         void* y, void* res_out, float* rstd,
+                         int64_t T, int64_t H, float eps, int dtype);
+
+/* Backward: given dy and the saved pre-norm input s (= res_out of fwd, or x
+ * when no residual) and rstd, computes
+ *   dx  = r*(w*dy - s_hat * mean(w*dy*s_hat))  with s_hat = s_fp32*rstd
+ *   dw  = sum_rows(dy * cast_to_dtype(s_hat))        (fp32 accumulation)
+ * dw_partial: (nblocks, H) fp32 scratch written by the kernel;
+ * dolomite_reduce_partials sums it into dw (H,) fp32.
+ * nblocks is returned by dolomite_rmsnorm_bwd_nblocks(T). */
+int dolomite_rmsnorm_bwd_nblocks(int64_t T);
+int dolomite_rmsnorm_bwd(dolomite_stream_t stream,
+                         const void* dy, const void* s, const void* w,
+                         const float* rstd, void* dx, float* dw_partial,
+                         int64_t T, int64_t H, int dtype);
+
+/* LayerNorm, same calling pattern (replaces the reference 'layernorm'/'torch'
+ * nn.LayerNorm path; mean/rstd cached fp32). db shares dw_partial layout:
+ * dw_partial holds (2*nblocks, H): [dw partials ; db partials]. */
+int dolomite_layernorm_fwd(dolomite_stream_t stream,
+                           const void* x, const void* res_in, const void* w, const void* b,
+                           void* y, void* res_out, float* mean, float* rstd,
+                           int64_t T, int64_t H, float eps, int dtype);
+int dolomite_layernorm_bwd(dolomite_stream_t stream,
+                           const void* dy, const void* s, const void* w,
+                           const float* mean, const float* rstd,
+                           void* dx, float* dwdb_partial,
+                           int64_t T, int64_t H, int dtype);
+
+/* Sum partials: out[h] = sum_i partial[i*H + h]; out fp32 (H,). */
+int dolomite_reduce_partials(dolomite_stream_t stream,
+                             const float* partial, float* out,
+                             int64_t nblocks, int64_t H);
+
+/* ------------------------------------------------------------------------
+ * RoPE on the packed QKV projection output, forward / backward.
+ * Replaces apply_rotary_pos_emb (position_embedding/rope.py:104-121) at the
+ * call site attention/padding_free.py:38-40, operating IN PLACE of the
+ * fused c_attn layout (attention/base.py:72-81) without unpacking:
+ *   qkv: (T, row_len) where row q-head h lives at
+ *        (h/G)*q_gstride + (h%G)*D  and kv-head j at k_off + j*kv_hstride
+ *        (v is untouched). G = n_head/num_kv_heads heads per kv group.
+ *   cos/sin: (T, D) fp32, already gathered per token position
+ *        (gpt_dolomite/base.py:289-296).
+ *   dir: +1 forward rotation, -1 inverse (exact backward, since the table
+ *        halves are duplicated: R^T = -R and S commutes with R).
+ * out != in allowed (copy-through), in == out allowed (in-place).
+ * ---------------------------------------------------------------------- */
+int dolomite_rope_qkv(dolomite_stream_t stream,
+                      const void* qkv_in, void* qkv_out,
+                      const float* cos_t, const float* sin_t,
+                      int64_t T, int64_t row_len,
+                      int H, int Hkv, int D, int G,
+                      int64_t q_gstride, int64_t k_off, int64_t kv_hstride,
+                      int dir, int rotate_v_copy, int dtype);
+
+/* ------------------------------------------------------------------------
+ * Varlen causal flash attention, forward.
+ * Replaces flash_attn_varlen_func (attention/padding_free.py:51-62).
+ * q/k/v address layout (strides in elements):
+ *   q[t, h, d]  = q_ptr + t*q_tstride + (h/G)*q_gstride + (h%G)*D + d
+ *   k[t, j, d]  = k_ptr + t*k_tstride + j*k_hstride + d
+ *   v[t, j, d]  = v_ptr + t*v_tstride + j*v_hstride + d
+ * (this addresses the packed c_attn output without copies for MHA/GQA/MQA)
+ * o: (T, H*D) row-major contiguous bf16/fp32; lse: (H, T) fp32 log-sum-exp
+ * (natural log, includes softmax scale), for backward.
+ * cu_seqlens: (batch+1,) int32 device pointer; causal only; dropout
+ * unsupported (training configs on this path run attn_pdrop = 0).
+ * ---------------------------------------------------------------------- */
+int dolomite_fa_varlen_fwd(dolomite_stream_t stream,
+                           const void* q, const void* k, const void* v,
+                           void* o, float* lse,
+                           const int32_t* cu_seqlens, int batch, int max_seqlen, int64_t T,
+                           int H, int Hkv, int D, int G,
+                           int64_t q_tstride, int64_t q_gstride,
+                           int64_t k_tstride, int64_t k_hstride,
+                           int64_t v_tstride, int64_t v_hstride,
+                           float scale, int dtype);
+
+/* Backward, three passes (all launched by the host in order on `stream`):
+ *  1. preprocess: delta[h,t] = rowsum(dO[t,h,:]*O[t,h,:]) fp32.
+ *  2. main: grid over (kv-tile, seq, kv-head); recomputes P from q,k,lse;
+ *     dk/dv accumulated in registers and written EXCLUSIVELY to the dk/dv
+ *     pointers (pre-offset views into the packed dqkv gradient buffer,
+ *     strides as fwd); dq accumulated into dq_acc (T,H,D) fp32 with
+ *     device-scope atomics (dq_acc must be zero-initialized).
+ *  3. dq_finalize: casts dq_acc into the q slots of the packed dqkv buffer.
+ */
+int dolomite_fa_bwd_preprocess(dolomite_stream_t stream,
+                               const void* o, const void* dout, float* delta,
+                               int64_t T, int H, int D,
+                               int64_t o_tstride, int64_t do_tstride, int dtype);
+int dolomite_fa_varlen_bwd(dolomite_stream_t stream,
+                           const void* q, const void* k, const void* v,
+                           const void* dout, const float* lse,
+                           const float* delta, float* dq_acc, void* dk, void* dv,
+                           const int32_t* cu_seqlens, int batch, int max_seqlen, int64_t T,
+                           int H, int Hkv, int D, int G,
+                           int64_t q_tstride, int64_t q_gstride,
+                           int64_t k_tstride, int64_t k_hstride,
+                           int64_t v_tstride, int64_t v_hstride,
+                           int64_t do_tstride,
+                           float scale, int dtype);
+int dolomite_fa_dq_finalize(dolomite_stream_t stream,
+                            const float* dq_acc, void* dqkv_q,
+                            int64_t T, int H, int D, int G,
+                            int64_t q_tstride, int64_t q_gstride, int dtype);
+
+/* MFMA fragment-layout self-test: computes C = A(16x32) @ B(32x16) with the
+ * compiled fragment mapping; host verifies against a CPU matmul.
+ * A/B bf16 row-major contiguous, C fp32 row-major. */
+int dolomite_mfma_probe(dolomite_stream_t stream,
+                        const void* A, const void* B, float* C);
+
+/* ------------------------------------------------------------------------
+ * Fused softmax cross-entropy over the vocab, forward / backward.
+ * Replaces F.cross_entropy at model_wrapper/pretraining.py:125 and
+ * gpt_dolomite/main.py:200 (mean reduction, ignore_index -100, fp32
+ * log-softmax as torch does for bf16 inputs).
+ * fwd: per-row loss (T,) fp32 (0 for ignored rows), per-row lse (T,) fp32,
+ *      n_valid is computed by the caller.
+ * bwd: dlogits = (softmax - onehot) * grad_scale, 0 for ignored rows;
+ *      dlogits may alias logits (in-place).
+ * logits rows at logits_ptr + t*row_stride.
+ * ---------------------------------------------------------------------- */
+int dolomite_ce_fwd(dolomite_stream_t stream,
+                    const void* logits, const int64_t* labels,
+                    float* row_loss, float* lse,
+                    int64_t T, int64_t V, int64_t row_stride,
+                    int ignore_index, int dtype);
+int dolomite_ce_bwd(dolomite_stream_t stream,
+                    const void* logits, const int64_t* labels, const float* lse,
+                    void* dlogits, float grad_scale,
+                    int64_t T, int64_t V, int64_t row_stride,
+                    int ignore_index, int dtype);
+
+/* ------------------------------------------------------------------------
+ * Fused AdamW on a flat fp32 master shard, with bf16 param write-out.
+ * Replaces the TorchAdamW step (optimization/optimizer.py:74, defaults
+ * arguments.py:235-249) on the ZeRO-2 local shard. torch.optim.AdamW
+ * semantics: decoupled wd, bias-corrected moments.
+ *   master/m/v: (n,) fp32.  grad: (n,) fp32 or bf16 (grad_dtype).
+ *   param_out: (n,) bf16 copy of updated master (NULL to skip).
+ * ---------------------------------------------------------------------- */
+int dolomite_adamw_step(dolomite_stream_t stream,
+                        float* master, void* param_out_bf16,
+                        const void* grad, int grad_dtype,
+                        float* m, float* v,
+                        int64_t n, float lr, float beta1, float beta2,
+                        float eps, float weight_decay, int step);
+
+/* Multiply a flat fp32 (or bf16) buffer by a scalar (grad clip apply). */
+int dolomite_scale_inplace(dolomite_stream_t stream, void* buf, int64_t n,
+                           float scale, int dtype);
+
+#ifdef __cplusplus
+}
+#endif
+
+#endif /* DOLOMITE_HIP_H */
