@@ -41,7 +41,14 @@ __global__ void __launch_bounds__(256) rmsnorm_fwd_kernel(
         for (int k = 0; k < V; ++k) {
             if (c + k < H) {
                 float v = load_as_f32(xr + c + k);
-                if (rr) v += load_as_f32(rr + c + k);
+                if (rr) {
+                    // the residual add is a tensor-dtype add in the reference
+                    // (layer.py:74,87): round the sum like a torch bf16 add
+                    v += load_as_f32(rr + c + k);
+                    T tmp;
+                    store_from_f32(&tmp, v);
+                    v = load_as_f32(&tmp);
+                }
                 row[c + k] = v;
                 ss += v * v;
             }
@@ -242,7 +249,12 @@ __global__ void __launch_bounds__(256) layernorm_fwd_kernel(
         for (int k = 0; k < V; ++k)
             if (c + k < H) {
                 float v = load_as_f32(xr + c + k);
-                if (rr) v += load_as_f32(rr + c + k);
+                if (rr) {
+                    v += load_as_f32(rr + c + k);
+                    T tmp;
+                    store_from_f32(&tmp, v);
+                    v = load_as_f32(&tmp);
+                }
                 row[c + k] = v;
                 sum += v;
             }
