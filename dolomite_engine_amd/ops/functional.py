@@ -112,7 +112,8 @@ class FusedRMSNorm(torch.autograd.Function):
             y = torch.empty_like(x2)
             s = torch.empty_like(x2) if residual is not None else x2
             rstd = torch.empty(x2.shape[0], dtype=torch.float32, device=x2.device)
-            hip.check(
+            with hip.prof("rmsnorm_fwd"):
+             hip.check(
                 hip.lib().dolomite_rmsnorm_fwd(
                     hip.stream(), hip.ptr(x2), hip.ptr(res2), hip.ptr(weight),
                     hip.ptr(y), hip.ptr(s) if residual is not None else hip.ptr(None),
@@ -137,7 +138,8 @@ class FusedRMSNorm(torch.autograd.Function):
             dx = torch.empty_like(s)
             nb = hip.lib().dolomite_rmsnorm_bwd_nblocks(s.shape[0])
             dw_partial = torch.empty(nb, H, dtype=torch.float32, device=s.device)
-            hip.check(
+            with hip.prof("rmsnorm_bwd"):
+             hip.check(
                 hip.lib().dolomite_rmsnorm_bwd(
                     hip.stream(), hip.ptr(dy2), hip.ptr(s), hip.ptr(weight),
                     hip.ptr(rstd), hip.ptr(dx), hip.ptr(dw_partial),
@@ -382,7 +384,8 @@ class VarlenAttention(torch.autograd.Function):
             qkv = qkv.contiguous()
             o = torch.empty(T, lo.H * lo.D, dtype=qkv.dtype, device=qkv.device)
             lse = torch.empty(lo.H, T, dtype=torch.float32, device=qkv.device)
-            hip.check(
+            with hip.prof("fa_varlen_fwd"):
+             hip.check(
                 hip.lib().dolomite_fa_varlen_fwd(
                     hip.stream(),
                     hip.ptr(qkv), hip.ptr(qkv, lo.k_off), hip.ptr(qkv, lo.v_off),
@@ -419,7 +422,8 @@ class VarlenAttention(torch.autograd.Function):
             )
             dq_acc = torch.zeros(T, lo.H, lo.D, dtype=torch.float32, device=qkv.device)
             dqkv = torch.empty_like(qkv)
-            hip.check(
+            with hip.prof("fa_varlen_bwd"):
+             hip.check(
                 hip.lib().dolomite_fa_varlen_bwd(
                     hip.stream(),
                     hip.ptr(qkv), hip.ptr(qkv, lo.k_off), hip.ptr(qkv, lo.v_off),
@@ -497,7 +501,8 @@ class FusedCrossEntropy(torch.autograd.Function):
             assert logits.stride(1) == 1
             row_loss = torch.empty(T, dtype=torch.float32, device=logits.device)
             lse = torch.empty(T, dtype=torch.float32, device=logits.device)
-            hip.check(
+            with hip.prof("ce_fwd"):
+             hip.check(
                 hip.lib().dolomite_ce_fwd(
                     hip.stream(), hip.ptr(logits), hip.ptr(labels), hip.ptr(row_loss),
                     hip.ptr(lse), T, V, row_stride, ignore_index, hip.dt(logits),
@@ -523,7 +528,8 @@ class FusedCrossEntropy(torch.autograd.Function):
         gs = float(gout) / float(n_valid)
         if logits.is_cuda:
             dlogits = torch.empty_like(logits)
-            hip.check(
+            with hip.prof("ce_bwd"):
+             hip.check(
                 hip.lib().dolomite_ce_bwd(
                     hip.stream(), hip.ptr(logits), hip.ptr(labels), hip.ptr(lse),
                     hip.ptr(dlogits), gs, T, V, logits.stride(0), ctx.ignore_index, hip.dt(logits),

@@ -105,3 +105,56 @@ def dt(t: torch.Tensor) -> int:
     if t.dtype == torch.float32:
         return F32
     raise TypeError(f"dolomite_hip ops support fp32/bf16, got {t.dtype}")
+
+
+# ---------------------------------------------------------------------------
+# Lightweight per-op HIP-event profiling (bench.py roofline evidence).
+# ---------------------------------------------------------------------------
+
+_PROF: dict | None = None
+
+
+def enable_profiling() -> None:
+    global _PROF
+    _PROF = {}
+
+
+def disable_profiling() -> None:
+    global _PROF
+    _PROF = None
+
+
+class _Region:
+    __slots__ = ("name", "start")
+
+    def __init__(self, name):
+        self.name = name
+
+    def __enter__(self):
+        if _PROF is not None:
+            self.start = torch.cuda.Event(enable_timing=True)
+            self.start.record()
+        return self
+
+    def __exit__(self, *a):
+        if _PROF is not None:
+            stop = torch.cuda.Event(enable_timing=True)
+            stop.record()
+            _PROF.setdefault(self.name, []).append((self.start, stop))
+        return False
+
+
+def prof(name: str) -> _Region:
+    return _Region(name)
+
+
+def collect_profile() -> dict:
+    """{name: {"count": N, "total_ms": t, "avg_ms": t/N}} — syncs the device."""
+    if _PROF is None:
+        return {}
+    torch.cuda.synchronize()
+    out = {}
+    for name, pairs in _PROF.items():
+        total = sum(s.elapsed_time(e) for s, e in pairs)
+        out[name] = {"count": len(pairs), "total_ms": total, "avg_ms": total / len(pairs)}
+    return out

@@ -1,0 +1,146 @@
+"""Pretraining entry point — drop-in for `python -m dolomite_engine.pretrain
+--config x.yml` (reference pretrain.py:283-375), launched under torchrun with
+one rank per GPU over RCCL."""
+
+import time
+
+import torch
+
+from .arguments import Mode, TrainingArgs, parse_args
+from .checkpointing import load_checkpoint_for_training, save_checkpoint
+from .data import SyntheticPretrainingDataLoader
+from .model_wrapper import ModelWrapperForPretraining
+from .optimization import get_scheduler
+from .train_utils import get_model_tflops, train_step
+from .utils import get_rank, get_world_size, init_distributed, log_rank_0
+from .zero import ZeRO2Engine
+
+
+def build_engine(args: TrainingArgs):
+    """model wrapper + ZeRO-2 engine + scheduler from TrainingArgs."""
+    tp = args.training_parameters
+    model_wrapper = ModelWrapperForPretraining(
+        micro_batch_size=tp.micro_batch_size,
+        sequence_length=tp.sequence_length,
+        reset_attention_mask=args.model_args.reset_attention_mask,
+        reset_position_ids=args.model_args.reset_position_ids,
+        model_name=args.model_args.model_name,
+        pretrained_config=args.model_args.pretrained_config,
+        dtype=args.mixed_precision_args.dtype,
+        attention_implementation=(
+            args.model_args.attention_implementation.value
+            if args.model_args.attention_implementation is not None
+            else "sdpa"
+        ),
+        use_padding_free_transformer=args.model_args.use_padding_free_transformer,
+    )
+    if torch.cuda.is_available():
+        model_wrapper.model.cuda()
+
+    oa = args.optimizer_args.class_args
+    engine = ZeRO2Engine(
+        model_wrapper.model,
+        lr=oa.get("lr", 1e-5),
+        betas=tuple(oa.get("betas", (0.9, 0.95))),
+        eps=oa.get("eps", 1e-10),
+        weight_decay=oa.get("weight_decay", 0.1),
+        overlap_comm=args.distributed_args.overlap_comm,
+    )
+    lr_scheduler = get_scheduler(oa.get("lr", 1e-5), args.lr_scheduler_args, tp.num_training_steps)
+    return model_wrapper, engine, lr_scheduler
+
+
+def train(args: TrainingArgs, model_wrapper, engine, lr_scheduler, train_loader, starting_step=0, metadata=None):
+    tp = args.training_parameters
+    ga = tp.gradient_accumulation_steps
+    world = get_world_size()
+    tokens_per_step = tp.micro_batch_size * tp.sequence_length * ga * world
+    tflops_per_step = (
+        get_model_tflops(
+            model_wrapper.config,
+            tp.micro_batch_size,
+            tp.sequence_length,
+            args.distributed_args.gradient_checkpointing_method is not None,
+        )
+        * ga
+        * world
+    )
+
+    train_iter = iter(train_loader)
+    model_wrapper.train()
+
+    for global_step in range(starting_step + 1, tp.num_training_steps + 1):
+        t0 = time.perf_counter()
+        loss, grad_norm = train_step(model_wrapper, engine, lr_scheduler, train_iter, ga, tp.gradient_clipping)
+        if torch.cuda.is_available():
+            torch.cuda.synchronize()
+        dt = time.perf_counter() - t0
+
+        if global_step % args.logging_args.log_interval == 0:
+            log_rank_0(
+                f"step = {global_step}, loss_step = {loss:.5f}, grad_norm = {grad_norm:.3f}, "
+                f"learning_rate = {lr_scheduler.get_lr():.3e}, step time (sec) = {dt:.3f}, "
+                f"throughput = {tokens_per_step / dt:.1f} tokens/s, FLOPS = {tflops_per_step / dt:.1f} TFLOPs"
+            )
+
+        if args.save_args is not None and args.save_args.save_interval and global_step % args.save_args.save_interval == 0:
+            save_checkpoint(
+                args.save_args.save_path,
+                global_step,
+                model_wrapper,
+                engine if args.save_args.save_optimizer else None,
+                lr_scheduler,
+                train_loader,
+                metadata={"consumed_samples": global_step * tp.micro_batch_size * ga * world},
+                args_dict=args.to_dict(),
+                save_optimizer=args.save_args.save_optimizer,
+            )
+
+    if args.save_args is not None:
+        save_checkpoint(
+            args.save_args.save_path,
+            tp.num_training_steps,
+            model_wrapper,
+            engine if args.save_args.save_optimizer else None,
+            lr_scheduler,
+            train_loader,
+            metadata={"consumed_samples": tp.num_training_steps * tp.micro_batch_size * ga * world},
+            args_dict=args.to_dict(),
+            save_optimizer=args.save_args.save_optimizer,
+        )
+
+
+def main(argv=None):
+    args = parse_args(argv)
+    init_distributed(args.distributed_args.timeout_minutes)
+    torch.manual_seed(args.random_args.seed + get_rank())
+
+    model_wrapper, engine, lr_scheduler = build_engine(args)
+
+    tp = args.training_parameters
+    train_loader = SyntheticPretrainingDataLoader(
+        tp.micro_batch_size, tp.sequence_length, model_wrapper.config.vocab_size, seed=args.random_args.seed
+    )
+
+    starting_step, metadata = 0, None
+    if args.load_args is not None:
+        starting_step, metadata = load_checkpoint_for_training(
+            args.load_args.load_path,
+            model_wrapper,
+            engine,
+            lr_scheduler,
+            train_loader,
+            iteration=args.load_args.iteration,
+            load_optimizer=args.load_args.load_optimizer,
+            load_lr_scheduler=args.load_args.load_lr_scheduler,
+            load_rng_state=args.load_args.load_rng_state,
+            load_dataloader_state=args.load_args.load_dataloader_state,
+        )
+        if not args.load_args.load_starting_iteration:
+            starting_step = 0
+
+    train(args, model_wrapper, engine, lr_scheduler, train_loader, starting_step, metadata)
+
+
+if __name__ == "__main__":
+    main()
