@@ -577,6 +577,12 @@ class GPTDolomiteForCausalLM(GPTDolomitePreTrainedModel):
     def set_input_embeddings(self, value):
         self.transformer.wte = value
 
+    def tie_weights(self, *args, **kwargs):
+        # when tied, logits are computed as F.linear(hs, wte.weight) directly
+        # (reference main.py:172-177) — there is no lm_head module to tie
+        if not self._tied_word_embeddings:
+            super().tie_weights(*args, **kwargs)
+
     def get_output_embeddings(self):
         if not self._tied_word_embeddings:
             return self.lm_head
