@@ -61,6 +61,7 @@ def build(args, device):
         dtype="bf16" if device.type == "cuda" else "fp32",
         attention_implementation="flash_attention_2" if device.type == "cuda" else "eager",
         use_padding_free_transformer=device.type == "cuda",
+        device=device,  # construct + init directly on the GPU
     )
     wrapper.model.to(device)
     engine = ZeRO2Engine(wrapper.model, lr=1e-5, betas=(0.9, 0.95), eps=1e-10, weight_decay=0.1)

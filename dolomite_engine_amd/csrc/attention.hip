@@ -311,7 +311,7 @@ template <int DPAD>
 __global__ void __launch_bounds__(256) fa_bwd_kernel(
     const __bf16* __restrict__ q, const __bf16* __restrict__ k, const __bf16* __restrict__ v,
     const __bf16* __restrict__ dout, const float* __restrict__ lse, const float* __restrict__ delta,
-    float* __restrict__ dq_acc, __bf16* __restrict__ dk, __bf16* __restrict__ dv,
+    float* __restrict__ dq_acc, float* __restrict__ dk_acc, float* __restrict__ dv_acc,
     const int32_t* __restrict__ cu, int H, int Hkv, int D, int G,
     int64_t q_ts, int64_t q_gs, int64_t k_ts, int64_t k_hs, int64_t v_ts, int64_t v_hs,
     int64_t do_ts, int64_t T_total, float scale) {
@@ -321,7 +321,8 @@ __global__ void __launch_bounds__(256) fa_bwd_kernel(
     constexpr int ST = 64 + 8;     // transposed [d][q] / [q][key] image stride
 
     const int b = blockIdx.y;
-    const int kvh = blockIdx.z;
+    const int h = blockIdx.z;        // q head: one workgroup per (kv-tile, q-head)
+    const int kvh = h / G;
     const int s0 = cu[b];
     const int L = cu[b + 1] - s0;
     const int ks = (int)blockIdx.x * 64;
@@ -370,21 +371,20 @@ __global__ void __launch_bounds__(256) fa_bwd_kernel(
         }
     }
 
-    f32x4 dv_acc[DCH], dk_acc[DCH];
+    f32x4 dvr[DCH], dkr[DCH];
 #pragma unroll
     for (int dc = 0; dc < DCH; ++dc) {
-        dv_acc[dc] = {0.f, 0.f, 0.f, 0.f};
-        dk_acc[dc] = {0.f, 0.f, 0.f, 0.f};
+        dvr[dc] = {0.f, 0.f, 0.f, 0.f};
+        dkr[dc] = {0.f, 0.f, 0.f, 0.f};
     }
 
     const int qt0 = ks / 64;
     const int nqt = (L + 63) / 64;
 
+    const int64_t q_hoff = (int64_t)(h / G) * q_gs + (int64_t)(h % G) * D;
     for (int qt = qt0; qt < nqt; ++qt) {
         const int qs = qt * 64;
-        for (int hg = 0; hg < G; ++hg) {
-            const int h = kvh * G + hg;
-            const int64_t q_hoff = (int64_t)kvh * q_gs + (int64_t)hg * D;
+        {
             // --- stage Q and dO tiles (both images each) ---
             __syncthreads();  // previous iteration's reads done
             {
@@ -451,9 +451,9 @@ __global__ void __launch_bounds__(256) fa_bwd_kernel(
 #pragma unroll
                 for (int dc = 0; dc < DCH; ++dc) {
                     bf16x8 dotb = *(const bf16x8*)&dOTl[(dc * 16 + lr) * ST + kc2 * 32 + lg * 8];
-                    dv_acc[dc] = MFMA16(ptf, dotb, dv_acc[dc]);
+                    dvr[dc] = MFMA16(ptf, dotb, dvr[dc]);
                     bf16x8 qtb = *(const bf16x8*)&QTl[(dc * 16 + lr) * ST + kc2 * 32 + lg * 8];
-                    dk_acc[dc] = MFMA16(dstf, qtb, dk_acc[dc]);
+                    dkr[dc] = MFMA16(dstf, qtb, dkr[dc]);
                 }
             }
 
@@ -478,7 +478,7 @@ __global__ void __launch_bounds__(256) fa_bwd_kernel(
         }
     }
 
-    // --- store dK/dV (exclusive ownership of this kv strip) ---
+    // --- accumulate dK/dV (G q-head workgroups contribute per kv strip) ---
 #pragma unroll
     for (int dc = 0; dc < DCH; ++dc) {
 #pragma unroll
@@ -486,8 +486,9 @@ __global__ void __launch_bounds__(256) fa_bwd_kernel(
             const int kpos = ks + wave * 16 + lg * 4 + r;
             const int d = dc * 16 + lr;
             if (kpos < kend && d < D) {
-                dk[(int64_t)(s0 + kpos) * k_ts + (int64_t)kvh * k_hs + d] = (__bf16)dk_acc[dc][r];
-                dv[(int64_t)(s0 + kpos) * v_ts + (int64_t)kvh * v_hs + d] = (__bf16)dv_acc[dc][r];
+                int64_t idx = ((int64_t)(s0 + kpos) * Hkv + kvh) * D + d;
+                atomicAdd(&dk_acc[idx], dkr[dc][r]);
+                atomicAdd(&dv_acc[idx], dvr[dc][r]);
             }
         }
     }
@@ -496,15 +497,15 @@ __global__ void __launch_bounds__(256) fa_bwd_kernel(
 template <int DPAD>
 static int launch_fa_bwd(hipStream_t stream, const __bf16* q, const __bf16* k, const __bf16* v,
                          const __bf16* dout, const float* lse, const float* delta,
-                         float* dq_acc, __bf16* dk, __bf16* dv,
+                         float* dq_acc, float* dk_acc, float* dv_acc,
                          const int32_t* cu, int batch, int64_t T, int H, int Hkv, int D, int G,
                          int64_t q_ts, int64_t q_gs, int64_t k_ts, int64_t k_hs,
                          int64_t v_ts, int64_t v_hs, int64_t do_ts, int max_tiles, float scale) {
-    dim3 grid(max_tiles, batch, Hkv), block(256);
+    dim3 grid(max_tiles, batch, H), block(256);
     constexpr int SQ = DPAD + 8, ST = 64 + 8;
     size_t shmem = (size_t)(64 * SQ * 2 + DPAD * ST * 3 + 64 * ST * 3) * sizeof(__bf16);
     hipLaunchKernelGGL((fa_bwd_kernel<DPAD>), grid, block, shmem, stream,
-                       q, k, v, dout, lse, delta, dq_acc, dk, dv, cu, H, Hkv, D, G,
+                       q, k, v, dout, lse, delta, dq_acc, dk_acc, dv_acc, cu, H, Hkv, D, G,
                        q_ts, q_gs, k_ts, k_hs, v_ts, v_hs, do_ts, T, scale);
     return dol_last_error();
 }
@@ -512,7 +513,7 @@ static int launch_fa_bwd(hipStream_t stream, const __bf16* q, const __bf16* k, c
 extern "C" int dolomite_fa_varlen_bwd(dolomite_stream_t stream,
                                       const void* q, const void* k, const void* v,
                                       const void* dout, const float* lse,
-                                      const float* delta, float* dq_acc, void* dk, void* dv,
+                                      const float* delta, float* dq_acc, float* dk_acc, float* dv_acc,
                                       const int32_t* cu_seqlens, int batch, int max_seqlen, int64_t T,
                                       int H, int Hkv, int D, int G,
                                       int64_t q_tstride, int64_t q_gstride,
@@ -526,7 +527,7 @@ extern "C" int dolomite_fa_varlen_bwd(dolomite_stream_t stream,
     hipStream_t s = (hipStream_t)stream;
 #define CASE(DP)                                                                                        \
     return launch_fa_bwd<DP>(s, (const __bf16*)q, (const __bf16*)k, (const __bf16*)v,                   \
-                             (const __bf16*)dout, lse, delta, dq_acc, (__bf16*)dk, (__bf16*)dv,         \
+                             (const __bf16*)dout, lse, delta, dq_acc, dk_acc, dv_acc,                   \
                              cu_seqlens, batch, T, H, Hkv, D, G, q_tstride, q_gstride, k_tstride,       \
                              k_hstride, v_tstride, v_hstride, do_tstride, max_tiles, scale)
     if (D <= 32) CASE(32);
@@ -537,35 +538,54 @@ extern "C" int dolomite_fa_varlen_bwd(dolomite_stream_t stream,
 }
 
 // ===========================================================================
-// Backward pass 3: cast dq_acc into the packed dqkv q-slots.
+// Backward pass 3: cast dq/dk/dv fp32 accumulators into the packed dqkv.
 // ===========================================================================
 
 template <typename T>
-__global__ void __launch_bounds__(256) fa_dq_finalize_kernel(
-    const float* __restrict__ dq_acc, T* __restrict__ dqkv_q,
-    int64_t total, int H, int D, int G, int64_t q_ts, int64_t q_gs) {
+__global__ void __launch_bounds__(256) fa_grad_finalize_kernel(
+    const float* __restrict__ dq_acc, const float* __restrict__ dk_acc, const float* __restrict__ dv_acc,
+    T* __restrict__ dqkv, int64_t total_q, int64_t total_kv,
+    int H, int Hkv, int D, int G, int64_t row_ts, int64_t q_gs, int64_t k_off, int64_t kv_hs, int64_t v_off) {
     int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-    if (idx >= total) return;
-    int64_t t = idx / ((int64_t)H * D);
-    int rem = (int)(idx % ((int64_t)H * D));
-    int h = rem / D;
+    if (idx < total_q) {
+        int64_t t = idx / ((int64_t)H * D);
+        int rem = (int)(idx % ((int64_t)H * D));
+        int h = rem / D;
+        int d = rem % D;
+        store_from_f32(&dqkv[t * row_ts + (int64_t)(h / G) * q_gs + (int64_t)(h % G) * D + d], dq_acc[idx]);
+        return;
+    }
+    int64_t kidx = idx - total_q;
+    bool is_v = kidx >= total_kv;
+    if (is_v) kidx -= total_kv;
+    if (kidx >= total_kv) return;
+    int64_t t = kidx / ((int64_t)Hkv * D);
+    int rem = (int)(kidx % ((int64_t)Hkv * D));
+    int j = rem / D;
     int d = rem % D;
-    store_from_f32(&dqkv_q[t * q_ts + (int64_t)(h / G) * q_gs + (int64_t)(h % G) * D + d], dq_acc[idx]);
+    const float* src = is_v ? dv_acc : dk_acc;
+    int64_t off = is_v ? v_off : k_off;
+    store_from_f32(&dqkv[t * row_ts + off + (int64_t)j * kv_hs + d], src[kidx]);
 }
 
-extern "C" int dolomite_fa_dq_finalize(dolomite_stream_t stream,
-                                       const float* dq_acc, void* dqkv_q,
-                                       int64_t T, int H, int D, int G,
-                                       int64_t q_tstride, int64_t q_gstride, int dtype) {
-    int64_t total = T * (int64_t)H * D;
+extern "C" int dolomite_fa_grad_finalize(dolomite_stream_t stream,
+                                         const float* dq_acc, const float* dk_acc, const float* dv_acc,
+                                         void* dqkv, int64_t T, int H, int Hkv, int D, int G,
+                                         int64_t row_tstride, int64_t q_gstride,
+                                         int64_t k_off, int64_t kv_hstride, int64_t v_off, int dtype) {
+    int64_t total_q = T * (int64_t)H * D;
+    int64_t total_kv = T * (int64_t)Hkv * D;
+    int64_t total = total_q + 2 * total_kv;
     if (total == 0) return 0;
     dim3 grid((uint32_t)((total + 255) / 256)), block(256);
     if (dtype == DOLOMITE_BF16)
-        hipLaunchKernelGGL((fa_dq_finalize_kernel<uint16_t>), grid, block, 0, (hipStream_t)stream,
-                           dq_acc, (uint16_t*)dqkv_q, total, H, D, G, q_tstride, q_gstride);
+        hipLaunchKernelGGL((fa_grad_finalize_kernel<uint16_t>), grid, block, 0, (hipStream_t)stream,
+                           dq_acc, dk_acc, dv_acc, (uint16_t*)dqkv, total_q, total_kv,
+                           H, Hkv, D, G, row_tstride, q_gstride, k_off, kv_hstride, v_off);
     else
-        hipLaunchKernelGGL((fa_dq_finalize_kernel<float>), grid, block, 0, (hipStream_t)stream,
-                           dq_acc, (float*)dqkv_q, total, H, D, G, q_tstride, q_gstride);
+        hipLaunchKernelGGL((fa_grad_finalize_kernel<float>), grid, block, 0, (hipStream_t)stream,
+                           dq_acc, dk_acc, dv_acc, (float*)dqkv, total_q, total_kv,
+                           H, Hkv, D, G, row_tstride, q_gstride, k_off, kv_hstride, v_off);
     return dol_last_error();
 }
 

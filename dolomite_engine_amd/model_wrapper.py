@@ -21,11 +21,13 @@ class ModelWrapper(nn.Module):
         attention_implementation: str,
         use_padding_free_transformer: bool,
         efficient_initialization: bool = False,
+        device: torch.device | str | None = None,
     ):
         super().__init__()
         if isinstance(dtype, str):
             dtype = string_to_torch_dtype(dtype)
         self.dtype = dtype
+        self._init_device = torch.device(device) if device is not None else None
         self.attention_implementation = attention_implementation
         self.use_padding_free_transformer = use_padding_free_transformer
 
@@ -38,7 +40,11 @@ class ModelWrapper(nn.Module):
             config = GPTDolomiteConfig(**pretrained_config)
             config._attn_implementation = attention_implementation
             self.config = config
-            self.model = GPTDolomiteForCausalLM(config, use_padding_free_transformer=use_padding_free_transformer)
+            import contextlib
+
+            dev_ctx = torch.device(self._init_device) if self._init_device is not None else contextlib.nullcontext()
+            with dev_ctx:
+                self.model = GPTDolomiteForCausalLM(config, use_padding_free_transformer=use_padding_free_transformer)
         else:
             config = GPTDolomiteConfig.from_pretrained(model_name)
             config._attn_implementation = attention_implementation

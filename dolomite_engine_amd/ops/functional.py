@@ -421,6 +421,8 @@ class VarlenAttention(torch.autograd.Function):
                 "fa_bwd_preprocess",
             )
             dq_acc = torch.zeros(T, lo.H, lo.D, dtype=torch.float32, device=qkv.device)
+            dk_acc = torch.zeros(T, lo.Hkv, lo.D, dtype=torch.float32, device=qkv.device)
+            dv_acc = torch.zeros(T, lo.Hkv, lo.D, dtype=torch.float32, device=qkv.device)
             dqkv = torch.empty_like(qkv)
             with hip.prof("fa_varlen_bwd"):
              hip.check(
@@ -428,7 +430,7 @@ class VarlenAttention(torch.autograd.Function):
                     hip.stream(),
                     hip.ptr(qkv), hip.ptr(qkv, lo.k_off), hip.ptr(qkv, lo.v_off),
                     hip.ptr(dout), hip.ptr(lse), hip.ptr(delta), hip.ptr(dq_acc),
-                    hip.ptr(dqkv, lo.k_off), hip.ptr(dqkv, lo.v_off),
+                    hip.ptr(dk_acc), hip.ptr(dv_acc),
                     hip.ptr(cu), batch, ctx.max_seqlen, T, lo.H, lo.Hkv, lo.D, lo.G,
                     lo.row_len, lo.q_gstride, lo.row_len, lo.kv_hstride,
                     lo.row_len, lo.kv_hstride, lo.H * lo.D, float(ctx.scale), hip.dt(qkv),
@@ -436,11 +438,12 @@ class VarlenAttention(torch.autograd.Function):
                 "fa_varlen_bwd",
             )
             hip.check(
-                hip.lib().dolomite_fa_dq_finalize(
-                    hip.stream(), hip.ptr(dq_acc), hip.ptr(dqkv),
-                    T, lo.H, lo.D, lo.G, lo.row_len, lo.q_gstride, hip.dt(qkv),
+                hip.lib().dolomite_fa_grad_finalize(
+                    hip.stream(), hip.ptr(dq_acc), hip.ptr(dk_acc), hip.ptr(dv_acc), hip.ptr(dqkv),
+                    T, lo.H, lo.Hkv, lo.D, lo.G, lo.row_len, lo.q_gstride,
+                    lo.k_off, lo.kv_hstride, lo.v_off, hip.dt(qkv),
                 ),
-                "fa_dq_finalize",
+                "fa_grad_finalize",
             )
             return dqkv, None, None, None, None
         # CPU: differentiate the CPU restatement with torch autograd
