@@ -93,9 +93,13 @@ __global__ void __launch_bounds__(256) fa_fwd_kernel(
 
     extern __shared__ char smem_raw[];
     __bf16* Klds = (__bf16*)smem_raw;              // [64][SK]
-    __bf16* Vlds = Klds + 64 * SK;                 // [DPAD][SV] (transposed)
+    __bf16* Vlds = Klds + 64 * SK;                 // [DPAD][SV] (transposed, swizzled)
     __bf16* Plds = Vlds + DPAD * SV;               // [4 waves][16][SV]
     __bf16* Pw = Plds + wave * 16 * SV;
+
+    // XOR block swizzle for the transposed V image (see fa_bwd_kernel)
+#define VSWZ(row, col) ((row) * SV + (((((col) >> 3) ^ (((row) >> 3) & 7)) << 3) | ((col) & 7)))
+#define VSWZ8(row, col0) ((row) * SV + (((((col0) >> 3) ^ (((row) >> 3) & 7)) << 3)))
 
     const int kvh = h / G;
     const int64_t q_hoff = (int64_t)(h / G) * q_gs + (int64_t)(h % G) * D;
@@ -136,7 +140,7 @@ __global__ void __launch_bounds__(256) fa_fwd_kernel(
                 const __bf16* vp = v + (int64_t)(s0 + (kv_valid ? ks + key : 0)) * v_ts + (int64_t)kvh * v_hs + d0;
                 bf16x8 vv = load_bf16x8_guard(vp, d0, D, kv_valid);
 #pragma unroll
-                for (int e = 0; e < 8; ++e) Vlds[(d0 + e) * SV + key] = vv[e];
+                for (int e = 0; e < 8; ++e) Vlds[VSWZ(d0 + e, key)] = vv[e];
             }
         }
         __syncthreads();
@@ -199,7 +203,7 @@ __global__ void __launch_bounds__(256) fa_fwd_kernel(
             bf16x8 pf = *(const bf16x8*)&Pw[lr * SV + kc2 * 32 + lg * 8];
 #pragma unroll
             for (int dc = 0; dc < DCH; ++dc) {
-                bf16x8 vf = *(const bf16x8*)&Vlds[(dc * 16 + lr) * SV + kc2 * 32 + lg * 8];
+                bf16x8 vf = *(const bf16x8*)&Vlds[VSWZ8(dc * 16 + lr, kc2 * 32 + lg * 8)];
                 o_acc[dc] = MFMA16(pf, vf, o_acc[dc]);
             }
         }
@@ -225,6 +229,8 @@ __global__ void __launch_bounds__(256) fa_fwd_kernel(
                 o[(int64_t)(s0 + qpos) * o_ts + (int64_t)h * D + d] = (__bf16)(o_acc[dc][r] * inv_l[r]);
         }
     }
+#undef VSWZ
+#undef VSWZ8
 }
 
 template <int DPAD>
@@ -317,8 +323,7 @@ __global__ void __launch_bounds__(256) fa_bwd_kernel(
     int64_t do_ts, int64_t T_total, float scale) {
     constexpr int KCH = DPAD / 32;
     constexpr int DCH = DPAD / 16;
-    constexpr int SQ = DPAD + 8;   // [q][d] and [key][d] image stride
-    constexpr int ST = 64 + 8;     // transposed [d][q] / [q][key] image stride
+    constexpr int ST = 64 + 8;     // transposed-image / [q][key] row stride
 
     const int b = blockIdx.y;
     const int h = blockIdx.z;        // q head: one workgroup per (kv-tile, q-head)
@@ -333,15 +338,21 @@ __global__ void __launch_bounds__(256) fa_bwd_kernel(
     const int lr = lane & 15;
     const int lg = lane >> 4;
 
+    // Transposed images use an XOR block swizzle: element (row, col) lives at
+    // row*ST + ((col>>3 ^ (row>>3)&7)<<3) + (col&7). This keeps every
+    // ds_read_b128 16-byte aligned while spreading the 8-consecutive-row
+    // scatter writes of the transpose staging over 8 bank groups (the linear
+    // layout put all of them on one bank: 12-way conflict).
     extern __shared__ char smem_raw[];
-    __bf16* Qlds = (__bf16*)smem_raw;          // [64 q][SQ]
-    __bf16* QTl = Qlds + 64 * SQ;              // [DPAD][ST]   (Q^T)
-    __bf16* dOl = QTl + DPAD * ST;             // [64 q][SQ]
-    __bf16* dOTl = dOl + 64 * SQ;              // [DPAD][ST]   (dO^T)
-    __bf16* KTl = dOTl + DPAD * ST;            // [DPAD][ST]   (K^T)
+    __bf16* QTl = (__bf16*)smem_raw;           // [DPAD][ST]   (Q^T, swizzled)
+    __bf16* dOTl = QTl + DPAD * ST;            // [DPAD][ST]   (dO^T, swizzled)
+    __bf16* KTl = dOTl + DPAD * ST;            // [DPAD][ST]   (K^T, swizzled)
     __bf16* dSl = KTl + DPAD * ST;             // [64 q][ST]   (dS, [q][key])
     __bf16* dSTl = dSl + 64 * ST;              // [64 key][ST] (dS^T, [key][q])
     __bf16* PTl = dSTl + 64 * ST;              // [64 key][ST] (P^T, [key][q])
+
+#define SWZ(row, col) ((row) * ST + (((((col) >> 3) ^ (((row) >> 3) & 7)) << 3) | ((col) & 7)))
+#define SWZ8(row, col0) ((row) * ST + (((((col0) >> 3) ^ (((row) >> 3) & 7)) << 3)))
 
     const int kend = min(L, ks + 64);
 
@@ -357,7 +368,7 @@ __global__ void __launch_bounds__(256) fa_bwd_kernel(
         const __bf16* vp = v + (int64_t)(s0 + (kvalid ? krow : 0)) * v_ts + (int64_t)kvh * v_hs + d0;
         vfr[kc] = load_bf16x8_guard(vp, d0, D, kvalid);
     }
-    // --- stage K^T image cooperatively ---
+    // --- stage K^T image once (swizzled scatter) ---
     {
         const int pieces = 64 * DPAD / 8;
         for (int pidx = threadIdx.x; pidx < pieces; pidx += 256) {
@@ -367,7 +378,7 @@ __global__ void __launch_bounds__(256) fa_bwd_kernel(
             const __bf16* kp = k + (int64_t)(s0 + (valid ? ks + key : 0)) * k_ts + (int64_t)kvh * k_hs + d0;
             bf16x8 kk = load_bf16x8_guard(kp, d0, D, valid);
 #pragma unroll
-            for (int e = 0; e < 8; ++e) KTl[(d0 + e) * ST + key] = kk[e];
+            for (int e = 0; e < 8; ++e) KTl[SWZ(d0 + e, key)] = kk[e];
         }
     }
 
@@ -380,100 +391,104 @@ __global__ void __launch_bounds__(256) fa_bwd_kernel(
 
     const int qt0 = ks / 64;
     const int nqt = (L + 63) / 64;
-
     const int64_t q_hoff = (int64_t)(h / G) * q_gs + (int64_t)(h % G) * D;
+    const int64_t do_hoff = (int64_t)h * D;
+
     for (int qt = qt0; qt < nqt; ++qt) {
         const int qs = qt * 64;
+        // --- stage Q^T and dO^T (swizzled transposes; row-major reads of
+        //     Q/dO go straight to global where they are 16B-contiguous) ---
+        __syncthreads();  // previous iteration's reads done
         {
-            // --- stage Q and dO tiles (both images each) ---
-            __syncthreads();  // previous iteration's reads done
-            {
-                const int pieces = 64 * DPAD / 8;
-                for (int pidx = threadIdx.x; pidx < pieces; pidx += 256) {
-                    int qq = pidx / (DPAD / 8);
-                    int d0 = (pidx % (DPAD / 8)) * 8;
-                    bool valid = (qs + qq) < L;
-                    const __bf16* qp = q + (int64_t)(s0 + (valid ? qs + qq : 0)) * q_ts + q_hoff + d0;
-                    bf16x8 qq8 = load_bf16x8_guard(qp, d0, D, valid);
-                    *(bf16x8*)&Qlds[qq * SQ + d0] = qq8;
+            const int pieces = 64 * DPAD / 8;
+            for (int pidx = threadIdx.x; pidx < pieces; pidx += 256) {
+                int qq = pidx / (DPAD / 8);
+                int d0 = (pidx % (DPAD / 8)) * 8;
+                bool valid = (qs + qq) < L;
+                const __bf16* qp = q + (int64_t)(s0 + (valid ? qs + qq : 0)) * q_ts + q_hoff + d0;
+                bf16x8 qq8 = load_bf16x8_guard(qp, d0, D, valid);
 #pragma unroll
-                    for (int e = 0; e < 8; ++e) QTl[(d0 + e) * ST + qq] = qq8[e];
-                    const __bf16* dp = dout + (int64_t)(s0 + (valid ? qs + qq : 0)) * do_ts + (int64_t)h * D + d0;
-                    bf16x8 dd8 = load_bf16x8_guard(dp, d0, D, valid);
-                    *(bf16x8*)&dOl[qq * SQ + d0] = dd8;
+                for (int e = 0; e < 8; ++e) QTl[SWZ(d0 + e, qq)] = qq8[e];
+                const __bf16* dp = dout + (int64_t)(s0 + (valid ? qs + qq : 0)) * do_ts + do_hoff + d0;
+                bf16x8 dd8 = load_bf16x8_guard(dp, d0, D, valid);
 #pragma unroll
-                    for (int e = 0; e < 8; ++e) dOTl[(d0 + e) * ST + qq] = dd8[e];
-                }
+                for (int e = 0; e < 8; ++e) dOTl[SWZ(d0 + e, qq)] = dd8[e];
             }
-            __syncthreads();
+        }
+        __syncthreads();
 
-            // --- S^T = K·Q^T, P^T = exp(scale*S^T - lse) ; dP^T = V·dO^T ---
-            // C layout: col = q = lr, row = key = 16*wave + lg*4 + r
-            f32x4 st[4], dpt[4];
+        // --- S^T = K*Q^T, P^T = exp(scale*S^T - lse) ; dP^T = V*dO^T ---
+        // C layout: col = q = lr, row = key = 16*wave + lg*4 + r
+        // B-frags (Q^T, dO^T as [k=d][j=q]) are 16B rows of Q/dO: read from
+        // global (L2-hot across the 4 waves / q-tile revisits)
+        f32x4 st[4], dpt[4];
 #pragma unroll
-            for (int cb = 0; cb < 4; ++cb) {
-                st[cb] = {0.f, 0.f, 0.f, 0.f};
-                dpt[cb] = {0.f, 0.f, 0.f, 0.f};
+        for (int cb = 0; cb < 4; ++cb) {
+            st[cb] = {0.f, 0.f, 0.f, 0.f};
+            dpt[cb] = {0.f, 0.f, 0.f, 0.f};
+            const int qrow = qs + cb * 16 + lr;
+            const bool qok = qrow < L;
+            const __bf16* qrp = q + (int64_t)(s0 + (qok ? qrow : 0)) * q_ts + q_hoff;
+            const __bf16* drp = dout + (int64_t)(s0 + (qok ? qrow : 0)) * do_ts + do_hoff;
 #pragma unroll
-                for (int kc = 0; kc < KCH; ++kc) {
-                    bf16x8 qtf = *(const bf16x8*)&Qlds[(cb * 16 + lr) * SQ + kc * 32 + lg * 8];
-                    st[cb] = MFMA16(kfr[kc], qtf, st[cb]);
-                    bf16x8 dtf = *(const bf16x8*)&dOl[(cb * 16 + lr) * SQ + kc * 32 + lg * 8];
-                    dpt[cb] = MFMA16(vfr[kc], dtf, dpt[cb]);
-                }
+            for (int kc = 0; kc < KCH; ++kc) {
+                int d0 = kc * 32 + lg * 8;
+                bf16x8 qtf = load_bf16x8_guard(qrp + d0, d0, D, qok);
+                st[cb] = MFMA16(kfr[kc], qtf, st[cb]);
+                bf16x8 dtf = load_bf16x8_guard(drp + d0, d0, D, qok);
+                dpt[cb] = MFMA16(vfr[kc], dtf, dpt[cb]);
             }
+        }
 
 #pragma unroll
-            for (int cb = 0; cb < 4; ++cb) {
-                const int qpos = qs + cb * 16 + lr;
-                const bool qok = qpos < L;
-                float lsev = qok ? lse[(int64_t)h * T_total + s0 + qpos] : 0.f;
-                float delv = qok ? delta[(int64_t)h * T_total + s0 + qpos] : 0.f;
+        for (int cb = 0; cb < 4; ++cb) {
+            const int qpos = qs + cb * 16 + lr;
+            const bool qok = qpos < L;
+            float lsev = qok ? lse[(int64_t)h * T_total + s0 + qpos] : 0.f;
+            float delv = qok ? delta[(int64_t)h * T_total + s0 + qpos] : 0.f;
 #pragma unroll
-                for (int r = 0; r < 4; ++r) {
-                    const int kpos = ks + wave * 16 + lg * 4 + r;
-                    bool ok = qok && kpos < kend && kpos <= qpos;
-                    float pv = ok ? __expf(st[cb][r] * scale - lsev) : 0.f;
-                    float ds = ok ? pv * (dpt[cb][r] - delv) * scale : 0.f;
-                    // stash P^T [key][q] and dS both images
-                    PTl[(wave * 16 + lg * 4 + r) * ST + cb * 16 + lr] = (__bf16)pv;
-                    dSTl[(wave * 16 + lg * 4 + r) * ST + cb * 16 + lr] = (__bf16)ds;
-                    dSl[(cb * 16 + lr) * ST + wave * 16 + lg * 4 + r] = (__bf16)ds;
-                }
+            for (int r = 0; r < 4; ++r) {
+                const int kpos = ks + wave * 16 + lg * 4 + r;
+                bool ok = qok && kpos < kend && kpos <= qpos;
+                float pv = ok ? __expf(st[cb][r] * scale - lsev) : 0.f;
+                float ds = ok ? pv * (dpt[cb][r] - delv) * scale : 0.f;
+                PTl[(wave * 16 + lg * 4 + r) * ST + cb * 16 + lr] = (__bf16)pv;
+                dSTl[(wave * 16 + lg * 4 + r) * ST + cb * 16 + lr] = (__bf16)ds;
+                dSl[(cb * 16 + lr) * ST + wave * 16 + lg * 4 + r] = (__bf16)ds;
             }
-            __syncthreads();
+        }
+        __syncthreads();
 
-            // --- dV += P^T·dO ; dK += dS^T·Q  (contraction over q) ---
+        // --- dV += P^T*dO ; dK += dS^T*Q  (contraction over q) ---
 #pragma unroll
-            for (int kc2 = 0; kc2 < 2; ++kc2) {
-                bf16x8 ptf = *(const bf16x8*)&PTl[(wave * 16 + lr) * ST + kc2 * 32 + lg * 8];
-                bf16x8 dstf = *(const bf16x8*)&dSTl[(wave * 16 + lr) * ST + kc2 * 32 + lg * 8];
-#pragma unroll
-                for (int dc = 0; dc < DCH; ++dc) {
-                    bf16x8 dotb = *(const bf16x8*)&dOTl[(dc * 16 + lr) * ST + kc2 * 32 + lg * 8];
-                    dvr[dc] = MFMA16(ptf, dotb, dvr[dc]);
-                    bf16x8 qtb = *(const bf16x8*)&QTl[(dc * 16 + lr) * ST + kc2 * 32 + lg * 8];
-                    dkr[dc] = MFMA16(dstf, qtb, dkr[dc]);
-                }
-            }
-
-            // --- dQ strip (this wave's 16 q rows): dS·K^T-image, atomics ---
+        for (int kc2 = 0; kc2 < 2; ++kc2) {
+            bf16x8 ptf = *(const bf16x8*)&PTl[(wave * 16 + lr) * ST + kc2 * 32 + lg * 8];
+            bf16x8 dstf = *(const bf16x8*)&dSTl[(wave * 16 + lr) * ST + kc2 * 32 + lg * 8];
 #pragma unroll
             for (int dc = 0; dc < DCH; ++dc) {
-                f32x4 dq = {0.f, 0.f, 0.f, 0.f};
+                bf16x8 dotb = *(const bf16x8*)&dOTl[SWZ8(dc * 16 + lr, kc2 * 32 + lg * 8)];
+                dvr[dc] = MFMA16(ptf, dotb, dvr[dc]);
+                bf16x8 qtb = *(const bf16x8*)&QTl[SWZ8(dc * 16 + lr, kc2 * 32 + lg * 8)];
+                dkr[dc] = MFMA16(dstf, qtb, dkr[dc]);
+            }
+        }
+
+        // --- dQ strip (this wave's 16 q rows): dS*K^T-image, atomics ---
 #pragma unroll
-                for (int kc2 = 0; kc2 < 2; ++kc2) {
-                    bf16x8 dsf = *(const bf16x8*)&dSl[(wave * 16 + lr) * ST + kc2 * 32 + lg * 8];
-                    bf16x8 ktb = *(const bf16x8*)&KTl[(dc * 16 + lr) * ST + kc2 * 32 + lg * 8];
-                    dq = MFMA16(dsf, ktb, dq);
-                }
+        for (int dc = 0; dc < DCH; ++dc) {
+            f32x4 dq = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
-                for (int r = 0; r < 4; ++r) {
-                    const int qpos = qs + wave * 16 + lg * 4 + r;
-                    const int d = dc * 16 + lr;
-                    if (qpos < L && d < D)
-                        atomicAdd(&dq_acc[((int64_t)(s0 + qpos) * H + h) * D + d], dq[r]);
-                }
+            for (int kc2 = 0; kc2 < 2; ++kc2) {
+                bf16x8 dsf = *(const bf16x8*)&dSl[(wave * 16 + lr) * ST + kc2 * 32 + lg * 8];
+                bf16x8 ktb = *(const bf16x8*)&KTl[SWZ8(dc * 16 + lr, kc2 * 32 + lg * 8)];
+                dq = MFMA16(dsf, ktb, dq);
+            }
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                const int qpos = qs + wave * 16 + lg * 4 + r;
+                const int d = dc * 16 + lr;
+                if (qpos < L && d < D)
+                    atomicAdd(&dq_acc[((int64_t)(s0 + qpos) * H + h) * D + d], dq[r]);
             }
         }
     }
@@ -492,6 +507,8 @@ __global__ void __launch_bounds__(256) fa_bwd_kernel(
             }
         }
     }
+#undef SWZ
+#undef SWZ8
 }
 
 template <int DPAD>
@@ -502,8 +519,8 @@ static int launch_fa_bwd(hipStream_t stream, const __bf16* q, const __bf16* k, c
                          int64_t q_ts, int64_t q_gs, int64_t k_ts, int64_t k_hs,
                          int64_t v_ts, int64_t v_hs, int64_t do_ts, int max_tiles, float scale) {
     dim3 grid(max_tiles, batch, H), block(256);
-    constexpr int SQ = DPAD + 8, ST = 64 + 8;
-    size_t shmem = (size_t)(64 * SQ * 2 + DPAD * ST * 3 + 64 * ST * 3) * sizeof(__bf16);
+    constexpr int ST = 64 + 8;
+    size_t shmem = (size_t)(DPAD * ST * 3 + 64 * ST * 3) * sizeof(__bf16);
     hipLaunchKernelGGL((fa_bwd_kernel<DPAD>), grid, block, shmem, stream,
                        q, k, v, dout, lse, delta, dq_acc, dk_acc, dv_acc, cu, H, Hkv, D, G,
                        q_ts, q_gs, k_ts, k_hs, v_ts, v_hs, do_ts, T, scale);
