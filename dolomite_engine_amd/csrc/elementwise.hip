@@ -19,273 +19,336 @@
 #include <math.h>
 
 // ===========================================================================
-// RMSNorm forward: one workgroup per row, input stashed in LDS as fp32.
+// RMSNorm / LayerNorm — wave-per-row, register-resident, no barriers.
+// HBM-bound: one coalesced read of the row into VGPRs, wave-shuffle
+// reductions, one coalesced write. (The first round's row-per-workgroup +
+// LDS-stash version measured 0.75 TB/s on (32k, 2560) bf16; this structure
+// removes the LDS round trip and both barriers.)
+// Rows per workgroup = 4 (one per wave); lane owns elements lane*V + i*64*V.
 // ===========================================================================
 
-template <typename T, int V>
+template <typename T, int V, int ITMAX, bool HAS_RES>
 __global__ void __launch_bounds__(256) rmsnorm_fwd_kernel(
     const T* __restrict__ x, const T* __restrict__ res_in, const T* __restrict__ w,
     T* __restrict__ y, T* __restrict__ res_out, float* __restrict__ rstd_out,
     int64_t T_rows, int H, float eps) {
-    extern __shared__ float smem[];           // H floats (row stash) + 8 reduce
-    float* row = smem;
-    float* red = smem + H;
+    int64_t row = (int64_t)blockIdx.x * 4 + (threadIdx.x >> 6);
+    if (row >= T_rows) return;
+    const int lane = threadIdx.x & 63;
+    const T* xr = x + row * (int64_t)H;
+    const T* rr = HAS_RES ? res_in + row * (int64_t)H : nullptr;
 
-    int64_t t = blockIdx.x;
-    const T* xr = x + t * (int64_t)H;
-    const T* rr = res_in ? res_in + t * (int64_t)H : nullptr;
-
+    float buf[ITMAX][V];
     float ss = 0.f;
-    for (int c = threadIdx.x * V; c < H; c += blockDim.x * V) {
 #pragma unroll
-        for (int k = 0; k < V; ++k) {
-            if (c + k < H) {
-                float v = load_as_f32(xr + c + k);
-                if (rr) {
-                    // the residual add is a tensor-dtype add in the reference
-                    // (layer.py:74,87): round the sum like a torch bf16 add
-                    v += load_as_f32(rr + c + k);
+    for (int i = 0; i < ITMAX; ++i) {
+        int c = (i * 64 + lane) * V;
+        if (c < H) {
+#pragma unroll
+            for (int kk = 0; kk < V; ++kk) {
+                float v = load_as_f32(xr + c + kk);
+                if (HAS_RES) {
+                    // tensor-dtype residual add (layer.py:74,87): round the sum
+                    v += load_as_f32(rr + c + kk);
                     T tmp;
                     store_from_f32(&tmp, v);
                     v = load_as_f32(&tmp);
                 }
-                row[c + k] = v;
+                buf[i][kk] = v;
                 ss += v * v;
             }
         }
     }
-    float total = block_reduce_sum(ss, red, blockDim.x);
-    float rstd = rsqrtf(total / (float)H + eps);
-    if (threadIdx.x == 0 && rstd_out) rstd_out[t] = rstd;
+    float rstd = rsqrtf(wave_reduce_sum(ss) / (float)H + eps);
+    if (lane == 0 && rstd_out) rstd_out[row] = rstd;
 
-    T* yr = y + t * (int64_t)H;
-    T* sr = res_out ? res_out + t * (int64_t)H : nullptr;
-    for (int c = threadIdx.x * V; c < H; c += blockDim.x * V) {
+    T* yr = y + row * (int64_t)H;
+    T* sr = HAS_RES ? res_out + row * (int64_t)H : nullptr;
 #pragma unroll
-        for (int k = 0; k < V; ++k) {
-            if (c + k < H) {
-                float s = row[c + k];
-                if (sr) store_from_f32(sr + c + k, s);
-                // reference order: cast normalized value to input dtype BEFORE
-                // multiplying by weight (rmsnorm/base.py:23-25)
-                float nhat;
+    for (int i = 0; i < ITMAX; ++i) {
+        int c = (i * 64 + lane) * V;
+        if (c < H) {
+#pragma unroll
+            for (int kk = 0; kk < V; ++kk) {
+                float v = buf[i][kk];
+                if (HAS_RES) store_from_f32(sr + c + kk, v);
                 T tmp;
-                store_from_f32(&tmp, s * rstd);
-                nhat = load_as_f32(&tmp);
-                store_from_f32(yr + c + k, load_as_f32(w + c + k) * nhat);
+                store_from_f32(&tmp, v * rstd);  // cast before weight (rmsnorm/base.py:23-25)
+                store_from_f32(yr + c + kk, load_as_f32(w + c + kk) * load_as_f32(&tmp));
             }
         }
     }
 }
+
+// Dispatch helper: pick the smallest ITMAX covering H at vector width V.
+// Requires H % V == 0 (true for every transformer width on this path);
+// falls back to V=1 otherwise.
+#define DOL_NORM_DISPATCH(KERN, T_, V_, HAS_RES_, ...)                                   \
+    do {                                                                                 \
+        int it = ((int)H + 64 * (V_)-1) / (64 * (V_));                                   \
+        if (it <= 2)                                                                     \
+            hipLaunchKernelGGL((KERN<T_, V_, 2, HAS_RES_>), grid, block, 0, stream_, __VA_ARGS__); \
+        else if (it <= 4)                                                                \
+            hipLaunchKernelGGL((KERN<T_, V_, 4, HAS_RES_>), grid, block, 0, stream_, __VA_ARGS__); \
+        else if (it <= 5)                                                                \
+            hipLaunchKernelGGL((KERN<T_, V_, 5, HAS_RES_>), grid, block, 0, stream_, __VA_ARGS__); \
+        else if (it <= 8)                                                                \
+            hipLaunchKernelGGL((KERN<T_, V_, 8, HAS_RES_>), grid, block, 0, stream_, __VA_ARGS__); \
+        else if (it <= 16)                                                               \
+            hipLaunchKernelGGL((KERN<T_, V_, 16, HAS_RES_>), grid, block, 0, stream_, __VA_ARGS__); \
+        else                                                                             \
+            return 9002;                                                                 \
+    } while (0)
 
 extern "C" int dolomite_rmsnorm_fwd(dolomite_stream_t stream,
                                     const void* x, const void* res_in, const void* w,
                                     void* y, void* res_out, float* rstd,
                                     int64_t T_rows, int64_t H, float eps, int dtype) {
-    if (H > 65536 / 8) { /* LDS cap: H*4 + 32 bytes must fit 160KB; enforce 16K */ }
-    dim3 grid((uint32_t)T_rows), block(256);
-    size_t shmem = (size_t)H * 4 + 8 * 4;
-    if (shmem > 160 * 1024) return 9001;
-    if (dtype == DOLOMITE_BF16) {
-        if (H % 8 == 0)
-            hipLaunchKernelGGL((rmsnorm_fwd_kernel<uint16_t, 8>), grid, block, shmem, (hipStream_t)stream,
-                               (const uint16_t*)x, (const uint16_t*)res_in, (const uint16_t*)w,
-                               (uint16_t*)y, (uint16_t*)res_out, rstd, T_rows, (int)H, eps);
+    if (T_rows == 0) return 0;
+    hipStream_t stream_ = (hipStream_t)stream;
+    dim3 grid((uint32_t)((T_rows + 3) / 4)), block(256);
+    bool has_res = res_in != nullptr;
+    if (dtype == DOLOMITE_BF16 && H % 8 == 0) {
+        if (has_res)
+            DOL_NORM_DISPATCH(rmsnorm_fwd_kernel, uint16_t, 8, true,
+                              (const uint16_t*)x, (const uint16_t*)res_in, (const uint16_t*)w,
+                              (uint16_t*)y, (uint16_t*)res_out, rstd, T_rows, (int)H, eps);
         else
-            hipLaunchKernelGGL((rmsnorm_fwd_kernel<uint16_t, 1>), grid, block, shmem, (hipStream_t)stream,
-                               (const uint16_t*)x, (const uint16_t*)res_in, (const uint16_t*)w,
-                               (uint16_t*)y, (uint16_t*)res_out, rstd, T_rows, (int)H, eps);
+            DOL_NORM_DISPATCH(rmsnorm_fwd_kernel, uint16_t, 8, false,
+                              (const uint16_t*)x, (const uint16_t*)res_in, (const uint16_t*)w,
+                              (uint16_t*)y, (uint16_t*)res_out, rstd, T_rows, (int)H, eps);
+    } else if (dtype == DOLOMITE_BF16) {
+        if (has_res)
+            DOL_NORM_DISPATCH(rmsnorm_fwd_kernel, uint16_t, 1, true,
+                              (const uint16_t*)x, (const uint16_t*)res_in, (const uint16_t*)w,
+                              (uint16_t*)y, (uint16_t*)res_out, rstd, T_rows, (int)H, eps);
+        else
+            DOL_NORM_DISPATCH(rmsnorm_fwd_kernel, uint16_t, 1, false,
+                              (const uint16_t*)x, (const uint16_t*)res_in, (const uint16_t*)w,
+                              (uint16_t*)y, (uint16_t*)res_out, rstd, T_rows, (int)H, eps);
+    } else if (H % 4 == 0) {
+        if (has_res)
+            DOL_NORM_DISPATCH(rmsnorm_fwd_kernel, float, 4, true,
+                              (const float*)x, (const float*)res_in, (const float*)w,
+                              (float*)y, (float*)res_out, rstd, T_rows, (int)H, eps);
+        else
+            DOL_NORM_DISPATCH(rmsnorm_fwd_kernel, float, 4, false,
+                              (const float*)x, (const float*)res_in, (const float*)w,
+                              (float*)y, (float*)res_out, rstd, T_rows, (int)H, eps);
     } else {
-        if (H % 4 == 0)
-            hipLaunchKernelGGL((rmsnorm_fwd_kernel<float, 4>), grid, block, shmem, (hipStream_t)stream,
-                               (const float*)x, (const float*)res_in, (const float*)w,
-                               (float*)y, (float*)res_out, rstd, T_rows, (int)H, eps);
+        if (has_res)
+            DOL_NORM_DISPATCH(rmsnorm_fwd_kernel, float, 1, true,
+                              (const float*)x, (const float*)res_in, (const float*)w,
+                              (float*)y, (float*)res_out, rstd, T_rows, (int)H, eps);
         else
-            hipLaunchKernelGGL((rmsnorm_fwd_kernel<float, 1>), grid, block, shmem, (hipStream_t)stream,
-                               (const float*)x, (const float*)res_in, (const float*)w,
-                               (float*)y, (float*)res_out, rstd, T_rows, (int)H, eps);
+            DOL_NORM_DISPATCH(rmsnorm_fwd_kernel, float, 1, false,
+                              (const float*)x, (const float*)res_in, (const float*)w,
+                              (float*)y, (float*)res_out, rstd, T_rows, (int)H, eps);
     }
     return dol_last_error();
 }
 
-// ===========================================================================
-// RMSNorm backward: fixed grid, per-thread column ownership for dw partials.
+// ---------------------------------------------------------------------------
+// RMSNorm backward — wave-per-row, per-wave dw partials (fixed grid).
 //   dx = rstd*(w*dy - s_hat * mean(w*dy*s_hat)),  s_hat = s_fp32*rstd
-//   dw = sum_rows dy * cast_to_dtype(s_hat)
-// ===========================================================================
+//   dw = sum_rows dy * cast_to_dtype(s_hat)          (fp32 partials)
+// ---------------------------------------------------------------------------
 
-#define RMS_BWD_BLOCKS 600  // > 256 CUs, fixed so dw_partial scratch is bounded
+#define RMS_BWD_BLOCKS 1024  // x4 waves = 4096 dw partial rows
 
 extern "C" int dolomite_rmsnorm_bwd_nblocks(int64_t T_rows) {
     (void)T_rows;
-    return RMS_BWD_BLOCKS;
+    return RMS_BWD_BLOCKS * 4;
 }
 
-template <typename T, int V, int ITMAX>
-__global__ void __launch_bounds__(256) rmsnorm_bwd_kernel(
+template <typename T, int V, int ITMAX, bool IS_LN>
+__global__ void __launch_bounds__(256) norm_bwd_kernel(
     const T* __restrict__ dy, const T* __restrict__ s, const T* __restrict__ w,
-    const float* __restrict__ rstd, T* __restrict__ dx, float* __restrict__ dw_partial,
+    const float* __restrict__ mean, const float* __restrict__ rstd,
+    T* __restrict__ dx, float* __restrict__ dw_partial,
     int64_t T_rows, int H) {
-    __shared__ float red[8];
+    const int lane = threadIdx.x & 63;
+    const int64_t wslot = (int64_t)blockIdx.x * 4 + (threadIdx.x >> 6);
+    const int64_t nslots = (int64_t)gridDim.x * 4;
 
-    float wv[ITMAX][V];
-    float dw_acc[ITMAX][V];
+    float wv[ITMAX][V], dw_acc[ITMAX][V], db_acc[IS_LN ? ITMAX : 1][IS_LN ? V : 1];
 #pragma unroll
     for (int i = 0; i < ITMAX; ++i)
 #pragma unroll
-        for (int k = 0; k < V; ++k) {
-            int c = i * (int)blockDim.x * V + (int)threadIdx.x * V + k;
-            wv[i][k] = (c < H) ? load_as_f32(w + c) : 0.f;
-            dw_acc[i][k] = 0.f;
+        for (int kk = 0; kk < V; ++kk) {
+            int c = (i * 64 + lane) * V + kk;
+            wv[i][kk] = (c < H) ? load_as_f32(w + c) : 0.f;
+            dw_acc[i][kk] = 0.f;
+            if (IS_LN) db_acc[i][kk] = 0.f;
         }
 
-    for (int64_t t = blockIdx.x; t < T_rows; t += gridDim.x) {
-        const T* dyr = dy + t * (int64_t)H;
-        const T* sr = s + t * (int64_t)H;
-        float r = rstd[t];
+    for (int64_t row = wslot; row < T_rows; row += nslots) {
+        const T* dyr = dy + row * (int64_t)H;
+        const T* sr = s + row * (int64_t)H;
+        float r = rstd[row];
+        float mu = IS_LN ? mean[row] : 0.f;
 
-        float sv[ITMAX][V], dyv[ITMAX][V];
-        float dot = 0.f;
+        float sh[ITMAX][V], dyv[ITMAX][V];
+        float d1 = 0.f, d2 = 0.f;
 #pragma unroll
-        for (int i = 0; i < ITMAX; ++i)
+        for (int i = 0; i < ITMAX; ++i) {
+            int c0 = (i * 64 + lane) * V;
+            if (c0 < H) {
 #pragma unroll
-            for (int k = 0; k < V; ++k) {
-                int c = i * (int)blockDim.x * V + (int)threadIdx.x * V + k;
-                if (c < H) {
-                    float svv = load_as_f32(sr + c);
-                    float dyy = load_as_f32(dyr + c);
-                    sv[i][k] = svv;
-                    dyv[i][k] = dyy;
-                    dot += wv[i][k] * dyy * (svv * r);
-                } else {
-                    sv[i][k] = 0.f;
-                    dyv[i][k] = 0.f;
+                for (int kk = 0; kk < V; ++kk) {
+                    float shat = (load_as_f32(sr + c0 + kk) - mu) * r;
+                    float dyy = load_as_f32(dyr + c0 + kk);
+                    float dyw = dyy * wv[i][kk];
+                    sh[i][kk] = shat;
+                    dyv[i][kk] = dyy;
+                    d2 += dyw * shat;
+                    if (IS_LN) d1 += dyw;
+                }
+            } else {
+#pragma unroll
+                for (int kk = 0; kk < V; ++kk) { sh[i][kk] = 0.f; dyv[i][kk] = 0.f; }
+            }
+        }
+        float m2 = wave_reduce_sum(d2) / (float)H;
+        float m1 = IS_LN ? wave_reduce_sum(d1) / (float)H : 0.f;
+
+        T* dxr = dx + row * (int64_t)H;
+#pragma unroll
+        for (int i = 0; i < ITMAX; ++i) {
+            int c0 = (i * 64 + lane) * V;
+            if (c0 < H) {
+#pragma unroll
+                for (int kk = 0; kk < V; ++kk) {
+                    float dxv;
+                    if (IS_LN) {
+                        dxv = r * (dyv[i][kk] * wv[i][kk] - m1 - sh[i][kk] * m2);
+                        dw_acc[i][kk] += dyv[i][kk] * sh[i][kk];
+                        db_acc[i][kk] += dyv[i][kk];
+                    } else {
+                        dxv = r * (wv[i][kk] * dyv[i][kk] - sh[i][kk] * m2);
+                        // dw uses the CAST normalized value (reference casts
+                        // before the weight multiply)
+                        T tmp;
+                        store_from_f32(&tmp, sh[i][kk]);
+                        dw_acc[i][kk] += dyv[i][kk] * load_as_f32(&tmp);
+                    }
+                    store_from_f32(dxr + c0 + kk, dxv);
                 }
             }
-        float dtot = block_reduce_sum(dot, red, blockDim.x) / (float)H;
-
-        T* dxr = dx + t * (int64_t)H;
-#pragma unroll
-        for (int i = 0; i < ITMAX; ++i)
-#pragma unroll
-            for (int k = 0; k < V; ++k) {
-                int c = i * (int)blockDim.x * V + (int)threadIdx.x * V + k;
-                if (c < H) {
-                    float shat = sv[i][k] * r;
-                    float dxv = r * (wv[i][k] * dyv[i][k] - shat * dtot);
-                    store_from_f32(dxr + c, dxv);
-                    // dw uses the CAST normalized value (reference casts before w-mul)
-                    T tmp;
-                    store_from_f32(&tmp, shat);
-                    dw_acc[i][k] += dyv[i][k] * load_as_f32(&tmp);
-                }
-            }
+        }
     }
 
-    float* dwp = dw_partial + (int64_t)blockIdx.x * H;
+    float* dwp = dw_partial + wslot * H;
+    float* dbp = IS_LN ? dw_partial + (nslots + wslot) * H : nullptr;
 #pragma unroll
-    for (int i = 0; i < ITMAX; ++i)
+    for (int i = 0; i < ITMAX; ++i) {
+        int c0 = (i * 64 + lane) * V;
+        if (c0 < H) {
 #pragma unroll
-        for (int k = 0; k < V; ++k) {
-            int c = i * (int)blockDim.x * V + (int)threadIdx.x * V + k;
-            if (c < H) dwp[c] = dw_acc[i][k];
+            for (int kk = 0; kk < V; ++kk) {
+                dwp[c0 + kk] = dw_acc[i][kk];
+                if (IS_LN) dbp[c0 + kk] = db_acc[i][kk];
+            }
         }
-}
-
-template <typename T>
-static int launch_rmsnorm_bwd(hipStream_t stream, const T* dy, const T* s, const T* w,
-                              const float* rstd, T* dx, float* dw_partial, int64_t T_rows, int H) {
-    dim3 grid(RMS_BWD_BLOCKS), block(256);
-    const int V = 4;
-    int it = (H + 256 * V - 1) / (256 * V);
-    if (it <= 1)
-        hipLaunchKernelGGL((rmsnorm_bwd_kernel<T, V, 1>), grid, block, 0, stream, dy, s, w, rstd, dx, dw_partial, T_rows, H);
-    else if (it <= 2)
-        hipLaunchKernelGGL((rmsnorm_bwd_kernel<T, V, 2>), grid, block, 0, stream, dy, s, w, rstd, dx, dw_partial, T_rows, H);
-    else if (it <= 4)
-        hipLaunchKernelGGL((rmsnorm_bwd_kernel<T, V, 4>), grid, block, 0, stream, dy, s, w, rstd, dx, dw_partial, T_rows, H);
-    else if (it <= 8)
-        hipLaunchKernelGGL((rmsnorm_bwd_kernel<T, V, 8>), grid, block, 0, stream, dy, s, w, rstd, dx, dw_partial, T_rows, H);
-    else
-        return 9002;  // H > 8192 unsupported by this kernel
-    return dol_last_error();
+    }
 }
 
 extern "C" int dolomite_rmsnorm_bwd(dolomite_stream_t stream,
                                     const void* dy, const void* s, const void* w,
                                     const float* rstd, void* dx, float* dw_partial,
                                     int64_t T_rows, int64_t H, int dtype) {
-    if (dtype == DOLOMITE_BF16)
-        return launch_rmsnorm_bwd((hipStream_t)stream, (const uint16_t*)dy, (const uint16_t*)s,
-                                  (const uint16_t*)w, rstd, (uint16_t*)dx, dw_partial, T_rows, (int)H);
-    return launch_rmsnorm_bwd((hipStream_t)stream, (const float*)dy, (const float*)s,
-                              (const float*)w, rstd, (float*)dx, dw_partial, T_rows, (int)H);
+    if (T_rows == 0) return 0;
+    hipStream_t stream_ = (hipStream_t)stream;
+    dim3 grid(RMS_BWD_BLOCKS), block(256);
+    if (dtype == DOLOMITE_BF16 && H % 8 == 0)
+        DOL_NORM_DISPATCH(norm_bwd_kernel, uint16_t, 8, false,
+                          (const uint16_t*)dy, (const uint16_t*)s, (const uint16_t*)w,
+                          nullptr, rstd, (uint16_t*)dx, dw_partial, T_rows, (int)H);
+    else if (dtype == DOLOMITE_BF16)
+        DOL_NORM_DISPATCH(norm_bwd_kernel, uint16_t, 1, false,
+                          (const uint16_t*)dy, (const uint16_t*)s, (const uint16_t*)w,
+                          nullptr, rstd, (uint16_t*)dx, dw_partial, T_rows, (int)H);
+    else if (H % 4 == 0)
+        DOL_NORM_DISPATCH(norm_bwd_kernel, float, 4, false,
+                          (const float*)dy, (const float*)s, (const float*)w,
+                          nullptr, rstd, (float*)dx, dw_partial, T_rows, (int)H);
+    else
+        DOL_NORM_DISPATCH(norm_bwd_kernel, float, 1, false,
+                          (const float*)dy, (const float*)s, (const float*)w,
+                          nullptr, rstd, (float*)dx, dw_partial, T_rows, (int)H);
+    return dol_last_error();
 }
 
-// ===========================================================================
-// LayerNorm forward / backward (same structure, plus mean and bias).
-//   y = cast((x32 - mu) * rstd * w32 + b32)   [torch F.layer_norm semantics:
-//   bf16 upcast to fp32 throughout, single final cast]
-// ===========================================================================
+// ---------------------------------------------------------------------------
+// LayerNorm forward (wave-per-row) and backward (shares norm_bwd_kernel).
+//   y = cast((x32 - mu)*rstd*w32 + b32)    [torch F.layer_norm semantics]
+// ---------------------------------------------------------------------------
 
-template <typename T, int V>
+template <typename T, int V, int ITMAX, bool HAS_RES>
 __global__ void __launch_bounds__(256) layernorm_fwd_kernel(
     const T* __restrict__ x, const T* __restrict__ res_in,
     const T* __restrict__ w, const T* __restrict__ b,
     T* __restrict__ y, T* __restrict__ res_out,
     float* __restrict__ mean_out, float* __restrict__ rstd_out,
     int64_t T_rows, int H, float eps) {
-    extern __shared__ float smem[];
-    float* row = smem;
-    float* red = smem + H;
+    int64_t row = (int64_t)blockIdx.x * 4 + (threadIdx.x >> 6);
+    if (row >= T_rows) return;
+    const int lane = threadIdx.x & 63;
+    const T* xr = x + row * (int64_t)H;
+    const T* rr = HAS_RES ? res_in + row * (int64_t)H : nullptr;
 
-    int64_t t = blockIdx.x;
-    const T* xr = x + t * (int64_t)H;
-    const T* rr = res_in ? res_in + t * (int64_t)H : nullptr;
-
+    float buf[ITMAX][V];
     float sum = 0.f;
-    for (int c = threadIdx.x * V; c < H; c += blockDim.x * V) {
 #pragma unroll
-        for (int k = 0; k < V; ++k)
-            if (c + k < H) {
-                float v = load_as_f32(xr + c + k);
-                if (rr) {
-                    v += load_as_f32(rr + c + k);
+    for (int i = 0; i < ITMAX; ++i) {
+        int c = (i * 64 + lane) * V;
+        if (c < H) {
+#pragma unroll
+            for (int kk = 0; kk < V; ++kk) {
+                float v = load_as_f32(xr + c + kk);
+                if (HAS_RES) {
+                    v += load_as_f32(rr + c + kk);
                     T tmp;
                     store_from_f32(&tmp, v);
                     v = load_as_f32(&tmp);
                 }
-                row[c + k] = v;
+                buf[i][kk] = v;
                 sum += v;
             }
+        }
     }
-    float mu = block_reduce_sum(sum, red, blockDim.x) / (float)H;
+    float mu = wave_reduce_sum(sum) / (float)H;
     float ss = 0.f;
-    for (int c = threadIdx.x * V; c < H; c += blockDim.x * V) {
 #pragma unroll
-        for (int k = 0; k < V; ++k)
-            if (c + k < H) {
-                float d = row[c + k] - mu;
+    for (int i = 0; i < ITMAX; ++i) {
+        int c = (i * 64 + lane) * V;
+        if (c < H)
+#pragma unroll
+            for (int kk = 0; kk < V; ++kk) {
+                float d = buf[i][kk] - mu;
                 ss += d * d;
             }
     }
-    float rstd = rsqrtf(block_reduce_sum(ss, red, blockDim.x) / (float)H + eps);
-    if (threadIdx.x == 0) {
-        if (mean_out) mean_out[t] = mu;
-        if (rstd_out) rstd_out[t] = rstd;
+    float rstd = rsqrtf(wave_reduce_sum(ss) / (float)H + eps);
+    if (lane == 0) {
+        if (mean_out) mean_out[row] = mu;
+        if (rstd_out) rstd_out[row] = rstd;
     }
 
-    T* yr = y + t * (int64_t)H;
-    T* sr = res_out ? res_out + t * (int64_t)H : nullptr;
-    for (int c = threadIdx.x * V; c < H; c += blockDim.x * V) {
+    T* yr = y + row * (int64_t)H;
+    T* sr = HAS_RES ? res_out + row * (int64_t)H : nullptr;
 #pragma unroll
-        for (int k = 0; k < V; ++k)
-            if (c + k < H) {
-                float s = row[c + k];
-                if (sr) store_from_f32(sr + c + k, s);
-                float v = (s - mu) * rstd * load_as_f32(w + c + k) + load_as_f32(b + c + k);
-                store_from_f32(yr + c + k, v);
+    for (int i = 0; i < ITMAX; ++i) {
+        int c = (i * 64 + lane) * V;
+        if (c < H) {
+#pragma unroll
+            for (int kk = 0; kk < V; ++kk) {
+                float v = buf[i][kk];
+                if (HAS_RES) store_from_f32(sr + c + kk, v);
+                float out = (v - mu) * rstd * load_as_f32(w + c + kk) + load_as_f32(b + c + kk);
+                store_from_f32(yr + c + kk, out);
             }
+        }
     }
 }
 
@@ -293,125 +356,47 @@ extern "C" int dolomite_layernorm_fwd(dolomite_stream_t stream,
                                       const void* x, const void* res_in, const void* w, const void* b,
                                       void* y, void* res_out, float* mean, float* rstd,
                                       int64_t T_rows, int64_t H, float eps, int dtype) {
-    dim3 grid((uint32_t)T_rows), block(256);
-    size_t shmem = (size_t)H * 4 + 8 * 4;
-    if (shmem > 160 * 1024) return 9001;
-    if (dtype == DOLOMITE_BF16) {
-        if (H % 8 == 0)
-            hipLaunchKernelGGL((layernorm_fwd_kernel<uint16_t, 8>), grid, block, shmem, (hipStream_t)stream,
-                               (const uint16_t*)x, (const uint16_t*)res_in, (const uint16_t*)w, (const uint16_t*)b,
-                               (uint16_t*)y, (uint16_t*)res_out, mean, rstd, T_rows, (int)H, eps);
+    if (T_rows == 0) return 0;
+    hipStream_t stream_ = (hipStream_t)stream;
+    dim3 grid((uint32_t)((T_rows + 3) / 4)), block(256);
+    bool has_res = res_in != nullptr;
+    if (dtype == DOLOMITE_BF16 && H % 8 == 0) {
+        if (has_res)
+            DOL_NORM_DISPATCH(layernorm_fwd_kernel, uint16_t, 8, true,
+                              (const uint16_t*)x, (const uint16_t*)res_in, (const uint16_t*)w, (const uint16_t*)b,
+                              (uint16_t*)y, (uint16_t*)res_out, mean, rstd, T_rows, (int)H, eps);
         else
-            hipLaunchKernelGGL((layernorm_fwd_kernel<uint16_t, 1>), grid, block, shmem, (hipStream_t)stream,
-                               (const uint16_t*)x, (const uint16_t*)res_in, (const uint16_t*)w, (const uint16_t*)b,
-                               (uint16_t*)y, (uint16_t*)res_out, mean, rstd, T_rows, (int)H, eps);
+            DOL_NORM_DISPATCH(layernorm_fwd_kernel, uint16_t, 8, false,
+                              (const uint16_t*)x, (const uint16_t*)res_in, (const uint16_t*)w, (const uint16_t*)b,
+                              (uint16_t*)y, (uint16_t*)res_out, mean, rstd, T_rows, (int)H, eps);
+    } else if (dtype == DOLOMITE_BF16) {
+        if (has_res)
+            DOL_NORM_DISPATCH(layernorm_fwd_kernel, uint16_t, 1, true,
+                              (const uint16_t*)x, (const uint16_t*)res_in, (const uint16_t*)w, (const uint16_t*)b,
+                              (uint16_t*)y, (uint16_t*)res_out, mean, rstd, T_rows, (int)H, eps);
+        else
+            DOL_NORM_DISPATCH(layernorm_fwd_kernel, uint16_t, 1, false,
+                              (const uint16_t*)x, (const uint16_t*)res_in, (const uint16_t*)w, (const uint16_t*)b,
+                              (uint16_t*)y, (uint16_t*)res_out, mean, rstd, T_rows, (int)H, eps);
+    } else if (H % 4 == 0) {
+        if (has_res)
+            DOL_NORM_DISPATCH(layernorm_fwd_kernel, float, 4, true,
+                              (const float*)x, (const float*)res_in, (const float*)w, (const float*)b,
+                              (float*)y, (float*)res_out, mean, rstd, T_rows, (int)H, eps);
+        else
+            DOL_NORM_DISPATCH(layernorm_fwd_kernel, float, 4, false,
+                              (const float*)x, (const float*)res_in, (const float*)w, (const float*)b,
+                              (float*)y, (float*)res_out, mean, rstd, T_rows, (int)H, eps);
     } else {
-        if (H % 4 == 0)
-            hipLaunchKernelGGL((layernorm_fwd_kernel<float, 4>), grid, block, shmem, (hipStream_t)stream,
-                               (const float*)x, (const float*)res_in, (const float*)w, (const float*)b,
-                               (float*)y, (float*)res_out, mean, rstd, T_rows, (int)H, eps);
+        if (has_res)
+            DOL_NORM_DISPATCH(layernorm_fwd_kernel, float, 1, true,
+                              (const float*)x, (const float*)res_in, (const float*)w, (const float*)b,
+                              (float*)y, (float*)res_out, mean, rstd, T_rows, (int)H, eps);
         else
-            hipLaunchKernelGGL((layernorm_fwd_kernel<float, 1>), grid, block, shmem, (hipStream_t)stream,
-                               (const float*)x, (const float*)res_in, (const float*)w, (const float*)b,
-                               (float*)y, (float*)res_out, mean, rstd, T_rows, (int)H, eps);
+            DOL_NORM_DISPATCH(layernorm_fwd_kernel, float, 1, false,
+                              (const float*)x, (const float*)res_in, (const float*)w, (const float*)b,
+                              (float*)y, (float*)res_out, mean, rstd, T_rows, (int)H, eps);
     }
-    return dol_last_error();
-}
-
-template <typename T, int V, int ITMAX>
-__global__ void __launch_bounds__(256) layernorm_bwd_kernel(
-    const T* __restrict__ dy, const T* __restrict__ s, const T* __restrict__ w,
-    const float* __restrict__ mean, const float* __restrict__ rstd,
-    T* __restrict__ dx, float* __restrict__ dwdb_partial,
-    int64_t T_rows, int H) {
-    __shared__ float red[8];
-
-    float wv[ITMAX][V], dw_acc[ITMAX][V], db_acc[ITMAX][V];
-#pragma unroll
-    for (int i = 0; i < ITMAX; ++i)
-#pragma unroll
-        for (int k = 0; k < V; ++k) {
-            int c = i * (int)blockDim.x * V + (int)threadIdx.x * V + k;
-            wv[i][k] = (c < H) ? load_as_f32(w + c) : 0.f;
-            dw_acc[i][k] = 0.f;
-            db_acc[i][k] = 0.f;
-        }
-
-    for (int64_t t = blockIdx.x; t < T_rows; t += gridDim.x) {
-        const T* dyr = dy + t * (int64_t)H;
-        const T* sr = s + t * (int64_t)H;
-        float mu = mean[t], r = rstd[t];
-
-        float xh[ITMAX][V], dyv[ITMAX][V];
-        float d1 = 0.f, d2 = 0.f;
-#pragma unroll
-        for (int i = 0; i < ITMAX; ++i)
-#pragma unroll
-            for (int k = 0; k < V; ++k) {
-                int c = i * (int)blockDim.x * V + (int)threadIdx.x * V + k;
-                if (c < H) {
-                    float xhat = (load_as_f32(sr + c) - mu) * r;
-                    float dyy = load_as_f32(dyr + c);
-                    float dyw = dyy * wv[i][k];
-                    xh[i][k] = xhat;
-                    dyv[i][k] = dyy;
-                    d1 += dyw;
-                    d2 += dyw * xhat;
-                } else {
-                    xh[i][k] = 0.f;
-                    dyv[i][k] = 0.f;
-                }
-            }
-        float m1 = block_reduce_sum(d1, red, blockDim.x) / (float)H;
-        __syncthreads();
-        float m2 = block_reduce_sum(d2, red, blockDim.x) / (float)H;
-
-        T* dxr = dx + t * (int64_t)H;
-#pragma unroll
-        for (int i = 0; i < ITMAX; ++i)
-#pragma unroll
-            for (int k = 0; k < V; ++k) {
-                int c = i * (int)blockDim.x * V + (int)threadIdx.x * V + k;
-                if (c < H) {
-                    float dxv = r * (dyv[i][k] * wv[i][k] - m1 - xh[i][k] * m2);
-                    store_from_f32(dxr + c, dxv);
-                    dw_acc[i][k] += dyv[i][k] * xh[i][k];
-                    db_acc[i][k] += dyv[i][k];
-                }
-            }
-    }
-
-    float* dwp = dwdb_partial + (int64_t)blockIdx.x * H;
-    float* dbp = dwdb_partial + ((int64_t)gridDim.x + blockIdx.x) * H;
-#pragma unroll
-    for (int i = 0; i < ITMAX; ++i)
-#pragma unroll
-        for (int k = 0; k < V; ++k) {
-            int c = i * (int)blockDim.x * V + (int)threadIdx.x * V + k;
-            if (c < H) {
-                dwp[c] = dw_acc[i][k];
-                dbp[c] = db_acc[i][k];
-            }
-        }
-}
-
-template <typename T>
-static int launch_layernorm_bwd(hipStream_t stream, const T* dy, const T* s, const T* w,
-                                const float* mean, const float* rstd, T* dx, float* dwdb_partial,
-                                int64_t T_rows, int H) {
-    dim3 grid(RMS_BWD_BLOCKS), block(256);
-    const int V = 4;
-    int it = (H + 256 * V - 1) / (256 * V);
-    if (it <= 1)
-        hipLaunchKernelGGL((layernorm_bwd_kernel<T, V, 1>), grid, block, 0, stream, dy, s, w, mean, rstd, dx, dwdb_partial, T_rows, H);
-    else if (it <= 2)
-        hipLaunchKernelGGL((layernorm_bwd_kernel<T, V, 2>), grid, block, 0, stream, dy, s, w, mean, rstd, dx, dwdb_partial, T_rows, H);
-    else if (it <= 4)
-        hipLaunchKernelGGL((layernorm_bwd_kernel<T, V, 4>), grid, block, 0, stream, dy, s, w, mean, rstd, dx, dwdb_partial, T_rows, H);
-    else if (it <= 8)
-        hipLaunchKernelGGL((layernorm_bwd_kernel<T, V, 8>), grid, block, 0, stream, dy, s, w, mean, rstd, dx, dwdb_partial, T_rows, H);
-    else
-        return 9002;
     return dol_last_error();
 }
 
@@ -420,30 +405,57 @@ extern "C" int dolomite_layernorm_bwd(dolomite_stream_t stream,
                                       const float* mean, const float* rstd,
                                       void* dx, float* dwdb_partial,
                                       int64_t T_rows, int64_t H, int dtype) {
-    if (dtype == DOLOMITE_BF16)
-        return launch_layernorm_bwd((hipStream_t)stream, (const uint16_t*)dy, (const uint16_t*)s,
-                                    (const uint16_t*)w, mean, rstd, (uint16_t*)dx, dwdb_partial, T_rows, (int)H);
-    return launch_layernorm_bwd((hipStream_t)stream, (const float*)dy, (const float*)s,
-                                (const float*)w, mean, rstd, (float*)dx, dwdb_partial, T_rows, (int)H);
+    if (T_rows == 0) return 0;
+    hipStream_t stream_ = (hipStream_t)stream;
+    dim3 grid(RMS_BWD_BLOCKS), block(256);
+    if (dtype == DOLOMITE_BF16 && H % 8 == 0)
+        DOL_NORM_DISPATCH(norm_bwd_kernel, uint16_t, 8, true,
+                          (const uint16_t*)dy, (const uint16_t*)s, (const uint16_t*)w,
+                          mean, rstd, (uint16_t*)dx, dwdb_partial, T_rows, (int)H);
+    else if (dtype == DOLOMITE_BF16)
+        DOL_NORM_DISPATCH(norm_bwd_kernel, uint16_t, 1, true,
+                          (const uint16_t*)dy, (const uint16_t*)s, (const uint16_t*)w,
+                          mean, rstd, (uint16_t*)dx, dwdb_partial, T_rows, (int)H);
+    else if (H % 4 == 0)
+        DOL_NORM_DISPATCH(norm_bwd_kernel, float, 4, true,
+                          (const float*)dy, (const float*)s, (const float*)w,
+                          mean, rstd, (float*)dx, dwdb_partial, T_rows, (int)H);
+    else
+        DOL_NORM_DISPATCH(norm_bwd_kernel, float, 1, true,
+                          (const float*)dy, (const float*)s, (const float*)w,
+                          mean, rstd, (float*)dx, dwdb_partial, T_rows, (int)H);
+    return dol_last_error();
 }
 
 // ===========================================================================
-// Partials reduction: out[h] = sum_i partial[i*H + h]
+// Partials reduction: out[h] = sum_i partial[i*H + h].
+// Parallel over column blocks AND row chunks (atomic join); `out` must be
+// zero-initialized by the caller. (The serial column-loop version filled
+// only H/256 workgroups: 40 GB/s on (4096, 2560).)
 // ===========================================================================
 
 __global__ void __launch_bounds__(256) reduce_partials_kernel(
     const float* __restrict__ partial, float* __restrict__ out, int64_t nblocks, int64_t H) {
     int64_t h = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
     if (h >= H) return;
+    int64_t chunk = (nblocks + gridDim.y - 1) / gridDim.y;
+    int64_t i0 = (int64_t)blockIdx.y * chunk;
+    int64_t i1 = min(i0 + chunk, nblocks);
     float acc = 0.f;
-    for (int64_t i = 0; i < nblocks; ++i) acc += partial[i * H + h];
-    out[h] = acc;
+    for (int64_t i = i0; i < i1; ++i) acc += partial[i * H + h];
+    if (gridDim.y == 1)
+        out[h] = acc;
+    else
+        atomicAdd(&out[h], acc);
 }
 
 extern "C" int dolomite_reduce_partials(dolomite_stream_t stream,
                                         const float* partial, float* out,
                                         int64_t nblocks, int64_t H) {
-    dim3 grid((uint32_t)((H + 255) / 256)), block(256);
+    int64_t nsplit64 = (nblocks + 127) / 128;
+    uint32_t nsplit = (uint32_t)(nsplit64 > 32 ? 32 : nsplit64);
+    if (nsplit < 1) nsplit = 1;
+    dim3 grid((uint32_t)((H + 255) / 256), nsplit), block(256);
     hipLaunchKernelGGL(reduce_partials_kernel, grid, block, 0, (hipStream_t)stream, partial, out, nblocks, H);
     return dol_last_error();
 }

@@ -147,7 +147,7 @@ class FusedRMSNorm(torch.autograd.Function):
                 ),
                 "rmsnorm_bwd",
             )
-            dw32 = torch.empty(H, dtype=torch.float32, device=s.device)
+            dw32 = torch.zeros(H, dtype=torch.float32, device=s.device)
             hip.check(
                 hip.lib().dolomite_reduce_partials(hip.stream(), hip.ptr(dw_partial), hip.ptr(dw32), nb, H),
                 "reduce_partials",
@@ -235,7 +235,7 @@ class FusedLayerNorm(torch.autograd.Function):
             )
             # dw partials are rows [0, nb), db rows [nb, 2nb) — reduce both
             # halves separately
-            dw32 = torch.empty(H, dtype=torch.float32, device=s.device)
+            dw32 = torch.zeros(H, dtype=torch.float32, device=s.device)
             db32 = torch.empty(H, dtype=torch.float32, device=s.device)
             hip.check(hip.lib().dolomite_reduce_partials(hip.stream(), hip.ptr(dwdb), hip.ptr(dw32), nb, H), "rp")
             hip.check(
