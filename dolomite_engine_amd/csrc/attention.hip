@@ -33,20 +33,19 @@ typedef float f32x4 __attribute__((ext_vector_type(4)));
 
 #define MFMA16(a, b, c) __builtin_amdgcn_mfma_f32_16x16x32_bf16((a), (b), (c), 0, 0, 0)
 
-// Load 8 consecutive bf16 from global with bounds/padding guard.
-// `base` points at element d0 of a row with `dmax` valid elements from d0's
-// row origin; d0 must be the element offset within the row.
+// Branchless guarded 16B load. Requires D % 8 == 0 (every head dim on this
+// path): a chunk is then fully inside [0, D) or fully in the pad, so the
+// guard reduces to ONE wave-divergent-free vector load from a clamped
+// address plus per-element selects. (The original branchy version emitted
+// 8 scalar loads + a vector load per fragment and forced hipcc to drain
+// vmcnt(0) before every dependent MFMA — the dominant stall of round 1's
+// backward kernel.)
 __device__ __forceinline__ bf16x8 load_bf16x8_guard(const __bf16* p, int d0, int D, bool valid) {
-    bf16x8 r;
-    if (valid && d0 + 8 <= D) {
-        r = *(const bf16x8*)p;
-    } else {
+    const __bf16* q = (d0 < D) ? p : (p - d0);  // clamp into the row
+    bool ok = valid && (d0 < D);
+    bf16x8 r = *(const bf16x8*)q;
 #pragma unroll
-        for (int e = 0; e < 8; ++e) {
-            float f = (valid && d0 + e < D) ? (float)p[e] : 0.f;
-            r[e] = (__bf16)f;
-        }
-    }
+    for (int e = 0; e < 8; ++e) r[e] = ok ? r[e] : (__bf16)0.f;
     return r;
 }
 
@@ -83,8 +82,11 @@ __global__ void __launch_bounds__(256) fa_fwd_kernel(
     const int h = blockIdx.z;
     const int s0 = cu[b];
     const int L = cu[b + 1] - s0;
-    const int qs = (int)blockIdx.x * 64;
-    if (qs >= L) return;
+    // heaviest tiles (largest qs -> most k-tiles) dispatch FIRST: in-order
+    // dispatch otherwise schedules the long-pole causal workgroups last
+    const int ntile_seq = (L + 63) / 64;
+    if ((int)blockIdx.x >= ntile_seq) return;
+    const int qs = (ntile_seq - 1 - (int)blockIdx.x) * 64;
 
     const int lane = threadIdx.x & 63;
     const int wave = threadIdx.x >> 6;
@@ -422,21 +424,46 @@ __global__ void __launch_bounds__(256) fa_bwd_kernel(
         // B-frags (Q^T, dO^T as [k=d][j=q]) are 16B rows of Q/dO: read from
         // global (L2-hot across the 4 waves / q-tile revisits)
         f32x4 st[4], dpt[4];
-#pragma unroll
-        for (int cb = 0; cb < 4; ++cb) {
-            st[cb] = {0.f, 0.f, 0.f, 0.f};
-            dpt[cb] = {0.f, 0.f, 0.f, 0.f};
-            const int qrow = qs + cb * 16 + lr;
+        // double-buffered B-frag loads: issue block cb+1's 2*KCH loads before
+        // block cb's MFMAs so the global latency hides under the matrix work
+        bf16x8 qcur[KCH], dcur[KCH], qnxt[KCH], dnxt[KCH];
+        {
+            const int qrow = qs + lr;
             const bool qok = qrow < L;
             const __bf16* qrp = q + (int64_t)(s0 + (qok ? qrow : 0)) * q_ts + q_hoff;
             const __bf16* drp = dout + (int64_t)(s0 + (qok ? qrow : 0)) * do_ts + do_hoff;
 #pragma unroll
             for (int kc = 0; kc < KCH; ++kc) {
                 int d0 = kc * 32 + lg * 8;
-                bf16x8 qtf = load_bf16x8_guard(qrp + d0, d0, D, qok);
-                st[cb] = MFMA16(kfr[kc], qtf, st[cb]);
-                bf16x8 dtf = load_bf16x8_guard(drp + d0, d0, D, qok);
-                dpt[cb] = MFMA16(vfr[kc], dtf, dpt[cb]);
+                qcur[kc] = load_bf16x8_guard(qrp + d0, d0, D, qok);
+                dcur[kc] = load_bf16x8_guard(drp + d0, d0, D, qok);
+            }
+        }
+#pragma unroll
+        for (int cb = 0; cb < 4; ++cb) {
+            if (cb < 3) {
+                const int qrow = qs + (cb + 1) * 16 + lr;
+                const bool qok = qrow < L;
+                const __bf16* qrp = q + (int64_t)(s0 + (qok ? qrow : 0)) * q_ts + q_hoff;
+                const __bf16* drp = dout + (int64_t)(s0 + (qok ? qrow : 0)) * do_ts + do_hoff;
+#pragma unroll
+                for (int kc = 0; kc < KCH; ++kc) {
+                    int d0 = kc * 32 + lg * 8;
+                    qnxt[kc] = load_bf16x8_guard(qrp + d0, d0, D, qok);
+                    dnxt[kc] = load_bf16x8_guard(drp + d0, d0, D, qok);
+                }
+            }
+            st[cb] = {0.f, 0.f, 0.f, 0.f};
+            dpt[cb] = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+            for (int kc = 0; kc < KCH; ++kc) {
+                st[cb] = MFMA16(kfr[kc], qcur[kc], st[cb]);
+                dpt[cb] = MFMA16(vfr[kc], dcur[kc], dpt[cb]);
+            }
+#pragma unroll
+            for (int kc = 0; kc < KCH; ++kc) {
+                qcur[kc] = qnxt[kc];
+                dcur[kc] = dnxt[kc];
             }
         }
 
