@@ -316,7 +316,7 @@ extern "C" int dolomite_fa_bwd_preprocess(dolomite_stream_t stream,
 // ===========================================================================
 
 template <int DPAD>
-__global__ void __launch_bounds__(256) fa_bwd_kernel(
+__global__ void __launch_bounds__(256, 2) fa_bwd_kernel(
     const __bf16* __restrict__ q, const __bf16* __restrict__ k, const __bf16* __restrict__ v,
     const __bf16* __restrict__ dout, const float* __restrict__ lse, const float* __restrict__ delta,
     float* __restrict__ dq_acc, float* __restrict__ dk_acc, float* __restrict__ dv_acc,

@@ -82,3 +82,54 @@ __device__ __forceinline__ float block_reduce_sum(float v, float* red, int nthre
     __syncthreads();
     return total;
 }
+
+// ---- vectorized row I/O: one wide load/store per V elements --------------
+// (hipcc does not merge scalar bf16 accesses well — guide §5 common mistake 2)
+
+typedef unsigned short us8_t __attribute__((ext_vector_type(8)));
+typedef float f4_t __attribute__((ext_vector_type(4)));
+
+template <typename T, int V>
+struct VecIO;
+
+template <>
+struct VecIO<uint16_t, 8> {
+    static __device__ __forceinline__ void load(const uint16_t* p, float* out) {
+        us8_t v = *(const us8_t*)p;
+#pragma unroll
+        for (int e = 0; e < 8; ++e) out[e] = bf16_to_f32((uint16_t)v[e]);
+    }
+    static __device__ __forceinline__ void store(uint16_t* p, const float* in) {
+        us8_t v;
+#pragma unroll
+        for (int e = 0; e < 8; ++e) v[e] = f32_to_bf16(in[e]);
+        *(us8_t*)p = v;
+    }
+};
+
+template <>
+struct VecIO<float, 4> {
+    static __device__ __forceinline__ void load(const float* p, float* out) {
+        f4_t v = *(const f4_t*)p;
+#pragma unroll
+        for (int e = 0; e < 4; ++e) out[e] = v[e];
+    }
+    static __device__ __forceinline__ void store(float* p, const float* in) {
+        f4_t v;
+#pragma unroll
+        for (int e = 0; e < 4; ++e) v[e] = in[e];
+        *(f4_t*)p = v;
+    }
+};
+
+template <>
+struct VecIO<uint16_t, 1> {
+    static __device__ __forceinline__ void load(const uint16_t* p, float* out) { out[0] = bf16_to_f32(*p); }
+    static __device__ __forceinline__ void store(uint16_t* p, const float* in) { *p = f32_to_bf16(in[0]); }
+};
+
+template <>
+struct VecIO<float, 1> {
+    static __device__ __forceinline__ void load(const float* p, float* out) { out[0] = *p; }
+    static __device__ __forceinline__ void store(float* p, const float* in) { *p = in[0]; }
+};

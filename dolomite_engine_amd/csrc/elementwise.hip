@@ -44,19 +44,21 @@ __global__ void __launch_bounds__(256) rmsnorm_fwd_kernel(
     for (int i = 0; i < ITMAX; ++i) {
         int c = (i * 64 + lane) * V;
         if (c < H) {
+            VecIO<T, V>::load(xr + c, buf[i]);
+            if (HAS_RES) {
+                float rv[V];
+                VecIO<T, V>::load(rr + c, rv);
 #pragma unroll
-            for (int kk = 0; kk < V; ++kk) {
-                float v = load_as_f32(xr + c + kk);
-                if (HAS_RES) {
+                for (int kk = 0; kk < V; ++kk) {
                     // tensor-dtype residual add (layer.py:74,87): round the sum
-                    v += load_as_f32(rr + c + kk);
+                    float v = buf[i][kk] + rv[kk];
                     T tmp;
                     store_from_f32(&tmp, v);
-                    v = load_as_f32(&tmp);
+                    buf[i][kk] = load_as_f32(&tmp);
                 }
-                buf[i][kk] = v;
-                ss += v * v;
             }
+#pragma unroll
+            for (int kk = 0; kk < V; ++kk) ss += buf[i][kk] * buf[i][kk];
         }
     }
     float rstd = rsqrtf(wave_reduce_sum(ss) / (float)H + eps);
@@ -68,14 +70,16 @@ __global__ void __launch_bounds__(256) rmsnorm_fwd_kernel(
     for (int i = 0; i < ITMAX; ++i) {
         int c = (i * 64 + lane) * V;
         if (c < H) {
+            if (HAS_RES) VecIO<T, V>::store(sr + c, buf[i]);
+            float wv[V], ov[V];
+            VecIO<T, V>::load(w + c, wv);
 #pragma unroll
             for (int kk = 0; kk < V; ++kk) {
-                float v = buf[i][kk];
-                if (HAS_RES) store_from_f32(sr + c + kk, v);
                 T tmp;
-                store_from_f32(&tmp, v * rstd);  // cast before weight (rmsnorm/base.py:23-25)
-                store_from_f32(yr + c + kk, load_as_f32(w + c + kk) * load_as_f32(&tmp));
+                store_from_f32(&tmp, buf[i][kk] * rstd);  // cast before weight (rmsnorm/base.py:23-25)
+                ov[kk] = wv[kk] * load_as_f32(&tmp);
             }
+            VecIO<T, V>::store(yr + c, ov);
         }
     }
 }
@@ -173,14 +177,19 @@ __global__ void __launch_bounds__(256) norm_bwd_kernel(
 
     float wv[ITMAX][V], dw_acc[ITMAX][V], db_acc[IS_LN ? ITMAX : 1][IS_LN ? V : 1];
 #pragma unroll
-    for (int i = 0; i < ITMAX; ++i)
+    for (int i = 0; i < ITMAX; ++i) {
+        int c0 = (i * 64 + lane) * V;
+        if (c0 < H)
+            VecIO<T, V>::load(w + c0, wv[i]);
+        else
+#pragma unroll
+            for (int kk = 0; kk < V; ++kk) wv[i][kk] = 0.f;
 #pragma unroll
         for (int kk = 0; kk < V; ++kk) {
-            int c = (i * 64 + lane) * V + kk;
-            wv[i][kk] = (c < H) ? load_as_f32(w + c) : 0.f;
             dw_acc[i][kk] = 0.f;
             if (IS_LN) db_acc[i][kk] = 0.f;
         }
+    }
 
     for (int64_t row = wslot; row < T_rows; row += nslots) {
         const T* dyr = dy + row * (int64_t)H;
@@ -194,13 +203,13 @@ __global__ void __launch_bounds__(256) norm_bwd_kernel(
         for (int i = 0; i < ITMAX; ++i) {
             int c0 = (i * 64 + lane) * V;
             if (c0 < H) {
+                VecIO<T, V>::load(sr + c0, sh[i]);
+                VecIO<T, V>::load(dyr + c0, dyv[i]);
 #pragma unroll
                 for (int kk = 0; kk < V; ++kk) {
-                    float shat = (load_as_f32(sr + c0 + kk) - mu) * r;
-                    float dyy = load_as_f32(dyr + c0 + kk);
-                    float dyw = dyy * wv[i][kk];
+                    float shat = (sh[i][kk] - mu) * r;
+                    float dyw = dyv[i][kk] * wv[i][kk];
                     sh[i][kk] = shat;
-                    dyv[i][kk] = dyy;
                     d2 += dyw * shat;
                     if (IS_LN) d1 += dyw;
                 }
@@ -217,23 +226,23 @@ __global__ void __launch_bounds__(256) norm_bwd_kernel(
         for (int i = 0; i < ITMAX; ++i) {
             int c0 = (i * 64 + lane) * V;
             if (c0 < H) {
+                float dxv[V];
 #pragma unroll
                 for (int kk = 0; kk < V; ++kk) {
-                    float dxv;
                     if (IS_LN) {
-                        dxv = r * (dyv[i][kk] * wv[i][kk] - m1 - sh[i][kk] * m2);
+                        dxv[kk] = r * (dyv[i][kk] * wv[i][kk] - m1 - sh[i][kk] * m2);
                         dw_acc[i][kk] += dyv[i][kk] * sh[i][kk];
                         db_acc[i][kk] += dyv[i][kk];
                     } else {
-                        dxv = r * (wv[i][kk] * dyv[i][kk] - sh[i][kk] * m2);
+                        dxv[kk] = r * (wv[i][kk] * dyv[i][kk] - sh[i][kk] * m2);
                         // dw uses the CAST normalized value (reference casts
                         // before the weight multiply)
                         T tmp;
                         store_from_f32(&tmp, sh[i][kk]);
                         dw_acc[i][kk] += dyv[i][kk] * load_as_f32(&tmp);
                     }
-                    store_from_f32(dxr + c0 + kk, dxv);
                 }
+                VecIO<T, V>::store(dxr + c0, dxv);
             }
         }
     }
@@ -303,18 +312,20 @@ __global__ void __launch_bounds__(256) layernorm_fwd_kernel(
     for (int i = 0; i < ITMAX; ++i) {
         int c = (i * 64 + lane) * V;
         if (c < H) {
+            VecIO<T, V>::load(xr + c, buf[i]);
+            if (HAS_RES) {
+                float rv[V];
+                VecIO<T, V>::load(rr + c, rv);
 #pragma unroll
-            for (int kk = 0; kk < V; ++kk) {
-                float v = load_as_f32(xr + c + kk);
-                if (HAS_RES) {
-                    v += load_as_f32(rr + c + kk);
+                for (int kk = 0; kk < V; ++kk) {
+                    float v = buf[i][kk] + rv[kk];
                     T tmp;
                     store_from_f32(&tmp, v);
-                    v = load_as_f32(&tmp);
+                    buf[i][kk] = load_as_f32(&tmp);
                 }
-                buf[i][kk] = v;
-                sum += v;
             }
+#pragma unroll
+            for (int kk = 0; kk < V; ++kk) sum += buf[i][kk];
         }
     }
     float mu = wave_reduce_sum(sum) / (float)H;
@@ -341,13 +352,13 @@ __global__ void __launch_bounds__(256) layernorm_fwd_kernel(
     for (int i = 0; i < ITMAX; ++i) {
         int c = (i * 64 + lane) * V;
         if (c < H) {
+            if (HAS_RES) VecIO<T, V>::store(sr + c, buf[i]);
+            float wv[V], bv[V], ov[V];
+            VecIO<T, V>::load(w + c, wv);
+            VecIO<T, V>::load(b + c, bv);
 #pragma unroll
-            for (int kk = 0; kk < V; ++kk) {
-                float v = buf[i][kk];
-                if (HAS_RES) store_from_f32(sr + c + kk, v);
-                float out = (v - mu) * rstd * load_as_f32(w + c + kk) + load_as_f32(b + c + kk);
-                store_from_f32(yr + c + kk, out);
-            }
+            for (int kk = 0; kk < V; ++kk) ov[kk] = (buf[i][kk] - mu) * rstd * wv[kk] + bv[kk];
+            VecIO<T, V>::store(yr + c, ov);
         }
     }
 }
@@ -538,16 +549,22 @@ __global__ void __launch_bounds__(256) ce_fwd_kernel(
 
     float m = -INFINITY, ssum = 0.f;
     for (int64_t c = threadIdx.x * V; c < V_dim; c += (int64_t)blockDim.x * V) {
+        float xv[V];
+        if (c + V <= V_dim) {
+            VecIO<T, V>::load(xr + c, xv);
+        } else {
 #pragma unroll
-        for (int k = 0; k < V; ++k)
-            if (c + k < V_dim) {
-                float v = load_as_f32(xr + c + k);
-                if (v > m) {
-                    ssum *= __expf(m - v);
-                    m = v;
-                }
-                ssum += __expf(v - m);
+            for (int k = 0; k < V; ++k) xv[k] = (c + k < V_dim) ? load_as_f32(xr + c + k) : -INFINITY;
+        }
+#pragma unroll
+        for (int k = 0; k < V; ++k) {
+            float v = xv[k];
+            if (v > m) {
+                ssum *= __expf(m - v);
+                m = v;
             }
+            ssum += __expf(v - m);
+        }
     }
     // block combine: global max then shift partial sums
     float mall = m;
@@ -608,20 +625,36 @@ __global__ void __launch_bounds__(256) ce_bwd_kernel(
     int64_t lab = labels[t];
     float l = lse[t];
     if (lab == ignore_index) {
-        for (int64_t c = threadIdx.x * V; c < V_dim; c += (int64_t)blockDim.x * V)
+        float zv[V] = {};
+        for (int64_t c = threadIdx.x * V; c < V_dim; c += (int64_t)blockDim.x * V) {
+            if (c + V <= V_dim)
+                VecIO<T, V>::store(dr + c, zv);
+            else
 #pragma unroll
-            for (int k = 0; k < V; ++k)
-                if (c + k < V_dim) store_from_f32(dr + c + k, 0.f);
+                for (int k = 0; k < V; ++k)
+                    if (c + k < V_dim) store_from_f32(dr + c + k, 0.f);
+        }
         return;
     }
     for (int64_t c = threadIdx.x * V; c < V_dim; c += (int64_t)blockDim.x * V) {
+        if (c + V <= V_dim) {
+            float xv[V], gv[V];
+            VecIO<T, V>::load(xr + c, xv);
 #pragma unroll
-        for (int k = 0; k < V; ++k)
-            if (c + k < V_dim) {
-                float p = __expf(load_as_f32(xr + c + k) - l);
-                float g = (c + k == lab) ? (p - 1.f) : p;
-                store_from_f32(dr + c + k, g * gs);
+            for (int k = 0; k < V; ++k) {
+                float p = __expf(xv[k] - l);
+                gv[k] = ((c + k == lab) ? (p - 1.f) : p) * gs;
             }
+            VecIO<T, V>::store(dr + c, gv);
+        } else {
+#pragma unroll
+            for (int k = 0; k < V; ++k)
+                if (c + k < V_dim) {
+                    float p = __expf(load_as_f32(xr + c + k) - l);
+                    float g = (c + k == lab) ? (p - 1.f) : p;
+                    store_from_f32(dr + c + k, g * gs);
+                }
+        }
     }
 }
 

@@ -228,15 +228,10 @@ class FusedLayerNorm(torch.autograd.Function):
                 ),
                 "layernorm_bwd",
             )
-            out32 = torch.empty(2, H, dtype=torch.float32, device=s.device)
-            hip.check(
-                hip.lib().dolomite_reduce_partials(hip.stream(), hip.ptr(dwdb), hip.ptr(out32), 2 * nb, H),
-                "reduce_partials",
-            )
             # dw partials are rows [0, nb), db rows [nb, 2nb) — reduce both
-            # halves separately
+            # halves separately (out buffers zeroed: atomic join)
             dw32 = torch.zeros(H, dtype=torch.float32, device=s.device)
-            db32 = torch.empty(H, dtype=torch.float32, device=s.device)
+            db32 = torch.zeros(H, dtype=torch.float32, device=s.device)
             hip.check(hip.lib().dolomite_reduce_partials(hip.stream(), hip.ptr(dwdb), hip.ptr(dw32), nb, H), "rp")
             hip.check(
                 hip.lib().dolomite_reduce_partials(hip.stream(), hip.ptr(dwdb, nb * H), hip.ptr(db32), nb, H), "rp"
