@@ -315,20 +315,29 @@ extern "C" int dolomite_fa_bwd_preprocess(dolomite_stream_t stream,
 // Backward pass 2: per (kv-tile, seq, kv-head) workgroup.
 // ===========================================================================
 
+// Backward is split FA2-style into two kernels (each recomputes P from
+// q/k/lse — cheaper than the atomics+barriers a fused version needs):
+//   fa_bwd_dkv_kernel : grid (kv-tile, seq, q-head); dK/dV accumulate in
+//     registers over the q-tile loop, joined into fp32 buffers by one atomic
+//     pass per workgroup (G q-head contributors per kv strip).
+//   fa_bwd_dq_kernel  : grid (q-tile, seq, q-head); dQ accumulates in
+//     registers over the kv-tile loop — no atomics at all — and is stored
+//     once. K^T is the only LDS image (swizzled), staged per kv tile.
+
 template <int DPAD>
-__global__ void __launch_bounds__(256, 2) fa_bwd_kernel(
+__global__ void __launch_bounds__(256, 2) fa_bwd_dkv_kernel(
     const __bf16* __restrict__ q, const __bf16* __restrict__ k, const __bf16* __restrict__ v,
     const __bf16* __restrict__ dout, const float* __restrict__ lse, const float* __restrict__ delta,
-    float* __restrict__ dq_acc, float* __restrict__ dk_acc, float* __restrict__ dv_acc,
+    float* __restrict__ dk_acc, float* __restrict__ dv_acc,
     const int32_t* __restrict__ cu, int H, int Hkv, int D, int G,
     int64_t q_ts, int64_t q_gs, int64_t k_ts, int64_t k_hs, int64_t v_ts, int64_t v_hs,
     int64_t do_ts, int64_t T_total, float scale) {
     constexpr int KCH = DPAD / 32;
     constexpr int DCH = DPAD / 16;
-    constexpr int ST = 64 + 8;     // transposed-image / [q][key] row stride
+    constexpr int ST = 64 + 8;
 
     const int b = blockIdx.y;
-    const int h = blockIdx.z;        // q head: one workgroup per (kv-tile, q-head)
+    const int h = blockIdx.z;
     const int kvh = h / G;
     const int s0 = cu[b];
     const int L = cu[b + 1] - s0;
@@ -340,17 +349,10 @@ __global__ void __launch_bounds__(256, 2) fa_bwd_kernel(
     const int lr = lane & 15;
     const int lg = lane >> 4;
 
-    // Transposed images use an XOR block swizzle: element (row, col) lives at
-    // row*ST + ((col>>3 ^ (row>>3)&7)<<3) + (col&7). This keeps every
-    // ds_read_b128 16-byte aligned while spreading the 8-consecutive-row
-    // scatter writes of the transpose staging over 8 bank groups (the linear
-    // layout put all of them on one bank: 12-way conflict).
     extern __shared__ char smem_raw[];
     __bf16* QTl = (__bf16*)smem_raw;           // [DPAD][ST]   (Q^T, swizzled)
     __bf16* dOTl = QTl + DPAD * ST;            // [DPAD][ST]   (dO^T, swizzled)
-    __bf16* KTl = dOTl + DPAD * ST;            // [DPAD][ST]   (K^T, swizzled)
-    __bf16* dSl = KTl + DPAD * ST;             // [64 q][ST]   (dS, [q][key])
-    __bf16* dSTl = dSl + 64 * ST;              // [64 key][ST] (dS^T, [key][q])
+    __bf16* dSTl = dOTl + DPAD * ST;           // [64 key][ST] (dS^T, [key][q])
     __bf16* PTl = dSTl + 64 * ST;              // [64 key][ST] (P^T, [key][q])
 
 #define SWZ(row, col) ((row) * ST + (((((col) >> 3) ^ (((row) >> 3) & 7)) << 3) | ((col) & 7)))
@@ -358,7 +360,6 @@ __global__ void __launch_bounds__(256, 2) fa_bwd_kernel(
 
     const int kend = min(L, ks + 64);
 
-    // --- this wave's K and V fragments (A-layout: i = lr -> key) ---
     const int krow = ks + wave * 16 + lr;
     const bool kvalid = krow < kend;
     bf16x8 kfr[KCH], vfr[KCH];
@@ -369,19 +370,6 @@ __global__ void __launch_bounds__(256, 2) fa_bwd_kernel(
         kfr[kc] = load_bf16x8_guard(kp, d0, D, kvalid);
         const __bf16* vp = v + (int64_t)(s0 + (kvalid ? krow : 0)) * v_ts + (int64_t)kvh * v_hs + d0;
         vfr[kc] = load_bf16x8_guard(vp, d0, D, kvalid);
-    }
-    // --- stage K^T image once (swizzled scatter) ---
-    {
-        const int pieces = 64 * DPAD / 8;
-        for (int pidx = threadIdx.x; pidx < pieces; pidx += 256) {
-            int key = pidx / (DPAD / 8);
-            int d0 = (pidx % (DPAD / 8)) * 8;
-            bool valid = (ks + key) < kend;
-            const __bf16* kp = k + (int64_t)(s0 + (valid ? ks + key : 0)) * k_ts + (int64_t)kvh * k_hs + d0;
-            bf16x8 kk = load_bf16x8_guard(kp, d0, D, valid);
-#pragma unroll
-            for (int e = 0; e < 8; ++e) KTl[SWZ(d0 + e, key)] = kk[e];
-        }
     }
 
     f32x4 dvr[DCH], dkr[DCH];
@@ -398,9 +386,7 @@ __global__ void __launch_bounds__(256, 2) fa_bwd_kernel(
 
     for (int qt = qt0; qt < nqt; ++qt) {
         const int qs = qt * 64;
-        // --- stage Q^T and dO^T (swizzled transposes; row-major reads of
-        //     Q/dO go straight to global where they are 16B-contiguous) ---
-        __syncthreads();  // previous iteration's reads done
+        __syncthreads();  // previous iteration's image reads done
         {
             const int pieces = 64 * DPAD / 8;
             for (int pidx = threadIdx.x; pidx < pieces; pidx += 256) {
@@ -419,13 +405,8 @@ __global__ void __launch_bounds__(256, 2) fa_bwd_kernel(
         }
         __syncthreads();
 
-        // --- S^T = K*Q^T, P^T = exp(scale*S^T - lse) ; dP^T = V*dO^T ---
-        // C layout: col = q = lr, row = key = 16*wave + lg*4 + r
-        // B-frags (Q^T, dO^T as [k=d][j=q]) are 16B rows of Q/dO: read from
-        // global (L2-hot across the 4 waves / q-tile revisits)
+        // S^T = K*Q^T, dP^T = V*dO^T (B-frags straight from global, dbuf'd)
         f32x4 st[4], dpt[4];
-        // double-buffered B-frag loads: issue block cb+1's 2*KCH loads before
-        // block cb's MFMAs so the global latency hides under the matrix work
         bf16x8 qcur[KCH], dcur[KCH], qnxt[KCH], dnxt[KCH];
         {
             const int qrow = qs + lr;
@@ -467,6 +448,8 @@ __global__ void __launch_bounds__(256, 2) fa_bwd_kernel(
             }
         }
 
+        // P^T, dS^T -> own-wave LDS strips (no barrier: each wave reads only
+        // the 16-key rows it wrote)
 #pragma unroll
         for (int cb = 0; cb < 4; ++cb) {
             const int qpos = qs + cb * 16 + lr;
@@ -481,12 +464,10 @@ __global__ void __launch_bounds__(256, 2) fa_bwd_kernel(
                 float ds = ok ? pv * (dpt[cb][r] - delv) * scale : 0.f;
                 PTl[(wave * 16 + lg * 4 + r) * ST + cb * 16 + lr] = (__bf16)pv;
                 dSTl[(wave * 16 + lg * 4 + r) * ST + cb * 16 + lr] = (__bf16)ds;
-                dSl[(cb * 16 + lr) * ST + wave * 16 + lg * 4 + r] = (__bf16)ds;
             }
         }
-        __syncthreads();
 
-        // --- dV += P^T*dO ; dK += dS^T*Q  (contraction over q) ---
+        // dV += P^T*dO ; dK += dS^T*Q (contraction over q)
 #pragma unroll
         for (int kc2 = 0; kc2 < 2; ++kc2) {
             bf16x8 ptf = *(const bf16x8*)&PTl[(wave * 16 + lr) * ST + kc2 * 32 + lg * 8];
@@ -499,28 +480,9 @@ __global__ void __launch_bounds__(256, 2) fa_bwd_kernel(
                 dkr[dc] = MFMA16(dstf, qtb, dkr[dc]);
             }
         }
-
-        // --- dQ strip (this wave's 16 q rows): dS*K^T-image, atomics ---
-#pragma unroll
-        for (int dc = 0; dc < DCH; ++dc) {
-            f32x4 dq = {0.f, 0.f, 0.f, 0.f};
-#pragma unroll
-            for (int kc2 = 0; kc2 < 2; ++kc2) {
-                bf16x8 dsf = *(const bf16x8*)&dSl[(wave * 16 + lr) * ST + kc2 * 32 + lg * 8];
-                bf16x8 ktb = *(const bf16x8*)&KTl[SWZ8(dc * 16 + lr, kc2 * 32 + lg * 8)];
-                dq = MFMA16(dsf, ktb, dq);
-            }
-#pragma unroll
-            for (int r = 0; r < 4; ++r) {
-                const int qpos = qs + wave * 16 + lg * 4 + r;
-                const int d = dc * 16 + lr;
-                if (qpos < L && d < D)
-                    atomicAdd(&dq_acc[((int64_t)(s0 + qpos) * H + h) * D + d], dq[r]);
-            }
-        }
     }
 
-    // --- accumulate dK/dV (G q-head workgroups contribute per kv strip) ---
+    // join dK/dV (G q-head contributors per kv strip)
 #pragma unroll
     for (int dc = 0; dc < DCH; ++dc) {
 #pragma unroll
@@ -539,17 +501,173 @@ __global__ void __launch_bounds__(256, 2) fa_bwd_kernel(
 }
 
 template <int DPAD>
+__global__ void __launch_bounds__(256) fa_bwd_dq_kernel(
+    const __bf16* __restrict__ q, const __bf16* __restrict__ k, const __bf16* __restrict__ v,
+    const __bf16* __restrict__ dout, const float* __restrict__ lse, const float* __restrict__ delta,
+    float* __restrict__ dq_acc,
+    const int32_t* __restrict__ cu, int H, int Hkv, int D, int G,
+    int64_t q_ts, int64_t q_gs, int64_t k_ts, int64_t k_hs, int64_t v_ts, int64_t v_hs,
+    int64_t do_ts, int64_t T_total, float scale) {
+    constexpr int KCH = DPAD / 32;
+    constexpr int DCH = DPAD / 16;
+    constexpr int ST = 64 + 8;
+
+    const int b = blockIdx.y;
+    const int h = blockIdx.z;
+    const int kvh = h / G;
+    const int s0 = cu[b];
+    const int L = cu[b + 1] - s0;
+    // heaviest q-tiles (most kv tiles) first
+    const int ntile_seq = (L + 63) / 64;
+    if ((int)blockIdx.x >= ntile_seq) return;
+    const int qs = (ntile_seq - 1 - (int)blockIdx.x) * 64;
+
+    const int lane = threadIdx.x & 63;
+    const int wave = threadIdx.x >> 6;
+    const int lr = lane & 15;
+    const int lg = lane >> 4;
+
+    extern __shared__ char smem_raw[];
+    __bf16* KTl = (__bf16*)smem_raw;           // [DPAD][ST] (K^T, swizzled)
+    __bf16* dSl = KTl + DPAD * ST;             // [4 waves][16 q][ST] (dS strips)
+    __bf16* dSw = dSl + wave * 16 * ST;
+
+#define SWZ(row, col) ((row) * ST + (((((col) >> 3) ^ (((row) >> 3) & 7)) << 3) | ((col) & 7)))
+#define SWZ8(row, col0) ((row) * ST + (((((col0) >> 3) ^ (((row) >> 3) & 7)) << 3)))
+
+    const int64_t q_hoff = (int64_t)(h / G) * q_gs + (int64_t)(h % G) * D;
+    const int64_t do_hoff = (int64_t)h * D;
+
+    // this wave's Q and dO fragments (A-layout: i = lr -> q row)
+    const int qrow = qs + wave * 16 + lr;
+    const bool qvalid = qrow < L;
+    bf16x8 qfr[KCH], dfr[KCH];
+#pragma unroll
+    for (int kc = 0; kc < KCH; ++kc) {
+        int d0 = kc * 32 + lg * 8;
+        const __bf16* qp = q + (int64_t)(s0 + (qvalid ? qrow : 0)) * q_ts + q_hoff + d0;
+        qfr[kc] = load_bf16x8_guard(qp, d0, D, qvalid);
+        const __bf16* dp = dout + (int64_t)(s0 + (qvalid ? qrow : 0)) * do_ts + do_hoff + d0;
+        dfr[kc] = load_bf16x8_guard(dp, d0, D, qvalid);
+    }
+    const int qpos_r[1] = {};  // (silence unused in some instantiations)
+    (void)qpos_r;
+
+    float lsev[4], delv[4];
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+        const int qp_ = qs + wave * 16 + lg * 4 + r;
+        bool ok = qp_ < L;
+        lsev[r] = ok ? lse[(int64_t)h * T_total + s0 + qp_] : 0.f;
+        delv[r] = ok ? delta[(int64_t)h * T_total + s0 + qp_] : 0.f;
+    }
+
+    f32x4 dq[DCH];
+#pragma unroll
+    for (int dc = 0; dc < DCH; ++dc) dq[dc] = {0.f, 0.f, 0.f, 0.f};
+
+    const int kend_total = min(L, qs + 64);
+    const int nkt = (kend_total + 63) / 64;
+
+    for (int kt = 0; kt < nkt; ++kt) {
+        const int ks = kt * 64;
+        __syncthreads();  // previous tile's K^T reads done
+        // stage K^T (swizzled scatter), cooperative
+        {
+            const int pieces = 64 * DPAD / 8;
+            for (int pidx = threadIdx.x; pidx < pieces; pidx += 256) {
+                int key = pidx / (DPAD / 8);
+                int d0 = (pidx % (DPAD / 8)) * 8;
+                bool valid = (ks + key) < kend_total;
+                const __bf16* kp = k + (int64_t)(s0 + (valid ? ks + key : 0)) * k_ts + (int64_t)kvh * k_hs + d0;
+                bf16x8 kk = load_bf16x8_guard(kp, d0, D, valid);
+#pragma unroll
+                for (int e = 0; e < 8; ++e) KTl[SWZ(d0 + e, key)] = kk[e];
+            }
+        }
+        __syncthreads();
+
+        // S = Q*K^T and dP = dO*V^T; B-frags are K/V rows read from global
+        f32x4 sc[4], dp[4];
+#pragma unroll
+        for (int cb = 0; cb < 4; ++cb) {
+            sc[cb] = {0.f, 0.f, 0.f, 0.f};
+            dp[cb] = {0.f, 0.f, 0.f, 0.f};
+            const int key = ks + cb * 16 + lr;
+            const bool kok = key < kend_total;
+            const __bf16* krp = k + (int64_t)(s0 + (kok ? key : 0)) * k_ts + (int64_t)kvh * k_hs;
+            const __bf16* vrp = v + (int64_t)(s0 + (kok ? key : 0)) * v_ts + (int64_t)kvh * v_hs;
+#pragma unroll
+            for (int kc = 0; kc < KCH; ++kc) {
+                int d0 = kc * 32 + lg * 8;
+                bf16x8 kb = load_bf16x8_guard(krp + d0, d0, D, kok);
+                sc[cb] = MFMA16(qfr[kc], kb, sc[cb]);
+                bf16x8 vb = load_bf16x8_guard(vrp + d0, d0, D, kok);
+                dp[cb] = MFMA16(dfr[kc], vb, dp[cb]);
+            }
+        }
+
+        // dS = P*(dP - delta)*scale, C-layout (row = q, col = key);
+        // stash into this wave's own dS strip (no cross-wave use)
+#pragma unroll
+        for (int cb = 0; cb < 4; ++cb) {
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                const int qpos = qs + wave * 16 + lg * 4 + r;
+                const int kpos = ks + cb * 16 + lr;
+                bool ok = qpos < L && kpos < kend_total && kpos <= qpos;
+                float pv = ok ? __expf(sc[cb][r] * scale - lsev[r]) : 0.f;
+                float ds = ok ? pv * (dp[cb][r] - delv[r]) * scale : 0.f;
+                dSw[(lg * 4 + r) * ST + cb * 16 + lr] = (__bf16)ds;
+            }
+        }
+
+        // dQ += dS*K (contraction over this tile's keys)
+#pragma unroll
+        for (int kc2 = 0; kc2 < 2; ++kc2) {
+            bf16x8 dsf = *(const bf16x8*)&dSw[lr * ST + kc2 * 32 + lg * 8];
+#pragma unroll
+            for (int dc = 0; dc < DCH; ++dc) {
+                bf16x8 ktb = *(const bf16x8*)&KTl[SWZ8(dc * 16 + lr, kc2 * 32 + lg * 8)];
+                dq[dc] = MFMA16(dsf, ktb, dq[dc]);
+            }
+        }
+    }
+
+    // single plain store of the accumulated dQ
+#pragma unroll
+    for (int dc = 0; dc < DCH; ++dc) {
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+            const int qpos = qs + wave * 16 + lg * 4 + r;
+            const int d = dc * 16 + lr;
+            if (qpos < L && d < D)
+                dq_acc[((int64_t)(s0 + qpos) * H + h) * D + d] = dq[dc][r];
+        }
+    }
+#undef SWZ
+#undef SWZ8
+}
+
+template <int DPAD>
 static int launch_fa_bwd(hipStream_t stream, const __bf16* q, const __bf16* k, const __bf16* v,
                          const __bf16* dout, const float* lse, const float* delta,
                          float* dq_acc, float* dk_acc, float* dv_acc,
                          const int32_t* cu, int batch, int64_t T, int H, int Hkv, int D, int G,
                          int64_t q_ts, int64_t q_gs, int64_t k_ts, int64_t k_hs,
                          int64_t v_ts, int64_t v_hs, int64_t do_ts, int max_tiles, float scale) {
-    dim3 grid(max_tiles, batch, H), block(256);
     constexpr int ST = 64 + 8;
-    size_t shmem = (size_t)(DPAD * ST * 3 + 64 * ST * 3) * sizeof(__bf16);
-    hipLaunchKernelGGL((fa_bwd_kernel<DPAD>), grid, block, shmem, stream,
-                       q, k, v, dout, lse, delta, dq_acc, dk_acc, dv_acc, cu, H, Hkv, D, G,
+    dim3 block(256);
+    dim3 grid(max_tiles, batch, H);
+    size_t shmem_dkv = (size_t)(DPAD * ST * 2 + 64 * ST * 2) * sizeof(__bf16);
+    hipLaunchKernelGGL((fa_bwd_dkv_kernel<DPAD>), grid, block, shmem_dkv, stream,
+                       q, k, v, dout, lse, delta, dk_acc, dv_acc, cu, H, Hkv, D, G,
+                       q_ts, q_gs, k_ts, k_hs, v_ts, v_hs, do_ts, T, scale);
+    int err = dol_last_error();
+    if (err) return err;
+    size_t shmem_dq = (size_t)(DPAD * ST + 64 * ST) * sizeof(__bf16);
+    hipLaunchKernelGGL((fa_bwd_dq_kernel<DPAD>), grid, block, shmem_dq, stream,
+                       q, k, v, dout, lse, delta, dq_acc, cu, H, Hkv, D, G,
                        q_ts, q_gs, k_ts, k_hs, v_ts, v_hs, do_ts, T, scale);
     return dol_last_error();
 }
