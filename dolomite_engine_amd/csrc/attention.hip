@@ -67,7 +67,7 @@ __device__ __forceinline__ float qwave_reduce_sum(float v) {
 // ===========================================================================
 
 template <int DPAD>
-__global__ void __launch_bounds__(256, 3) fa_fwd_kernel(
+__global__ void __launch_bounds__(256) fa_fwd_kernel(
     const __bf16* __restrict__ q, const __bf16* __restrict__ k, const __bf16* __restrict__ v,
     __bf16* __restrict__ o, float* __restrict__ lse,
     const int32_t* __restrict__ cu, int H, int Hkv, int D, int G,
@@ -127,44 +127,24 @@ __global__ void __launch_bounds__(256, 3) fa_fwd_kernel(
     const int kend = min(L, qs + 64);
     const int ntiles = (kend + 63) / 64;
 
-    // T14 staging split: tile kt+1's K/V pieces are loaded to registers
-    // while tile kt computes; the write pass lands after the read barrier.
-    constexpr int PIECES = 64 * DPAD / 8;
-    constexpr int PPT = (PIECES + 255) / 256;  // pieces per thread
-    bf16x8 kstage[PPT], vstage[PPT];
-
-#define FWD_STAGE_LOAD(KS)                                                                              \
-    _Pragma("unroll") for (int pi = 0; pi < PPT; ++pi) {                                                \
-        int pidx = (int)threadIdx.x + pi * 256;                                                         \
-        if (pidx < PIECES) {                                                                            \
-            int key = pidx / (DPAD / 8);                                                                \
-            int d0 = (pidx % (DPAD / 8)) * 8;                                                           \
-            bool kv_valid = ((KS) + key) < kend;                                                        \
-            const __bf16* kp = k + (int64_t)(s0 + (kv_valid ? (KS) + key : 0)) * k_ts + (int64_t)kvh * k_hs + d0; \
-            kstage[pi] = load_bf16x8_guard(kp, d0, D, kv_valid);                                        \
-            const __bf16* vp = v + (int64_t)(s0 + (kv_valid ? (KS) + key : 0)) * v_ts + (int64_t)kvh * v_hs + d0; \
-            vstage[pi] = load_bf16x8_guard(vp, d0, D, kv_valid);                                        \
-        }                                                                                               \
-    }
-
-    FWD_STAGE_LOAD(0)
-
     for (int kt = 0; kt < ntiles; ++kt) {
         const int ks = kt * 64;
-        // write the pre-loaded tile, then issue next tile's loads (they stay
-        // in flight across the barrier and the compute phase)
-#pragma unroll
-        for (int pi = 0; pi < PPT; ++pi) {
-            int pidx = (int)threadIdx.x + pi * 256;
-            if (pidx < PIECES) {
+        // --- cooperative staging: K -> [key][d], V -> transposed [d][key] ---
+        {
+            const int pieces = 64 * DPAD / 8;  // 8-elem pieces
+            for (int pidx = threadIdx.x; pidx < pieces; pidx += 256) {
                 int key = pidx / (DPAD / 8);
                 int d0 = (pidx % (DPAD / 8)) * 8;
-                *(bf16x8*)&Klds[key * SK + d0] = kstage[pi];
+                bool kv_valid = (ks + key) < kend;
+                const __bf16* kp = k + (int64_t)(s0 + (kv_valid ? ks + key : 0)) * k_ts + (int64_t)kvh * k_hs + d0;
+                bf16x8 kk = load_bf16x8_guard(kp, d0, D, kv_valid);
+                *(bf16x8*)&Klds[key * SK + d0] = kk;
+                const __bf16* vp = v + (int64_t)(s0 + (kv_valid ? ks + key : 0)) * v_ts + (int64_t)kvh * v_hs + d0;
+                bf16x8 vv = load_bf16x8_guard(vp, d0, D, kv_valid);
 #pragma unroll
-                for (int e = 0; e < 8; ++e) Vlds[VSWZ(d0 + e, key)] = vstage[pi][e];
+                for (int e = 0; e < 8; ++e) Vlds[VSWZ(d0 + e, key)] = vv[e];
             }
         }
-        if (kt + 1 < ntiles) FWD_STAGE_LOAD(ks + 64)
         __syncthreads();
 
         // --- QK^T: 4 key-blocks of 16, accumulate over KCH chunks ---
@@ -180,9 +160,6 @@ __global__ void __launch_bounds__(256, 3) fa_fwd_kernel(
         }
 
         // --- mask + online softmax (fp32, per C-row) ---
-        // full tiles (every kpos <= every qpos of this wave, no kend clip)
-        // skip the per-element masking entirely — wave-uniform condition
-        const bool full_tile = (ks + 64 <= qs + wave * 16 + 1) && (ks + 64 <= kend) && (qs + wave * 16 + 16 <= L);
         float p_val[4][4];
         float alpha[4];
 #pragma unroll
@@ -193,7 +170,7 @@ __global__ void __launch_bounds__(256, 3) fa_fwd_kernel(
             for (int cb = 0; cb < 4; ++cb) {
                 const int kpos = ks + cb * 16 + lr;
                 float s = sc[cb][r] * scale;
-                if (!full_tile && (kpos > qpos || kpos >= kend || qpos >= L)) s = -INFINITY;
+                if (kpos > qpos || kpos >= kend || qpos >= L) s = -INFINITY;
                 p_val[cb][r] = s;
                 rowmax = fmaxf(rowmax, s);
             }
@@ -256,7 +233,6 @@ __global__ void __launch_bounds__(256, 3) fa_fwd_kernel(
     }
 #undef VSWZ
 #undef VSWZ8
-#undef FWD_STAGE_LOAD
 }
 
 template <int DPAD>
