@@ -116,18 +116,25 @@ class RoPE(nn.Module):
 
     @torch.no_grad()
     def _set_cache(self, seq_len: int, device) -> None:
+        # plain attributes, NOT registered buffers: transformers-5
+        # from_pretrained materializes non-persistent buffers from the meta
+        # device UNINITIALIZED (to_empty), which silently poisons the tables.
+        # Plain attributes are untouched by the module system; device moves
+        # and meta-context construction are handled by the recompute check.
         self.max_seq_len_cached = seq_len
+        if device is not None and torch.device(device).type == "meta":
+            device = "cpu"
         inv_freq = 1.0 / (
             self.base ** (torch.arange(0, self.head_dim, 2, dtype=torch.float32, device=device) / self.head_dim)
         )
         t = torch.arange(seq_len, dtype=torch.float32, device=device)
         freqs = torch.outer(t, inv_freq)
         emb = torch.cat((freqs, freqs), dim=-1)
-        self.register_buffer("cos_cached", emb.cos(), persistent=False)
-        self.register_buffer("sin_cached", emb.sin(), persistent=False)
+        self.cos_cached = emb.cos()
+        self.sin_cached = emb.sin()
 
     def forward(self, seq_len: int, device) -> tuple[torch.Tensor, torch.Tensor]:
-        if seq_len > self.max_seq_len_cached or self.cos_cached.device != device:
+        if seq_len > self.max_seq_len_cached or self.cos_cached.device != torch.device(device):
             self._set_cache(max(seq_len, self.max_seq_len_cached), device)
         return self.cos_cached[:seq_len], self.sin_cached[:seq_len]
 
