@@ -118,9 +118,23 @@ def main(argv=None):
     model_wrapper, engine, lr_scheduler = build_engine(args)
 
     tp = args.training_parameters
-    train_loader = SyntheticPretrainingDataLoader(
-        tp.micro_batch_size, tp.sequence_length, model_wrapper.config.vocab_size, seed=args.random_args.seed
-    )
+    if args.datasets and args.datasets[0].class_name == "MegatronDataset":
+        from .megatron import GPTDataset, MegatronDataLoader, MMapIndexedDataset
+
+        ca = args.datasets[0].class_args
+        indexed = MMapIndexedDataset(ca["data_path"])
+        world = get_world_size()
+        gpt = GPTDataset(
+            indexed,
+            num_samples=tp.num_training_steps * tp.micro_batch_size * tp.gradient_accumulation_steps * world,
+            seq_length=tp.sequence_length,
+            seed=args.random_args.seed,
+        )
+        train_loader = MegatronDataLoader(gpt, tp.micro_batch_size)
+    else:
+        train_loader = SyntheticPretrainingDataLoader(
+            tp.micro_batch_size, tp.sequence_length, model_wrapper.config.vocab_size, seed=args.random_args.seed
+        )
 
     starting_step, metadata = 0, None
     if args.load_args is not None:

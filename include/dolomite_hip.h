@@ -199,6 +199,28 @@ int dolomite_adamw_step(dolomite_stream_t stream,
 int dolomite_scale_inplace(dolomite_stream_t stream, void* buf, int64_t n,
                            float scale, int dtype);
 
+/* ------------------------------------------------------------------------
+ * Host-side dataset index builders (CPU; the reference's only native
+ * component, data/megatron/utils/helpers.cpp).
+ * build_sample_idx: pack epoch-replicated documents into seq_length+1 token
+ *   windows; sample_idx is [(num_samples+1) x 2] rows of (doc_idx position,
+ *   token offset); caller computes num_samples =
+ *   (num_epochs*tokens_per_epoch - 1) / seq_length and allocates.
+ * build_blending_indices: greedy max-error dataset interleave. Buffers are
+ * HOST pointers.
+ * ---------------------------------------------------------------------- */
+int dolomite_build_sample_idx_i32(const int32_t* sizes, const int32_t* doc_idx,
+                                  int32_t seq_length, int32_t num_epochs,
+                                  int64_t tokens_per_epoch, int32_t* sample_idx,
+                                  int64_t num_samples);
+int dolomite_build_sample_idx_i64(const int32_t* sizes, const int32_t* doc_idx,
+                                  int32_t seq_length, int32_t num_epochs,
+                                  int64_t tokens_per_epoch, int64_t* sample_idx,
+                                  int64_t num_samples);
+int dolomite_build_blending_indices(int16_t* dataset_index, int64_t* dataset_sample_index,
+                                    const double* weights, int32_t num_datasets,
+                                    int64_t size);
+
 #ifdef __cplusplus
 }
 #endif
