@@ -62,6 +62,28 @@ __device__ __forceinline__ float qwave_reduce_sum(float v) {
     return v;
 }
 
+
+// XCD-aware block remap (guide T1): the dispatcher places flat block id b on
+// XCD b%8, so the raw grid (tile, seq, head) spreads one (seq, head)'s tiles
+// over all 8 L2s and every tile re-fetches the same Q/dO/K/V slices from HBM
+// (measured: 16 GB/launch vs ~1 GB algorithmic on the backward dkv kernel).
+// Remap so ALL tiles of one (seq, head) stay on ONE XCD and run
+// back-to-back: its ~1.3 MB working set then lives in that XCD's 4 MB L2.
+// Pure permutation — correctness never depends on placement.
+__device__ __forceinline__ void xcd_remap_tile_bh(int& tile, int& b, int& h) {
+    const int ntile = gridDim.x, nb = gridDim.y, nh = gridDim.z;
+    const int64_t nbh = (int64_t)nb * nh;
+    if (nbh % 8 != 0) return;  // identity fallback for tiny grids
+    int64_t raw = blockIdx.x + (int64_t)ntile * (blockIdx.y + (int64_t)nb * blockIdx.z);
+    int xcd = (int)(raw % 8);
+    int64_t idx = raw / 8;             // per-XCD sequence number
+    int64_t pair_local = idx / ntile;  // which (seq, head) pair on this XCD
+    tile = (int)(idx % ntile);
+    int64_t pair = pair_local * 8 + xcd;
+    b = (int)(pair % nb);
+    h = (int)(pair / nb);
+}
+
 // ===========================================================================
 // Forward
 // ===========================================================================
@@ -78,15 +100,15 @@ __global__ void __launch_bounds__(256) fa_fwd_kernel(
     constexpr int SK = DPAD + 8;     // K LDS row stride (elems), +16B pad
     constexpr int SV = 64 + 8;       // V^T / P LDS row stride
 
-    const int b = blockIdx.y;
-    const int h = blockIdx.z;
+    int tile_id = blockIdx.x, b = blockIdx.y, h = blockIdx.z;
+    xcd_remap_tile_bh(tile_id, b, h);
     const int s0 = cu[b];
     const int L = cu[b + 1] - s0;
     // heaviest tiles (largest qs -> most k-tiles) dispatch FIRST: in-order
     // dispatch otherwise schedules the long-pole causal workgroups last
     const int ntile_seq = (L + 63) / 64;
-    if ((int)blockIdx.x >= ntile_seq) return;
-    const int qs = (ntile_seq - 1 - (int)blockIdx.x) * 64;
+    if (tile_id >= ntile_seq) return;
+    const int qs = (ntile_seq - 1 - tile_id) * 64;
 
     const int lane = threadIdx.x & 63;
     const int wave = threadIdx.x >> 6;
@@ -336,12 +358,12 @@ __global__ void __launch_bounds__(256, 2) fa_bwd_dkv_kernel(
     constexpr int DCH = DPAD / 16;
     constexpr int ST = 64 + 8;
 
-    const int b = blockIdx.y;
-    const int h = blockIdx.z;
+    int tile_id = blockIdx.x, b = blockIdx.y, h = blockIdx.z;
+    xcd_remap_tile_bh(tile_id, b, h);
     const int kvh = h / G;
     const int s0 = cu[b];
     const int L = cu[b + 1] - s0;
-    const int ks = (int)blockIdx.x * 64;
+    const int ks = tile_id * 64;
     if (ks >= L) return;
 
     const int lane = threadIdx.x & 63;
@@ -514,15 +536,15 @@ __global__ void __launch_bounds__(256) fa_bwd_dq_kernel(
     constexpr int DCH = DPAD / 16;
     constexpr int ST = 64 + 8;
 
-    const int b = blockIdx.y;
-    const int h = blockIdx.z;
+    int tile_id = blockIdx.x, b = blockIdx.y, h = blockIdx.z;
+    xcd_remap_tile_bh(tile_id, b, h);
     const int kvh = h / G;
     const int s0 = cu[b];
     const int L = cu[b + 1] - s0;
     // heaviest q-tiles (most kv tiles) first
     const int ntile_seq = (L + 63) / 64;
-    if ((int)blockIdx.x >= ntile_seq) return;
-    const int qs = (ntile_seq - 1 - (int)blockIdx.x) * 64;
+    if (tile_id >= ntile_seq) return;
+    const int qs = (ntile_seq - 1 - tile_id) * 64;
 
     const int lane = threadIdx.x & 63;
     const int wave = threadIdx.x >> 6;
