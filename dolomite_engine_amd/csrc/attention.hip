@@ -528,7 +528,7 @@ template <int DPAD>
 __global__ void __launch_bounds__(256) fa_bwd_dq_kernel(
     const __bf16* __restrict__ q, const __bf16* __restrict__ k, const __bf16* __restrict__ v,
     const __bf16* __restrict__ dout, const float* __restrict__ lse, const float* __restrict__ delta,
-    float* __restrict__ dq_acc,
+    __bf16* __restrict__ dqkv_q,
     const int32_t* __restrict__ cu, int H, int Hkv, int D, int G,
     int64_t q_ts, int64_t q_gs, int64_t k_ts, int64_t k_hs, int64_t v_ts, int64_t v_hs,
     int64_t do_ts, int64_t T_total, float scale) {
@@ -659,7 +659,8 @@ __global__ void __launch_bounds__(256) fa_bwd_dq_kernel(
         }
     }
 
-    // single plain store of the accumulated dQ
+    // single bf16 store of the accumulated dQ straight into the packed dqkv
+    // (each (t, h, d) is owned by exactly one q-tile workgroup)
 #pragma unroll
     for (int dc = 0; dc < DCH; ++dc) {
 #pragma unroll
@@ -667,7 +668,7 @@ __global__ void __launch_bounds__(256) fa_bwd_dq_kernel(
             const int qpos = qs + wave * 16 + lg * 4 + r;
             const int d = dc * 16 + lr;
             if (qpos < L && d < D)
-                dq_acc[((int64_t)(s0 + qpos) * H + h) * D + d] = dq[dc][r];
+                dqkv_q[(int64_t)(s0 + qpos) * q_ts + q_hoff + d] = (__bf16)dq[dc][r];
         }
     }
 #undef SWZ
@@ -677,7 +678,7 @@ __global__ void __launch_bounds__(256) fa_bwd_dq_kernel(
 template <int DPAD>
 static int launch_fa_bwd(hipStream_t stream, const __bf16* q, const __bf16* k, const __bf16* v,
                          const __bf16* dout, const float* lse, const float* delta,
-                         float* dq_acc, float* dk_acc, float* dv_acc,
+                         __bf16* dqkv_q, float* dk_acc, float* dv_acc,
                          const int32_t* cu, int batch, int64_t T, int H, int Hkv, int D, int G,
                          int64_t q_ts, int64_t q_gs, int64_t k_ts, int64_t k_hs,
                          int64_t v_ts, int64_t v_hs, int64_t do_ts, int max_tiles, float scale) {
@@ -692,7 +693,7 @@ static int launch_fa_bwd(hipStream_t stream, const __bf16* q, const __bf16* k, c
     if (err) return err;
     size_t shmem_dq = (size_t)(DPAD * ST + 64 * ST) * sizeof(__bf16);
     hipLaunchKernelGGL((fa_bwd_dq_kernel<DPAD>), grid, block, shmem_dq, stream,
-                       q, k, v, dout, lse, delta, dq_acc, cu, H, Hkv, D, G,
+                       q, k, v, dout, lse, delta, dqkv_q, cu, H, Hkv, D, G,
                        q_ts, q_gs, k_ts, k_hs, v_ts, v_hs, do_ts, T, scale);
     return dol_last_error();
 }
@@ -700,7 +701,7 @@ static int launch_fa_bwd(hipStream_t stream, const __bf16* q, const __bf16* k, c
 extern "C" int dolomite_fa_varlen_bwd(dolomite_stream_t stream,
                                       const void* q, const void* k, const void* v,
                                       const void* dout, const float* lse,
-                                      const float* delta, float* dq_acc, float* dk_acc, float* dv_acc,
+                                      const float* delta, void* dqkv_q, float* dk_acc, float* dv_acc,
                                       const int32_t* cu_seqlens, int batch, int max_seqlen, int64_t T,
                                       int H, int Hkv, int D, int G,
                                       int64_t q_tstride, int64_t q_gstride,
@@ -714,7 +715,7 @@ extern "C" int dolomite_fa_varlen_bwd(dolomite_stream_t stream,
     hipStream_t s = (hipStream_t)stream;
 #define CASE(DP)                                                                                        \
     return launch_fa_bwd<DP>(s, (const __bf16*)q, (const __bf16*)k, (const __bf16*)v,                   \
-                             (const __bf16*)dout, lse, delta, dq_acc, dk_acc, dv_acc,                   \
+                             (const __bf16*)dout, lse, delta, (__bf16*)dqkv_q, dk_acc, dv_acc,          \
                              cu_seqlens, batch, T, H, Hkv, D, G, q_tstride, q_gstride, k_tstride,       \
                              k_hstride, v_tstride, v_hstride, do_tstride, max_tiles, scale)
     if (D <= 32) CASE(32);
@@ -730,19 +731,10 @@ extern "C" int dolomite_fa_varlen_bwd(dolomite_stream_t stream,
 
 template <typename T>
 __global__ void __launch_bounds__(256) fa_grad_finalize_kernel(
-    const float* __restrict__ dq_acc, const float* __restrict__ dk_acc, const float* __restrict__ dv_acc,
-    T* __restrict__ dqkv, int64_t total_q, int64_t total_kv,
-    int H, int Hkv, int D, int G, int64_t row_ts, int64_t q_gs, int64_t k_off, int64_t kv_hs, int64_t v_off) {
-    int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-    if (idx < total_q) {
-        int64_t t = idx / ((int64_t)H * D);
-        int rem = (int)(idx % ((int64_t)H * D));
-        int h = rem / D;
-        int d = rem % D;
-        store_from_f32(&dqkv[t * row_ts + (int64_t)(h / G) * q_gs + (int64_t)(h % G) * D + d], dq_acc[idx]);
-        return;
-    }
-    int64_t kidx = idx - total_q;
+    const float* __restrict__ dk_acc, const float* __restrict__ dv_acc,
+    T* __restrict__ dqkv, int64_t total_kv,
+    int Hkv, int D, int64_t row_ts, int64_t k_off, int64_t kv_hs, int64_t v_off) {
+    int64_t kidx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
     bool is_v = kidx >= total_kv;
     if (is_v) kidx -= total_kv;
     if (kidx >= total_kv) return;
@@ -756,23 +748,22 @@ __global__ void __launch_bounds__(256) fa_grad_finalize_kernel(
 }
 
 extern "C" int dolomite_fa_grad_finalize(dolomite_stream_t stream,
-                                         const float* dq_acc, const float* dk_acc, const float* dv_acc,
-                                         void* dqkv, int64_t T, int H, int Hkv, int D, int G,
-                                         int64_t row_tstride, int64_t q_gstride,
+                                         const float* dk_acc, const float* dv_acc,
+                                         void* dqkv, int64_t T, int Hkv, int D,
+                                         int64_t row_tstride,
                                          int64_t k_off, int64_t kv_hstride, int64_t v_off, int dtype) {
-    int64_t total_q = T * (int64_t)H * D;
     int64_t total_kv = T * (int64_t)Hkv * D;
-    int64_t total = total_q + 2 * total_kv;
+    int64_t total = 2 * total_kv;
     if (total == 0) return 0;
     dim3 grid((uint32_t)((total + 255) / 256)), block(256);
     if (dtype == DOLOMITE_BF16)
         hipLaunchKernelGGL((fa_grad_finalize_kernel<uint16_t>), grid, block, 0, (hipStream_t)stream,
-                           dq_acc, dk_acc, dv_acc, (uint16_t*)dqkv, total_q, total_kv,
-                           H, Hkv, D, G, row_tstride, q_gstride, k_off, kv_hstride, v_off);
+                           dk_acc, dv_acc, (uint16_t*)dqkv, total_kv,
+                           Hkv, D, row_tstride, k_off, kv_hstride, v_off);
     else
         hipLaunchKernelGGL((fa_grad_finalize_kernel<float>), grid, block, 0, (hipStream_t)stream,
-                           dq_acc, dk_acc, dv_acc, (float*)dqkv, total_q, total_kv,
-                           H, Hkv, D, G, row_tstride, q_gstride, k_off, kv_hstride, v_off);
+                           dk_acc, dv_acc, (float*)dqkv, total_kv,
+                           Hkv, D, row_tstride, k_off, kv_hstride, v_off);
     return dol_last_error();
 }
 

@@ -133,3 +133,15 @@ struct VecIO<float, 1> {
     static __device__ __forceinline__ void load(const float* p, float* out) { out[0] = *p; }
     static __device__ __forceinline__ void store(float* p, const float* in) { *p = in[0]; }
 };
+
+template <>
+struct VecIO<float, 8> {
+    static __device__ __forceinline__ void load(const float* p, float* out) {
+        VecIO<float, 4>::load(p, out);
+        VecIO<float, 4>::load(p + 4, out + 4);
+    }
+    static __device__ __forceinline__ void store(float* p, const float* in) {
+        VecIO<float, 4>::store(p, in);
+        VecIO<float, 4>::store(p + 4, in + 4);
+    }
+};

@@ -476,6 +476,50 @@ extern "C" int dolomite_reduce_partials(dolomite_stream_t stream,
 // One thread per (token, head, pair). position_embedding/rope.py:104-121.
 // ===========================================================================
 
+// vector variant: each thread rotates 8 consecutive pairs with two 16B
+// loads/stores (requires (D/2) % 8 == 0 — true of every head dim here)
+template <typename T>
+__global__ void __launch_bounds__(256) rope_qkv_vec8_kernel(
+    const T* __restrict__ qkv_in, T* __restrict__ qkv_out,
+    const float* __restrict__ cos_t, const float* __restrict__ sin_t,
+    int64_t total8, int64_t row_len, int H, int Hkv, int D, int G,
+    int64_t q_gstride, int64_t k_off, int64_t kv_hstride, float dir) {
+    int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (idx >= total8) return;
+    int halfD8 = (D >> 1) >> 3;
+    int nheads = H + Hkv;
+    int64_t per_tok = (int64_t)nheads * halfD8;
+    int64_t t = idx / per_tok;
+    int rem = (int)(idx % per_tok);
+    int hh = rem / halfD8;
+    int i0 = (rem % halfD8) * 8;
+
+    int64_t base = t * row_len;
+    if (hh < H)
+        base += (int64_t)(hh / G) * q_gstride + (int64_t)(hh % G) * D;
+    else
+        base += k_off + (int64_t)(hh - H) * kv_hstride;
+
+    float x1[8], x2[8], c1[8], s1[8], c2[8], s2[8], y1[8], y2[8];
+    VecIO<T, 8>::load(qkv_in + base + i0, x1);
+    VecIO<T, 8>::load(qkv_in + base + i0 + (D >> 1), x2);
+    VecIO<float, 4>::load(cos_t + t * D + i0, c1);
+    VecIO<float, 4>::load(cos_t + t * D + i0 + 4, c1 + 4);
+    VecIO<float, 4>::load(sin_t + t * D + i0, s1);
+    VecIO<float, 4>::load(sin_t + t * D + i0 + 4, s1 + 4);
+    VecIO<float, 4>::load(cos_t + t * D + i0 + (D >> 1), c2);
+    VecIO<float, 4>::load(cos_t + t * D + i0 + (D >> 1) + 4, c2 + 4);
+    VecIO<float, 4>::load(sin_t + t * D + i0 + (D >> 1), s2);
+    VecIO<float, 4>::load(sin_t + t * D + i0 + (D >> 1) + 4, s2 + 4);
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+        y1[e] = x1[e] * c1[e] - x2[e] * (s1[e] * dir);
+        y2[e] = x2[e] * c2[e] + x1[e] * (s2[e] * dir);
+    }
+    VecIO<T, 8>::store(qkv_out + base + i0, y1);
+    VecIO<T, 8>::store(qkv_out + base + i0 + (D >> 1), y2);
+}
+
 template <typename T>
 __global__ void __launch_bounds__(256) rope_qkv_kernel(
     const T* __restrict__ qkv_in, T* __restrict__ qkv_out,
@@ -518,10 +562,24 @@ extern "C" int dolomite_rope_qkv(dolomite_stream_t stream,
                                  int dir, int rotate_v_copy, int dtype) {
     (void)rotate_v_copy;
     if (D % 2 != 0) return 9003;
+    float fdir = (dir >= 0) ? 1.f : -1.f;
+    if ((D / 2) % 8 == 0) {
+        int64_t total8 = T_rows * (int64_t)(H + Hkv) * (D / 2 / 8);
+        if (total8 == 0) return 0;
+        dim3 grid((uint32_t)((total8 + 255) / 256)), block(256);
+        if (dtype == DOLOMITE_BF16)
+            hipLaunchKernelGGL((rope_qkv_vec8_kernel<uint16_t>), grid, block, 0, (hipStream_t)stream,
+                               (const uint16_t*)qkv_in, (uint16_t*)qkv_out, cos_t, sin_t,
+                               total8, row_len, H, Hkv, D, G, q_gstride, k_off, kv_hstride, fdir);
+        else
+            hipLaunchKernelGGL((rope_qkv_vec8_kernel<float>), grid, block, 0, (hipStream_t)stream,
+                               (const float*)qkv_in, (float*)qkv_out, cos_t, sin_t,
+                               total8, row_len, H, Hkv, D, G, q_gstride, k_off, kv_hstride, fdir);
+        return dol_last_error();
+    }
     int64_t total = T_rows * (int64_t)(H + Hkv) * (D / 2);
     if (total == 0) return 0;
     dim3 grid((uint32_t)((total + 255) / 256)), block(256);
-    float fdir = (dir >= 0) ? 1.f : -1.f;
     if (dtype == DOLOMITE_BF16)
         hipLaunchKernelGGL((rope_qkv_kernel<uint16_t>), grid, block, 0, (hipStream_t)stream,
                            (const uint16_t*)qkv_in, (uint16_t*)qkv_out, cos_t, sin_t,
@@ -687,6 +745,8 @@ extern "C" int dolomite_ce_bwd(dolomite_stream_t stream,
 // with optional bf16 write-out. HBM-bound; 4 elems per thread, coalesced.
 // ===========================================================================
 
+typedef unsigned short us4_t __attribute__((ext_vector_type(4)));
+
 template <int GRAD_BF16>
 __global__ void __launch_bounds__(256) adamw_kernel(
     float* __restrict__ master, uint16_t* __restrict__ param_out,
@@ -694,6 +754,37 @@ __global__ void __launch_bounds__(256) adamw_kernel(
     int64_t n, float lr, float b1, float b2, float eps, float wd,
     float bc1, float bc2) {
     int64_t i0 = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) * 4;
+    if (i0 >= n) return;
+    if (i0 + 4 <= n) {  // vector fast path (16B loads/stores per stream)
+        f4_t p4 = *(f4_t*)&master[i0];
+        f4_t m4 = *(f4_t*)&m[i0];
+        f4_t v4 = *(f4_t*)&v[i0];
+        f4_t g4;
+        if (GRAD_BF16) {
+            us4_t gb = *(const us4_t*)&((const uint16_t*)grad_v)[i0];
+#pragma unroll
+            for (int k = 0; k < 4; ++k) g4[k] = bf16_to_f32((uint16_t)gb[k]);
+        } else {
+            g4 = *(const f4_t*)&((const float*)grad_v)[i0];
+        }
+        us4_t o4;
+#pragma unroll
+        for (int k = 0; k < 4; ++k) {
+            float p = p4[k] * (1.f - lr * wd);
+            float mi = m4[k] * b1 + (1.f - b1) * g4[k];
+            float vi = v4[k] * b2 + (1.f - b2) * g4[k] * g4[k];
+            m4[k] = mi;
+            v4[k] = vi;
+            p -= (lr / bc1) * mi / (sqrtf(vi / bc2) + eps);
+            p4[k] = p;
+            o4[k] = f32_to_bf16(p);
+        }
+        *(f4_t*)&master[i0] = p4;
+        *(f4_t*)&m[i0] = m4;
+        *(f4_t*)&v[i0] = v4;
+        if (param_out) *(us4_t*)&param_out[i0] = o4;
+        return;
+    }
 #pragma unroll
     for (int k = 0; k < 4; ++k) {
         int64_t i = i0 + k;

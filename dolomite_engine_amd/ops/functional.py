@@ -415,7 +415,6 @@ class VarlenAttention(torch.autograd.Function):
                 ),
                 "fa_bwd_preprocess",
             )
-            dq_acc = torch.zeros(T, lo.H, lo.D, dtype=torch.float32, device=qkv.device)
             dk_acc = torch.zeros(T, lo.Hkv, lo.D, dtype=torch.float32, device=qkv.device)
             dv_acc = torch.zeros(T, lo.Hkv, lo.D, dtype=torch.float32, device=qkv.device)
             dqkv = torch.empty_like(qkv)
@@ -424,7 +423,7 @@ class VarlenAttention(torch.autograd.Function):
                 hip.lib().dolomite_fa_varlen_bwd(
                     hip.stream(),
                     hip.ptr(qkv), hip.ptr(qkv, lo.k_off), hip.ptr(qkv, lo.v_off),
-                    hip.ptr(dout), hip.ptr(lse), hip.ptr(delta), hip.ptr(dq_acc),
+                    hip.ptr(dout), hip.ptr(lse), hip.ptr(delta), hip.ptr(dqkv),
                     hip.ptr(dk_acc), hip.ptr(dv_acc),
                     hip.ptr(cu), batch, ctx.max_seqlen, T, lo.H, lo.Hkv, lo.D, lo.G,
                     lo.row_len, lo.q_gstride, lo.row_len, lo.kv_hstride,
@@ -434,8 +433,8 @@ class VarlenAttention(torch.autograd.Function):
             )
             hip.check(
                 hip.lib().dolomite_fa_grad_finalize(
-                    hip.stream(), hip.ptr(dq_acc), hip.ptr(dk_acc), hip.ptr(dv_acc), hip.ptr(dqkv),
-                    T, lo.H, lo.Hkv, lo.D, lo.G, lo.row_len, lo.q_gstride,
+                    hip.stream(), hip.ptr(dk_acc), hip.ptr(dv_acc), hip.ptr(dqkv),
+                    T, lo.Hkv, lo.D, lo.row_len,
                     lo.k_off, lo.kv_hstride, lo.v_off, hip.dt(qkv),
                 ),
                 "fa_grad_finalize",

@@ -126,10 +126,11 @@ int dolomite_fa_varlen_fwd(dolomite_stream_t stream,
  *  1. preprocess: delta[h,t] = rowsum(dO[t,h,:]*O[t,h,:]) fp32.
  *  2. main: grid over (kv-tile, seq, kv-head); recomputes P from q,k,lse;
  *     dk/dv accumulated in registers over the q-tile loop, then added to the
- *     pointers; one workgroup per (kv-tile, seq, Q-HEAD) so MQA/GQA fill the
- *     chip; dq/dk/dv accumulate into zero-initialized fp32 buffers
- *     ((T,H,D) and (T,Hkv,D)) with device-scope atomics.
- *  3. grad_finalize: casts the fp32 accumulators into the packed dqkv.
+ *     pointers; dkv: one workgroup per (kv-tile, seq, Q-HEAD); dk/dv join
+ *     zero-initialized (T,Hkv,D) fp32 buffers with device-scope atomics;
+ *     dq: one workgroup per (q-tile, seq, q-head), register-accumulated and
+ *     stored bf16 directly into the packed dqkv q-slots.
+ *  3. grad_finalize: casts the dk/dv fp32 accumulators into the packed dqkv.
  */
 int dolomite_fa_bwd_preprocess(dolomite_stream_t stream,
                                const void* o, const void* dout, float* delta,
@@ -138,7 +139,7 @@ int dolomite_fa_bwd_preprocess(dolomite_stream_t stream,
 int dolomite_fa_varlen_bwd(dolomite_stream_t stream,
                            const void* q, const void* k, const void* v,
                            const void* dout, const float* lse,
-                           const float* delta, float* dq_acc, float* dk_acc, float* dv_acc,
+                           const float* delta, void* dqkv_q, float* dk_acc, float* dv_acc,
                            const int32_t* cu_seqlens, int batch, int max_seqlen, int64_t T,
                            int H, int Hkv, int D, int G,
                            int64_t q_tstride, int64_t q_gstride,
@@ -147,9 +148,9 @@ int dolomite_fa_varlen_bwd(dolomite_stream_t stream,
                            int64_t do_tstride,
                            float scale, int dtype);
 int dolomite_fa_grad_finalize(dolomite_stream_t stream,
-                              const float* dq_acc, const float* dk_acc, const float* dv_acc,
-                              void* dqkv, int64_t T, int H, int Hkv, int D, int G,
-                              int64_t row_tstride, int64_t q_gstride,
+                              const float* dk_acc, const float* dv_acc,
+                              void* dqkv, int64_t T, int Hkv, int D,
+                              int64_t row_tstride,
                               int64_t k_off, int64_t kv_hstride, int64_t v_off, int dtype);
 
 /* MFMA fragment-layout self-test: computes C = A(16x32) @ B(32x16) with the
