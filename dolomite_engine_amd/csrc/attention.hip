@@ -371,11 +371,14 @@ __global__ void __launch_bounds__(256, 2) fa_bwd_dkv_kernel(
     const int lr = lane & 15;
     const int lg = lane >> 4;
 
+    constexpr int SQ = DPAD + 8;               // row-major image stride
     extern __shared__ char smem_raw[];
     __bf16* QTl = (__bf16*)smem_raw;           // [DPAD][ST]   (Q^T, swizzled)
     __bf16* dOTl = QTl + DPAD * ST;            // [DPAD][ST]   (dO^T, swizzled)
     __bf16* dSTl = dOTl + DPAD * ST;           // [64 key][ST] (dS^T, [key][q])
     __bf16* PTl = dSTl + 64 * ST;              // [64 key][ST] (P^T, [key][q])
+    __bf16* Qlds = PTl + 64 * ST;              // [64 q][SQ]   (row-major)
+    __bf16* dOl = Qlds + 64 * SQ;              // [64 q][SQ]   (row-major)
 
 #define SWZ(row, col) ((row) * ST + (((((col) >> 3) ^ (((row) >> 3) & 7)) << 3) | ((col) & 7)))
 #define SWZ8(row, col0) ((row) * ST + (((((col0) >> 3) ^ (((row) >> 3) & 7)) << 3)))
@@ -417,56 +420,32 @@ __global__ void __launch_bounds__(256, 2) fa_bwd_dkv_kernel(
                 bool valid = (qs + qq) < L;
                 const __bf16* qp = q + (int64_t)(s0 + (valid ? qs + qq : 0)) * q_ts + q_hoff + d0;
                 bf16x8 qq8 = load_bf16x8_guard(qp, d0, D, valid);
+                *(bf16x8*)&Qlds[qq * SQ + d0] = qq8;
 #pragma unroll
                 for (int e = 0; e < 8; ++e) QTl[SWZ(d0 + e, qq)] = qq8[e];
                 const __bf16* dp = dout + (int64_t)(s0 + (valid ? qs + qq : 0)) * do_ts + do_hoff + d0;
                 bf16x8 dd8 = load_bf16x8_guard(dp, d0, D, valid);
+                *(bf16x8*)&dOl[qq * SQ + d0] = dd8;
 #pragma unroll
                 for (int e = 0; e < 8; ++e) dOTl[SWZ(d0 + e, qq)] = dd8[e];
             }
         }
         __syncthreads();
 
-        // S^T = K*Q^T, dP^T = V*dO^T (B-frags straight from global, dbuf'd)
+        // S^T = K*Q^T, dP^T = V*dO^T; B-frags (identical across the 4
+        // waves) come from the row-major LDS images: Q[q=lr][d0..d0+8)
         f32x4 st[4], dpt[4];
-        bf16x8 qcur[KCH], dcur[KCH], qnxt[KCH], dnxt[KCH];
-        {
-            const int qrow = qs + lr;
-            const bool qok = qrow < L;
-            const __bf16* qrp = q + (int64_t)(s0 + (qok ? qrow : 0)) * q_ts + q_hoff;
-            const __bf16* drp = dout + (int64_t)(s0 + (qok ? qrow : 0)) * do_ts + do_hoff;
-#pragma unroll
-            for (int kc = 0; kc < KCH; ++kc) {
-                int d0 = kc * 32 + lg * 8;
-                qcur[kc] = load_bf16x8_guard(qrp + d0, d0, D, qok);
-                dcur[kc] = load_bf16x8_guard(drp + d0, d0, D, qok);
-            }
-        }
 #pragma unroll
         for (int cb = 0; cb < 4; ++cb) {
-            if (cb < 3) {
-                const int qrow = qs + (cb + 1) * 16 + lr;
-                const bool qok = qrow < L;
-                const __bf16* qrp = q + (int64_t)(s0 + (qok ? qrow : 0)) * q_ts + q_hoff;
-                const __bf16* drp = dout + (int64_t)(s0 + (qok ? qrow : 0)) * do_ts + do_hoff;
-#pragma unroll
-                for (int kc = 0; kc < KCH; ++kc) {
-                    int d0 = kc * 32 + lg * 8;
-                    qnxt[kc] = load_bf16x8_guard(qrp + d0, d0, D, qok);
-                    dnxt[kc] = load_bf16x8_guard(drp + d0, d0, D, qok);
-                }
-            }
             st[cb] = {0.f, 0.f, 0.f, 0.f};
             dpt[cb] = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
             for (int kc = 0; kc < KCH; ++kc) {
-                st[cb] = MFMA16(kfr[kc], qcur[kc], st[cb]);
-                dpt[cb] = MFMA16(vfr[kc], dcur[kc], dpt[cb]);
-            }
-#pragma unroll
-            for (int kc = 0; kc < KCH; ++kc) {
-                qcur[kc] = qnxt[kc];
-                dcur[kc] = dnxt[kc];
+                int d0 = kc * 32 + lg * 8;
+                bf16x8 qb = *(const bf16x8*)&Qlds[(cb * 16 + lr) * SQ + d0];
+                st[cb] = MFMA16(kfr[kc], qb, st[cb]);
+                bf16x8 db = *(const bf16x8*)&dOl[(cb * 16 + lr) * SQ + d0];
+                dpt[cb] = MFMA16(vfr[kc], db, dpt[cb]);
             }
         }
 
@@ -551,10 +530,13 @@ __global__ void __launch_bounds__(256) fa_bwd_dq_kernel(
     const int lr = lane & 15;
     const int lg = lane >> 4;
 
+    constexpr int SQ = DPAD + 8;
     extern __shared__ char smem_raw[];
     __bf16* KTl = (__bf16*)smem_raw;           // [DPAD][ST] (K^T, swizzled)
     __bf16* dSl = KTl + DPAD * ST;             // [4 waves][16 q][ST] (dS strips)
     __bf16* dSw = dSl + wave * 16 * ST;
+    __bf16* Klds = dSl + 4 * 16 * ST;          // [64 key][SQ] (row-major)
+    __bf16* Vlds = Klds + 64 * SQ;             // [64 key][SQ] (row-major)
 
 #define SWZ(row, col) ((row) * ST + (((((col) >> 3) ^ (((row) >> 3) & 7)) << 3) | ((col) & 7)))
 #define SWZ8(row, col0) ((row) * ST + (((((col0) >> 3) ^ (((row) >> 3) & 7)) << 3)))
@@ -596,7 +578,7 @@ __global__ void __launch_bounds__(256) fa_bwd_dq_kernel(
     for (int kt = 0; kt < nkt; ++kt) {
         const int ks = kt * 64;
         __syncthreads();  // previous tile's K^T reads done
-        // stage K^T (swizzled scatter), cooperative
+        // stage K^T (swizzled) + row-major K/V images, cooperative
         {
             const int pieces = 64 * DPAD / 8;
             for (int pidx = threadIdx.x; pidx < pieces; pidx += 256) {
@@ -605,28 +587,27 @@ __global__ void __launch_bounds__(256) fa_bwd_dq_kernel(
                 bool valid = (ks + key) < kend_total;
                 const __bf16* kp = k + (int64_t)(s0 + (valid ? ks + key : 0)) * k_ts + (int64_t)kvh * k_hs + d0;
                 bf16x8 kk = load_bf16x8_guard(kp, d0, D, valid);
+                *(bf16x8*)&Klds[key * SQ + d0] = kk;
 #pragma unroll
                 for (int e = 0; e < 8; ++e) KTl[SWZ(d0 + e, key)] = kk[e];
+                const __bf16* vp = v + (int64_t)(s0 + (valid ? ks + key : 0)) * v_ts + (int64_t)kvh * v_hs + d0;
+                *(bf16x8*)&Vlds[key * SQ + d0] = load_bf16x8_guard(vp, d0, D, valid);
             }
         }
         __syncthreads();
 
-        // S = Q*K^T and dP = dO*V^T; B-frags are K/V rows read from global
+        // S = Q*K^T and dP = dO*V^T; B-frags are K/V rows from the LDS images
         f32x4 sc[4], dp[4];
 #pragma unroll
         for (int cb = 0; cb < 4; ++cb) {
             sc[cb] = {0.f, 0.f, 0.f, 0.f};
             dp[cb] = {0.f, 0.f, 0.f, 0.f};
-            const int key = ks + cb * 16 + lr;
-            const bool kok = key < kend_total;
-            const __bf16* krp = k + (int64_t)(s0 + (kok ? key : 0)) * k_ts + (int64_t)kvh * k_hs;
-            const __bf16* vrp = v + (int64_t)(s0 + (kok ? key : 0)) * v_ts + (int64_t)kvh * v_hs;
 #pragma unroll
             for (int kc = 0; kc < KCH; ++kc) {
                 int d0 = kc * 32 + lg * 8;
-                bf16x8 kb = load_bf16x8_guard(krp + d0, d0, D, kok);
+                bf16x8 kb = *(const bf16x8*)&Klds[(cb * 16 + lr) * SQ + d0];
                 sc[cb] = MFMA16(qfr[kc], kb, sc[cb]);
-                bf16x8 vb = load_bf16x8_guard(vrp + d0, d0, D, kok);
+                bf16x8 vb = *(const bf16x8*)&Vlds[(cb * 16 + lr) * SQ + d0];
                 dp[cb] = MFMA16(dfr[kc], vb, dp[cb]);
             }
         }
@@ -685,13 +666,14 @@ static int launch_fa_bwd(hipStream_t stream, const __bf16* q, const __bf16* k, c
     constexpr int ST = 64 + 8;
     dim3 block(256);
     dim3 grid(max_tiles, batch, H);
-    size_t shmem_dkv = (size_t)(DPAD * ST * 2 + 64 * ST * 2) * sizeof(__bf16);
+    constexpr int SQ = DPAD + 8;
+    size_t shmem_dkv = (size_t)(DPAD * ST * 2 + 64 * ST * 2 + 64 * SQ * 2) * sizeof(__bf16);
     hipLaunchKernelGGL((fa_bwd_dkv_kernel<DPAD>), grid, block, shmem_dkv, stream,
                        q, k, v, dout, lse, delta, dk_acc, dv_acc, cu, H, Hkv, D, G,
                        q_ts, q_gs, k_ts, k_hs, v_ts, v_hs, do_ts, T, scale);
     int err = dol_last_error();
     if (err) return err;
-    size_t shmem_dq = (size_t)(DPAD * ST + 64 * ST) * sizeof(__bf16);
+    size_t shmem_dq = (size_t)(DPAD * ST + 64 * ST + 64 * SQ * 2) * sizeof(__bf16);
     hipLaunchKernelGGL((fa_bwd_dq_kernel<DPAD>), grid, block, shmem_dq, stream,
                        q, k, v, dout, lse, delta, dqkv_q, cu, H, Hkv, D, G,
                        q_ts, q_gs, k_ts, k_hs, v_ts, v_hs, do_ts, T, scale);
