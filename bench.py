@@ -217,12 +217,24 @@ def main():
     fa_fwd_flops = 2.0 * H * D * B * S * S  # QK^T + PV, causal half x2 matmuls, per layer-launch
     fa_bwd_flops = 2.5 * fa_fwd_flops  # 5 contractions vs 2 in fwd
     flops_per_launch = {"fa_varlen_fwd": fa_fwd_flops, "fa_varlen_bwd": fa_bwd_flops}
+    # per-launch HBM traffic measured in a separate rocprofv3 --pmc pass
+    # (profiles/pmc_traffic.json, corrections documented there); null when
+    # no calibration exists for the dominant kernel
+    traffic_cal = {}
+    try:
+        import pathlib
+
+        cal = json.load(open(pathlib.Path(__file__).parent / "profiles" / "pmc_traffic.json"))
+        traffic_cal = cal.get("kernels", {})
+    except Exception:
+        pass
     roofline = None
     if profile:
         dom = max(profile, key=lambda k: profile[k]["total_ms"])
         p = profile[dom]
         if dom in flops_per_launch:
             achieved = flops_per_launch[dom] / (p["avg_ms"] / 1e3)
+            tr = traffic_cal.get("fa_bwd_dkv" if dom == "fa_varlen_bwd" else dom)
             roofline = {
                 "kernel": dom,
                 "bound": "mfma",
@@ -230,7 +242,7 @@ def main():
                 "peak": 2500.0,
                 "unit": "TFLOP/s",
                 "frac": achieved / 1e12 / 2500.0,
-                "traffic": None,
+                "traffic": tr["hbm_read_bytes_per_launch"] if tr else None,
                 "avg_launch_ms": p["avg_ms"],
                 "launches": p["count"],
             }
