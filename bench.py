@@ -45,7 +45,7 @@ MODEL_3B = dict(
 )
 
 SEQ_LEN = 4096
-MICRO_BATCH = 8  # tokens per rank per step = 8 * 4096 = 32768
+MICRO_BATCH = 16  # tokens per rank per step = 16 * 4096 = 65536 (fills HBM better; +5% vs B=8)
 
 
 def build(args, device):
