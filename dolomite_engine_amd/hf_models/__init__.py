@@ -2,6 +2,7 @@
 
 from .config import GPTDolomiteConfig
 from .modeling import (
+    apply_gradient_checkpointing,
     GPTDolomiteBlock,
     GPTDolomiteForCausalLM,
     GPTDolomiteModel,

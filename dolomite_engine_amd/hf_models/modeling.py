@@ -521,7 +521,13 @@ class GPTDolomiteModel(GPTDolomitePreTrainedModel):
 
         delta, residual = hs, None
         for block in self.h:
-            delta, residual = block.forward_padding_free(delta, residual, rope_cos_sin, cu_seqlens, max_seqlen)
+            if getattr(block, "_gradient_checkpointing", False) and self.training:
+                delta, residual = torch.utils.checkpoint.checkpoint(
+                    block.forward_padding_free, delta, residual, rope_cos_sin, cu_seqlens, max_seqlen,
+                    use_reentrant=False,
+                )
+            else:
+                delta, residual = block.forward_padding_free(delta, residual, rope_cos_sin, cu_seqlens, max_seqlen)
         hidden_states, _ = self.ln_f(delta, residual)
         return hidden_states
 
@@ -559,7 +565,13 @@ class GPTDolomiteModel(GPTDolomitePreTrainedModel):
             attention_bias = bias.unsqueeze(1)
 
         for block in self.h:
-            hs = block.forward_dense(hs, attention_bias, rope_cos_sin, self.attention_implementation)
+            if getattr(block, "_gradient_checkpointing", False) and self.training:
+                hs = torch.utils.checkpoint.checkpoint(
+                    block.forward_dense, hs, attention_bias, rope_cos_sin, self.attention_implementation,
+                    use_reentrant=False,
+                )
+            else:
+                hs = block.forward_dense(hs, attention_bias, rope_cos_sin, self.attention_implementation)
         hs, _ = self.ln_f(hs)
         return hs
 
@@ -673,3 +685,16 @@ class GPTDolomiteForCausalLM(GPTDolomitePreTrainedModel):
         if self.upcast_logits_for_loss:
             shift_logits = shift_logits.float()
         return fused_cross_entropy(shift_logits.reshape(-1, shift_logits.size(-1)), shift_labels.reshape(-1))
+
+
+def apply_gradient_checkpointing(model, gradient_checkpointing_method: str = "block", checkpoint_every: int = 1, **_):
+    """Per-block activation checkpointing (reference
+    gradient_checkpointing/block.py:13-34: every `checkpoint_every`-th
+    GPTDolomiteBlock recomputes its forward in backward)."""
+    assert gradient_checkpointing_method == "block", gradient_checkpointing_method
+    idx = 0
+    for module in model.modules():
+        if isinstance(module, GPTDolomiteBlock):
+            if idx % checkpoint_every == 0:
+                module._gradient_checkpointing = True
+            idx += 1

@@ -34,6 +34,14 @@ def build_engine(args: TrainingArgs):
         ),
         use_padding_free_transformer=args.model_args.use_padding_free_transformer,
     )
+    if args.distributed_args.gradient_checkpointing_method is not None:
+        from .hf_models import apply_gradient_checkpointing
+
+        apply_gradient_checkpointing(
+            model_wrapper.model,
+            args.distributed_args.gradient_checkpointing_method,
+            **args.distributed_args.gradient_checkpointing_args,
+        )
     if torch.cuda.is_available():
         model_wrapper.model.cuda()
 

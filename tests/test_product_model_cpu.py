@@ -136,3 +136,34 @@ def test_label_ignore_positions_respected(golden_dir):
     shift_labels[torch.tensor([3]) - 0] = -100  # boundary: position cu[1]-1 = 3
     ref = F.cross_entropy(shift_logits, shift_labels)
     torch.testing.assert_close(out.loss, ref, rtol=1e-6, atol=1e-6)
+
+
+def test_gradient_checkpointing_matches_plain(golden_dir):
+    """Block activation checkpointing (reference gradient_checkpointing/
+    block.py) must not change gradients — padding-free path, every block."""
+    from dolomite_engine_amd.hf_models import apply_gradient_checkpointing
+
+    fx = _load(golden_dir, "model_mqa_rope_rmsnorm_gelu.pt")
+
+    def run(checkpointed):
+        model = build_model(fx, "flash_attention_2", padding_free=True)
+        if checkpointed:
+            apply_gradient_checkpointing(model, "block", checkpoint_every=1)
+        model.train()
+        B, S = fx["input_ids"].shape
+        out = model(
+            input_ids=fx["input_ids"].reshape(-1),
+            position_ids=torch.arange(S).repeat(B),
+            cu_seqlens=torch.arange(0, B * S + 1, S, dtype=torch.int32),
+            max_seqlen=S,
+            labels=fx["labels"].reshape(-1),
+        )
+        out.loss.backward()
+        return out.loss.detach(), {k: p.grad.clone() for k, p in model.named_parameters() if p.grad is not None}
+
+    loss_a, grads_a = run(False)
+    loss_b, grads_b = run(True)
+    torch.testing.assert_close(loss_a, loss_b, rtol=0, atol=0)
+    assert grads_a.keys() == grads_b.keys()
+    for k in grads_a:
+        torch.testing.assert_close(grads_a[k], grads_b[k], rtol=1e-6, atol=1e-7, msg=lambda m: f"{k}: {m}")
