@@ -77,9 +77,23 @@ def train(args: TrainingArgs, model_wrapper, engine, lr_scheduler, train_loader,
     train_iter = iter(train_loader)
     model_wrapper.train()
 
+    # torch profiler hook (reference train_utils.py:182-194: wait 5 / warmup
+    # 5 / active 1, rank 0, tensorboard trace)
+    profiler = None
+    if args.logging_args.torch_profiler_trace_path is not None and get_rank() == 0:
+        profiler = torch.profiler.profile(
+            schedule=torch.profiler.schedule(wait=5, warmup=5, active=1, repeat=1),
+            on_trace_ready=torch.profiler.tensorboard_trace_handler(args.logging_args.torch_profiler_trace_path),
+            record_shapes=True,
+            profile_memory=True,
+        )
+        profiler.__enter__()
+
     for global_step in range(starting_step + 1, tp.num_training_steps + 1):
         t0 = time.perf_counter()
         loss, grad_norm = train_step(model_wrapper, engine, lr_scheduler, train_iter, ga, tp.gradient_clipping)
+        if profiler is not None:
+            profiler.step()
         if torch.cuda.is_available():
             torch.cuda.synchronize()
         dt = time.perf_counter() - t0
@@ -103,6 +117,9 @@ def train(args: TrainingArgs, model_wrapper, engine, lr_scheduler, train_loader,
                 args_dict=args.to_dict(),
                 save_optimizer=args.save_args.save_optimizer,
             )
+
+    if profiler is not None:
+        profiler.__exit__(None, None, None)
 
     if args.save_args is not None:
         save_checkpoint(
