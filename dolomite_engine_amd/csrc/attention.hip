@@ -89,7 +89,7 @@ __device__ __forceinline__ void xcd_remap_tile_bh(int& tile, int& b, int& h) {
 // ===========================================================================
 
 template <int DPAD>
-__global__ void __launch_bounds__(256) fa_fwd_kernel(
+__global__ void __launch_bounds__(512) fa_fwd_kernel(
     const __bf16* __restrict__ q, const __bf16* __restrict__ k, const __bf16* __restrict__ v,
     __bf16* __restrict__ o, float* __restrict__ lse,
     const int32_t* __restrict__ cu, int H, int Hkv, int D, int G,
@@ -106,9 +106,12 @@ __global__ void __launch_bounds__(256) fa_fwd_kernel(
     const int L = cu[b + 1] - s0;
     // heaviest tiles (largest qs -> most k-tiles) dispatch FIRST: in-order
     // dispatch otherwise schedules the long-pole causal workgroups last
-    const int ntile_seq = (L + 63) / 64;
+    // 8 waves x 16 rows = 128 q rows per workgroup: K/V staging and
+    // barriers amortize over twice the compute at the same per-wave
+    // register budget
+    const int ntile_seq = (L + 127) / 128;
     if (tile_id >= ntile_seq) return;
-    const int qs = (ntile_seq - 1 - tile_id) * 64;
+    const int qs = (ntile_seq - 1 - tile_id) * 128;
 
     const int lane = threadIdx.x & 63;
     const int wave = threadIdx.x >> 6;
@@ -118,7 +121,7 @@ __global__ void __launch_bounds__(256) fa_fwd_kernel(
     extern __shared__ char smem_raw[];
     __bf16* Klds = (__bf16*)smem_raw;              // [64][SK]
     __bf16* Vlds = Klds + 64 * SK;                 // [DPAD][SV] (transposed, swizzled)
-    __bf16* Plds = Vlds + DPAD * SV;               // [4 waves][16][SV]
+    __bf16* Plds = Vlds + DPAD * SV;               // [8 waves][16][SV]
     __bf16* Pw = Plds + wave * 16 * SV;
 
     // XOR block swizzle for the transposed V image (see fa_bwd_kernel)
@@ -146,7 +149,7 @@ __global__ void __launch_bounds__(256) fa_fwd_kernel(
 #pragma unroll
     for (int dc = 0; dc < DCH; ++dc) o_acc[dc] = {0.f, 0.f, 0.f, 0.f};
 
-    const int kend = min(L, qs + 64);
+    const int kend = min(L, qs + 128);
     const int ntiles = (kend + 63) / 64;
 
     for (int kt = 0; kt < ntiles; ++kt) {
@@ -154,7 +157,7 @@ __global__ void __launch_bounds__(256) fa_fwd_kernel(
         // --- cooperative staging: K -> [key][d], V -> transposed [d][key] ---
         {
             const int pieces = 64 * DPAD / 8;  // 8-elem pieces
-            for (int pidx = threadIdx.x; pidx < pieces; pidx += 256) {
+            for (int pidx = threadIdx.x; pidx < pieces; pidx += 512) {
                 int key = pidx / (DPAD / 8);
                 int d0 = (pidx % (DPAD / 8)) * 8;
                 bool kv_valid = (ks + key) < kend;
@@ -263,8 +266,8 @@ static int launch_fa_fwd(hipStream_t stream, const __bf16* q, const __bf16* k, c
                          int H, int Hkv, int D, int G,
                          int64_t q_ts, int64_t q_gs, int64_t k_ts, int64_t k_hs,
                          int64_t v_ts, int64_t v_hs, int max_tiles, float scale) {
-    dim3 grid(max_tiles, batch, H), block(256);
-    size_t shmem = (64 * (DPAD + 8) + DPAD * 72 + 4 * 16 * 72) * sizeof(__bf16);
+    dim3 grid(max_tiles, batch, H), block(512);
+    size_t shmem = (64 * (DPAD + 8) + DPAD * 72 + 8 * 16 * 72) * sizeof(__bf16);
     hipLaunchKernelGGL((fa_fwd_kernel<DPAD>), grid, block, shmem, stream,
                        q, k, v, o, lse, cu, H, Hkv, D, G,
                        q_ts, q_gs, k_ts, k_hs, v_ts, v_hs, (int64_t)H * D, T, scale);
@@ -284,7 +287,7 @@ extern "C" int dolomite_fa_varlen_fwd(dolomite_stream_t stream,
     if (D > 128) return 9011;
     // grid.x = tiles of the longest sequence; shorter sequences' surplus
     // workgroups exit on the cu_seqlens check.
-    int max_tiles = (max_seqlen + 63) / 64;
+    int max_tiles = (max_seqlen + 127) / 128;
     hipStream_t s = (hipStream_t)stream;
 #define CASE(DP)                                                                                     \
     return launch_fa_fwd<DP>(s, (const __bf16*)q, (const __bf16*)k, (const __bf16*)v, (__bf16*)o,    \
@@ -347,7 +350,7 @@ extern "C" int dolomite_fa_bwd_preprocess(dolomite_stream_t stream,
 //     once. K^T is the only LDS image (swizzled), staged per kv tile.
 
 template <int DPAD>
-__global__ void __launch_bounds__(256, 2) fa_bwd_dkv_kernel(
+__global__ void __launch_bounds__(512, 2) fa_bwd_dkv_kernel(
     const __bf16* __restrict__ q, const __bf16* __restrict__ k, const __bf16* __restrict__ v,
     const __bf16* __restrict__ dout, const float* __restrict__ lse, const float* __restrict__ delta,
     float* __restrict__ dk_acc, float* __restrict__ dv_acc,
@@ -363,7 +366,8 @@ __global__ void __launch_bounds__(256, 2) fa_bwd_dkv_kernel(
     const int kvh = h / G;
     const int s0 = cu[b];
     const int L = cu[b + 1] - s0;
-    const int ks = tile_id * 64;
+    // 8 waves x 16 keys = 128-key strip per workgroup
+    const int ks = tile_id * 128;
     if (ks >= L) return;
 
     const int lane = threadIdx.x & 63;
@@ -373,17 +377,17 @@ __global__ void __launch_bounds__(256, 2) fa_bwd_dkv_kernel(
 
     constexpr int SQ = DPAD + 8;               // row-major image stride
     extern __shared__ char smem_raw[];
-    __bf16* QTl = (__bf16*)smem_raw;           // [DPAD][ST]   (Q^T, swizzled)
-    __bf16* dOTl = QTl + DPAD * ST;            // [DPAD][ST]   (dO^T, swizzled)
-    __bf16* dSTl = dOTl + DPAD * ST;           // [64 key][ST] (dS^T, [key][q])
-    __bf16* PTl = dSTl + 64 * ST;              // [64 key][ST] (P^T, [key][q])
-    __bf16* Qlds = PTl + 64 * ST;              // [64 q][SQ]   (row-major)
-    __bf16* dOl = Qlds + 64 * SQ;              // [64 q][SQ]   (row-major)
+    __bf16* QTl = (__bf16*)smem_raw;           // [DPAD][ST]    (Q^T, swizzled)
+    __bf16* dOTl = QTl + DPAD * ST;            // [DPAD][ST]    (dO^T, swizzled)
+    __bf16* dSTl = dOTl + DPAD * ST;           // [128 key][ST] (dS^T, [key][q])
+    __bf16* PTl = dSTl + 128 * ST;             // [128 key][ST] (P^T, [key][q])
+    __bf16* Qlds = PTl + 128 * ST;             // [64 q][SQ]    (row-major)
+    __bf16* dOl = Qlds + 64 * SQ;              // [64 q][SQ]    (row-major)
 
 #define SWZ(row, col) ((row) * ST + (((((col) >> 3) ^ (((row) >> 3) & 7)) << 3) | ((col) & 7)))
 #define SWZ8(row, col0) ((row) * ST + (((((col0) >> 3) ^ (((row) >> 3) & 7)) << 3)))
 
-    const int kend = min(L, ks + 64);
+    const int kend = min(L, ks + 128);
 
     const int krow = ks + wave * 16 + lr;
     const bool kvalid = krow < kend;
@@ -414,7 +418,7 @@ __global__ void __launch_bounds__(256, 2) fa_bwd_dkv_kernel(
         __syncthreads();  // previous iteration's image reads done
         {
             const int pieces = 64 * DPAD / 8;
-            for (int pidx = threadIdx.x; pidx < pieces; pidx += 256) {
+            for (int pidx = threadIdx.x; pidx < pieces; pidx += 512) {
                 int qq = pidx / (DPAD / 8);
                 int d0 = (pidx % (DPAD / 8)) * 8;
                 bool valid = (qs + qq) < L;
@@ -504,7 +508,7 @@ __global__ void __launch_bounds__(256, 2) fa_bwd_dkv_kernel(
 }
 
 template <int DPAD>
-__global__ void __launch_bounds__(256) fa_bwd_dq_kernel(
+__global__ void __launch_bounds__(512) fa_bwd_dq_kernel(
     const __bf16* __restrict__ q, const __bf16* __restrict__ k, const __bf16* __restrict__ v,
     const __bf16* __restrict__ dout, const float* __restrict__ lse, const float* __restrict__ delta,
     __bf16* __restrict__ dqkv_q,
@@ -520,10 +524,10 @@ __global__ void __launch_bounds__(256) fa_bwd_dq_kernel(
     const int kvh = h / G;
     const int s0 = cu[b];
     const int L = cu[b + 1] - s0;
-    // heaviest q-tiles (most kv tiles) first
-    const int ntile_seq = (L + 63) / 64;
+    // 8 waves x 16 rows = 128 q rows per workgroup; heaviest tiles first
+    const int ntile_seq = (L + 127) / 128;
     if (tile_id >= ntile_seq) return;
-    const int qs = (ntile_seq - 1 - tile_id) * 64;
+    const int qs = (ntile_seq - 1 - tile_id) * 128;
 
     const int lane = threadIdx.x & 63;
     const int wave = threadIdx.x >> 6;
@@ -533,10 +537,10 @@ __global__ void __launch_bounds__(256) fa_bwd_dq_kernel(
     constexpr int SQ = DPAD + 8;
     extern __shared__ char smem_raw[];
     __bf16* KTl = (__bf16*)smem_raw;           // [DPAD][ST] (K^T, swizzled)
-    __bf16* dSl = KTl + DPAD * ST;             // [4 waves][16 q][ST] (dS strips)
-    __bf16* dSw = dSl + wave * 16 * ST;
-    __bf16* Klds = dSl + 4 * 16 * ST;          // [64 key][SQ] (row-major)
+    __bf16* Klds = KTl + DPAD * ST;            // [64 key][SQ] (row-major)
     __bf16* Vlds = Klds + 64 * SQ;             // [64 key][SQ] (row-major)
+    __bf16* dSl = Vlds + 64 * SQ;              // [8 waves][16 q][ST] (dS strips)
+    __bf16* dSw = dSl + wave * 16 * ST;
 
 #define SWZ(row, col) ((row) * ST + (((((col) >> 3) ^ (((row) >> 3) & 7)) << 3) | ((col) & 7)))
 #define SWZ8(row, col0) ((row) * ST + (((((col0) >> 3) ^ (((row) >> 3) & 7)) << 3)))
@@ -572,7 +576,7 @@ __global__ void __launch_bounds__(256) fa_bwd_dq_kernel(
 #pragma unroll
     for (int dc = 0; dc < DCH; ++dc) dq[dc] = {0.f, 0.f, 0.f, 0.f};
 
-    const int kend_total = min(L, qs + 64);
+    const int kend_total = min(L, qs + 128);
     const int nkt = (kend_total + 63) / 64;
 
     for (int kt = 0; kt < nkt; ++kt) {
@@ -581,7 +585,7 @@ __global__ void __launch_bounds__(256) fa_bwd_dq_kernel(
         // stage K^T (swizzled) + row-major K/V images, cooperative
         {
             const int pieces = 64 * DPAD / 8;
-            for (int pidx = threadIdx.x; pidx < pieces; pidx += 256) {
+            for (int pidx = threadIdx.x; pidx < pieces; pidx += 512) {
                 int key = pidx / (DPAD / 8);
                 int d0 = (pidx % (DPAD / 8)) * 8;
                 bool valid = (ks + key) < kend_total;
@@ -664,16 +668,16 @@ static int launch_fa_bwd(hipStream_t stream, const __bf16* q, const __bf16* k, c
                          int64_t q_ts, int64_t q_gs, int64_t k_ts, int64_t k_hs,
                          int64_t v_ts, int64_t v_hs, int64_t do_ts, int max_tiles, float scale) {
     constexpr int ST = 64 + 8;
-    dim3 block(256);
+    dim3 block(512);
     dim3 grid(max_tiles, batch, H);
     constexpr int SQ = DPAD + 8;
-    size_t shmem_dkv = (size_t)(DPAD * ST * 2 + 64 * ST * 2 + 64 * SQ * 2) * sizeof(__bf16);
+    size_t shmem_dkv = (size_t)(DPAD * ST * 2 + 128 * ST * 2 + 64 * SQ * 2) * sizeof(__bf16);
     hipLaunchKernelGGL((fa_bwd_dkv_kernel<DPAD>), grid, block, shmem_dkv, stream,
                        q, k, v, dout, lse, delta, dk_acc, dv_acc, cu, H, Hkv, D, G,
                        q_ts, q_gs, k_ts, k_hs, v_ts, v_hs, do_ts, T, scale);
     int err = dol_last_error();
     if (err) return err;
-    size_t shmem_dq = (size_t)(DPAD * ST + 64 * ST + 64 * SQ * 2) * sizeof(__bf16);
+    size_t shmem_dq = (size_t)(DPAD * ST + 64 * SQ * 2 + 8 * 16 * ST) * sizeof(__bf16);
     hipLaunchKernelGGL((fa_bwd_dq_kernel<DPAD>), grid, block, shmem_dq, stream,
                        q, k, v, dout, lse, delta, dqkv_q, cu, H, Hkv, D, G,
                        q_ts, q_gs, k_ts, k_hs, v_ts, v_hs, do_ts, T, scale);
@@ -693,7 +697,7 @@ extern "C" int dolomite_fa_varlen_bwd(dolomite_stream_t stream,
                                       float scale, int dtype) {
     if (dtype != DOLOMITE_BF16) return 9010;
     if (D > 128) return 9011;
-    int max_tiles = (max_seqlen + 63) / 64;
+    int max_tiles = (max_seqlen + 127) / 128;
     hipStream_t s = (hipStream_t)stream;
 #define CASE(DP)                                                                                        \
     return launch_fa_bwd<DP>(s, (const __bf16*)q, (const __bf16*)k, (const __bf16*)v,                   \
