@@ -408,6 +408,13 @@ __global__ void __launch_bounds__(512, 2) fa_bwd_dkv_kernel(
         dkr[dc] = {0.f, 0.f, 0.f, 0.f};
     }
 
+    // T5 static priority (guide §5.5): at 2 waves/SIMD the later-dispatched
+    // half of an 8-wave workgroup loses VALU arbitration to the older half;
+    // one setprio(1) for that half removes its start-of-segment penalty.
+    // The guard must be provably wave-uniform (readfirstlane) or s_setprio
+    // is emitted unconditionally.
+    if (__builtin_amdgcn_readfirstlane(threadIdx.x) >= 256) __builtin_amdgcn_s_setprio(1);
+
     const int qt0 = ks / 64;
     const int nqt = (L + 63) / 64;
     const int64_t q_hoff = (int64_t)(h / G) * q_gs + (int64_t)(h % G) * D;
