@@ -376,19 +376,13 @@ __global__ void __launch_bounds__(512, 2) fa_bwd_dkv_kernel(
     const int lg = lane >> 4;
 
     constexpr int SQ = DPAD + 8;               // row-major image stride
-    constexpr int IMG = DPAD * ST + 64 * SQ;   // one Q-or-dO image slab
     extern __shared__ char smem_raw[];
-    // double-buffered per-q-tile images: tile t+1's loads are issued before
-    // tile t's compute (T14) and written into the other buffer after it —
-    // one barrier per iteration, load latency hidden under the MFMAs
-    __bf16* Qimg[2];
-    __bf16* dOimg[2];
-    Qimg[0] = (__bf16*)smem_raw;               // [DPAD][ST] QT + [64][SQ] Q
-    dOimg[0] = Qimg[0] + IMG;
-    Qimg[1] = dOimg[0] + IMG;
-    dOimg[1] = Qimg[1] + IMG;
-    __bf16* dSTl = dOimg[1] + IMG;             // [128 key][ST] (dS^T, [key][q])
+    __bf16* QTl = (__bf16*)smem_raw;           // [DPAD][ST]    (Q^T, swizzled)
+    __bf16* dOTl = QTl + DPAD * ST;            // [DPAD][ST]    (dO^T, swizzled)
+    __bf16* dSTl = dOTl + DPAD * ST;           // [128 key][ST] (dS^T, [key][q])
     __bf16* PTl = dSTl + 128 * ST;             // [128 key][ST] (P^T, [key][q])
+    __bf16* Qlds = PTl + 128 * ST;             // [64 q][SQ]    (row-major)
+    __bf16* dOl = Qlds + 64 * SQ;              // [64 q][SQ]    (row-major)
 
 #define SWZ(row, col) ((row) * ST + (((((col) >> 3) ^ (((row) >> 3) & 7)) << 3) | ((col) & 7)))
 #define SWZ8(row, col0) ((row) * ST + (((((col0) >> 3) ^ (((row) >> 3) & 7)) << 3)))
@@ -426,53 +420,28 @@ __global__ void __launch_bounds__(512, 2) fa_bwd_dkv_kernel(
     const int64_t q_hoff = (int64_t)(h / G) * q_gs + (int64_t)(h % G) * D;
     const int64_t do_hoff = (int64_t)h * D;
 
-    constexpr int PIECES = 64 * DPAD / 8;
-    constexpr int PPT = (PIECES + 511) / 512;  // pieces per thread
-    bf16x8 qst[PPT], dst_[PPT];
-
-#define DKV_STAGE_LOAD(QS)                                                                              \
-    _Pragma("unroll") for (int pi = 0; pi < PPT; ++pi) {                                                \
-        int pidx = (int)threadIdx.x + pi * 512;                                                         \
-        if (pidx < PIECES) {                                                                            \
-            int qq = pidx / (DPAD / 8);                                                                 \
-            int d0 = (pidx % (DPAD / 8)) * 8;                                                           \
-            bool valid = ((QS) + qq) < L;                                                               \
-            const __bf16* qp = q + (int64_t)(s0 + (valid ? (QS) + qq : 0)) * q_ts + q_hoff + d0;        \
-            qst[pi] = load_bf16x8_guard(qp, d0, D, valid);                                              \
-            const __bf16* dp = dout + (int64_t)(s0 + (valid ? (QS) + qq : 0)) * do_ts + do_hoff + d0;   \
-            dst_[pi] = load_bf16x8_guard(dp, d0, D, valid);                                             \
-        }                                                                                               \
-    }
-
-#define DKV_STAGE_WRITE(BUF)                                                                            \
-    _Pragma("unroll") for (int pi = 0; pi < PPT; ++pi) {                                                \
-        int pidx = (int)threadIdx.x + pi * 512;                                                         \
-        if (pidx < PIECES) {                                                                            \
-            int qq = pidx / (DPAD / 8);                                                                 \
-            int d0 = (pidx % (DPAD / 8)) * 8;                                                           \
-            __bf16* qi = Qimg[BUF];                                                                     \
-            __bf16* di = dOimg[BUF];                                                                    \
-            *(bf16x8*)&qi[DPAD * ST + qq * SQ + d0] = qst[pi];                                          \
-            *(bf16x8*)&di[DPAD * ST + qq * SQ + d0] = dst_[pi];                                         \
-            _Pragma("unroll") for (int e = 0; e < 8; ++e) {                                             \
-                qi[SWZ(d0 + e, qq)] = qst[pi][e];                                                       \
-                di[SWZ(d0 + e, qq)] = dst_[pi][e];                                                      \
-            }                                                                                           \
-        }                                                                                               \
-    }
-
-    int cur = 0;
-    DKV_STAGE_LOAD(qt0 * 64)
-    DKV_STAGE_WRITE(0)
-    __syncthreads();
-
     for (int qt = qt0; qt < nqt; ++qt) {
         const int qs = qt * 64;
-        const __bf16* QTl = Qimg[cur];
-        const __bf16* Qlds = Qimg[cur] + DPAD * ST;
-        const __bf16* dOTl = dOimg[cur];
-        const __bf16* dOl = dOimg[cur] + DPAD * ST;
-        if (qt + 1 < nqt) DKV_STAGE_LOAD((qt + 1) * 64)  // in flight over compute
+        __syncthreads();  // previous iteration's image reads done
+        {
+            const int pieces = 64 * DPAD / 8;
+            for (int pidx = threadIdx.x; pidx < pieces; pidx += 512) {
+                int qq = pidx / (DPAD / 8);
+                int d0 = (pidx % (DPAD / 8)) * 8;
+                bool valid = (qs + qq) < L;
+                const __bf16* qp = q + (int64_t)(s0 + (valid ? qs + qq : 0)) * q_ts + q_hoff + d0;
+                bf16x8 qq8 = load_bf16x8_guard(qp, d0, D, valid);
+                *(bf16x8*)&Qlds[qq * SQ + d0] = qq8;
+#pragma unroll
+                for (int e = 0; e < 8; ++e) QTl[SWZ(d0 + e, qq)] = qq8[e];
+                const __bf16* dp = dout + (int64_t)(s0 + (valid ? qs + qq : 0)) * do_ts + do_hoff + d0;
+                bf16x8 dd8 = load_bf16x8_guard(dp, d0, D, valid);
+                *(bf16x8*)&dOl[qq * SQ + d0] = dd8;
+#pragma unroll
+                for (int e = 0; e < 8; ++e) dOTl[SWZ(d0 + e, qq)] = dd8[e];
+            }
+        }
+        __syncthreads();
 
         // S^T = K*Q^T, dP^T = V*dO^T; B-frags (identical across the 4
         // waves) come from the row-major LDS images: Q[q=lr][d0..d0+8)
@@ -525,12 +494,7 @@ __global__ void __launch_bounds__(512, 2) fa_bwd_dkv_kernel(
                 dkr[dc] = MFMA16(dstf, qtb, dkr[dc]);
             }
         }
-        if (qt + 1 < nqt) DKV_STAGE_WRITE(cur ^ 1)  // land t+1 after compute (T14)
-        __syncthreads();
-        cur ^= 1;
     }
-#undef DKV_STAGE_LOAD
-#undef DKV_STAGE_WRITE
 
     // join dK/dV (G q-head contributors per kv strip)
 #pragma unroll
@@ -714,7 +678,7 @@ static int launch_fa_bwd(hipStream_t stream, const __bf16* q, const __bf16* k, c
     dim3 block(512);
     dim3 grid(max_tiles, batch, H);
     constexpr int SQ = DPAD + 8;
-    size_t shmem_dkv = (size_t)((DPAD * ST + 64 * SQ) * 4 + 128 * ST * 2) * sizeof(__bf16);
+    size_t shmem_dkv = (size_t)(DPAD * ST * 2 + 128 * ST * 2 + 64 * SQ * 2) * sizeof(__bf16);
     hipLaunchKernelGGL((fa_bwd_dkv_kernel<DPAD>), grid, block, shmem_dkv, stream,
                        q, k, v, dout, lse, delta, dk_acc, dv_acc, cu, H, Hkv, D, G,
                        q_ts, q_gs, k_ts, k_hs, v_ts, v_hs, do_ts, T, scale);
