@@ -69,3 +69,64 @@ def test_native_extension_actually_loaded():
 
     assert hip._lib is not None, "HIP extension was never called in the GPU model tests"
     assert "libdolomite_hip.so" in str(hip.so_path())
+
+
+def test_llama_shaped_finetune_step_8k(golden_dir):
+    """Config #4 shape class on hardware: GQA d_head=128, swiglu, rmsnorm,
+    one 8192-token packed sequence, block gradient checkpointing — a full
+    bf16 training step with finite loss/grads and loss matching the oracle."""
+    from dolomite_engine_amd.hf_models import (
+        GPTDolomiteConfig,
+        GPTDolomiteForCausalLM,
+        apply_gradient_checkpointing,
+    )
+    from oracle import OracleConfig, OracleGPTDolomiteForCausalLM
+
+    torch.manual_seed(5)
+    kw = dict(
+        vocab_size=2048,
+        n_positions=8192,
+        n_embd=512,
+        n_layer=2,
+        n_head=8,
+        num_key_value_heads=2,
+        attention_head_type="gqa",
+        n_inner=1024,
+        activation_function="swiglu",
+        normalization_function="rmsnorm",
+        position_embedding_type="rope",
+        resid_pdrop=0.0,
+        embd_pdrop=0.0,
+        attn_pdrop=0.0,
+        tie_word_embeddings=False,
+    )
+    cfg = GPTDolomiteConfig(**kw)
+    cfg._attn_implementation = "flash_attention_2"
+    model = GPTDolomiteForCausalLM(cfg, use_padding_free_transformer=True)
+    apply_gradient_checkpointing(model, "block", checkpoint_every=1)
+    model = model.to(torch.bfloat16).cuda()
+    model.train()
+
+    S = 8192
+    g = torch.Generator().manual_seed(9)
+    ids = torch.randint(0, 2048, (S,), generator=g).cuda()
+    pos = torch.arange(S).cuda()
+    cu = torch.tensor([0, S], dtype=torch.int32).cuda()
+    out = model(input_ids=ids, position_ids=pos, cu_seqlens=cu, max_seqlen=S, labels=ids)
+    out.loss.backward()
+    torch.cuda.synchronize()
+    assert torch.isfinite(out.loss)
+    assert all(torch.isfinite(p.grad).all() for p in model.parameters() if p.grad is not None)
+
+    ocfg = OracleConfig(
+        vocab_size=2048, n_positions=8192, n_embd=512, n_layer=2, n_head=8,
+        num_key_value_heads=2, attention_head_type="gqa", n_inner=1024,
+        activation_function="swiglu", normalization_function="rmsnorm",
+        position_embedding_type="rope", tie_word_embeddings=False,
+    )
+    omodel = OracleGPTDolomiteForCausalLM(ocfg)
+    omodel.load_state_dict({k: v.float().cpu() for k, v in model.state_dict().items()}, strict=False)
+    with torch.no_grad():
+        _, oloss = omodel(ids.cpu(), pos.cpu(), cu.cpu(), S, labels=ids.cpu())
+    rel = abs(float(out.loss) - float(oloss)) / abs(float(oloss))
+    assert rel < 2e-2, (float(out.loss), float(oloss))
