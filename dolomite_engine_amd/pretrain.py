@@ -58,7 +58,32 @@ def build_engine(args: TrainingArgs):
     return model_wrapper, engine, lr_scheduler
 
 
-def train(args: TrainingArgs, model_wrapper, engine, lr_scheduler, train_loader, starting_step=0, metadata=None):
+@torch.no_grad()
+def evaluate(val_iter, model_wrapper, global_step: int, eval_steps: int) -> float:
+    """Validation loop (reference pretrain.py:221-279): mean loss over
+    eval_steps batches, AVG-reduced over the data-parallel group."""
+    import torch.distributed as dist
+
+    from .utils import is_initialized
+
+    model_wrapper.eval()
+    loss_sum = 0
+    for _ in range(eval_steps):
+        batch = next(val_iter)
+        loss_sum = loss_sum + model_wrapper(batch)
+    loss_mean = loss_sum / eval_steps
+    if is_initialized():
+        dist.all_reduce(loss_mean, op=dist.ReduceOp.AVG if dist.get_backend() == "nccl" else dist.ReduceOp.SUM)
+        if dist.get_backend() != "nccl":
+            loss_mean = loss_mean / dist.get_world_size()
+    loss_mean = loss_mean.item()
+    log_rank_0(f"step = {global_step}, val_loss = {loss_mean:.5f}")
+    model_wrapper.train()
+    return loss_mean
+
+
+def train(args: TrainingArgs, model_wrapper, engine, lr_scheduler, train_loader, starting_step=0, metadata=None,
+          val_loader=None):
     tp = args.training_parameters
     ga = tp.gradient_accumulation_steps
     world = get_world_size()
@@ -104,6 +129,14 @@ def train(args: TrainingArgs, model_wrapper, engine, lr_scheduler, train_loader,
                 f"learning_rate = {lr_scheduler.get_lr():.3e}, step time (sec) = {dt:.3f}, "
                 f"throughput = {tokens_per_step / dt:.1f} tokens/s, FLOPS = {tflops_per_step / dt:.1f} TFLOPs"
             )
+
+        if (
+            tp.eval_during_training
+            and tp.eval_interval is not None
+            and val_loader is not None
+            and global_step % tp.eval_interval == 0
+        ):
+            evaluate(iter(val_loader), model_wrapper, global_step, eval_steps=8)
 
         if args.save_args is not None and args.save_args.save_interval and global_step % args.save_args.save_interval == 0:
             save_checkpoint(
@@ -178,7 +211,15 @@ def main(argv=None):
         if not args.load_args.load_starting_iteration:
             starting_step = 0
 
-    train(args, model_wrapper, engine, lr_scheduler, train_loader, starting_step, metadata)
+    val_loader = None
+    if args.training_parameters.eval_during_training and args.training_parameters.eval_interval is not None:
+        # held-out synthetic stream (different seed space than training)
+        val_loader = SyntheticPretrainingDataLoader(
+            tp.micro_batch_size, tp.sequence_length, model_wrapper.config.vocab_size,
+            seed=args.random_args.seed + 777777,
+        )
+
+    train(args, model_wrapper, engine, lr_scheduler, train_loader, starting_step, metadata, val_loader=val_loader)
 
 
 if __name__ == "__main__":

@@ -264,3 +264,19 @@ def test_reset_attention_mask_document_boundaries():
     batch = {"text": torch.randint(0, 512, (2, 9))}
     loss = w(batch)
     assert torch.isfinite(loss)
+
+
+def test_pretrain_with_eval_interval(tmp_path, capsys):
+    import yaml
+
+    from dolomite_engine_amd import pretrain
+
+    cfg = json.loads(json.dumps(TINY_CONFIG))
+    cfg["training_parameters"]["num_training_steps"] = 4
+    cfg["training_parameters"]["eval_interval"] = 2
+    p = tmp_path / "cfg.yml"
+    with open(p, "w") as f:
+        yaml.safe_dump(cfg, f)
+    pretrain.main(["--config", str(p)])
+    out = capsys.readouterr().out
+    assert "val_loss" in out
