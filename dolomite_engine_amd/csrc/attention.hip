@@ -33,6 +33,11 @@ typedef float f32x4 __attribute__((ext_vector_type(4)));
 
 #define MFMA16(a, b, c) __builtin_amdgcn_mfma_f32_16x16x32_bf16((a), (b), (c), 0, 0, 0)
 
+// softmax runs in the exp2 domain: v_exp_f32 natively computes 2^x, so
+// folding log2(e) into the softmax scale removes one multiply per element
+#define DOL_LOG2E 1.44269504088896340736f
+#define DOL_LN2 0.69314718055994530942f
+
 // Branchless guarded 16B load. Requires D % 8 == 0 (every head dim on this
 // path): a chunk is then fully inside [0, D) or fully in the pad, so the
 // guard reduces to ONE wave-divergent-free vector load from a clamped
@@ -187,6 +192,7 @@ __global__ void __launch_bounds__(512) fa_fwd_kernel(
         // --- mask + online softmax (fp32, per C-row) ---
         float p_val[4][4];
         float alpha[4];
+        const float scale2 = scale * DOL_LOG2E;  // exp2-domain scores
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
             const int qpos = qs + wave * 16 + lg * 4 + r;
@@ -194,7 +200,7 @@ __global__ void __launch_bounds__(512) fa_fwd_kernel(
 #pragma unroll
             for (int cb = 0; cb < 4; ++cb) {
                 const int kpos = ks + cb * 16 + lr;
-                float s = sc[cb][r] * scale;
+                float s = sc[cb][r] * scale2;
                 if (kpos > qpos || kpos >= kend || qpos >= L) s = -INFINITY;
                 p_val[cb][r] = s;
                 rowmax = fmaxf(rowmax, s);
@@ -202,11 +208,11 @@ __global__ void __launch_bounds__(512) fa_fwd_kernel(
             rowmax = qwave_reduce_max(rowmax);
             float mnew = fmaxf(m_run[r], rowmax);
             if (mnew == -INFINITY) mnew = 0.f;  // fully-masked row guard
-            alpha[r] = (m_run[r] == -INFINITY) ? 0.f : __expf(m_run[r] - mnew);
+            alpha[r] = (m_run[r] == -INFINITY) ? 0.f : exp2f(m_run[r] - mnew);
             float rsum = 0.f;
 #pragma unroll
             for (int cb = 0; cb < 4; ++cb) {
-                float e = (p_val[cb][r] == -INFINITY) ? 0.f : __expf(p_val[cb][r] - mnew);
+                float e = (p_val[cb][r] == -INFINITY) ? 0.f : exp2f(p_val[cb][r] - mnew);
                 p_val[cb][r] = e;
                 rsum += e;
             }
@@ -244,7 +250,7 @@ __global__ void __launch_bounds__(512) fa_fwd_kernel(
         inv_l[r] = (l_run[r] > 0.f) ? 1.f / l_run[r] : 0.f;
         const int qpos = qs + wave * 16 + lg * 4 + r;
         if (lr == 0 && qpos < L)
-            lse[(int64_t)h * T_total + s0 + qpos] = m_run[r] + __logf(l_run[r]);
+            lse[(int64_t)h * T_total + s0 + qpos] = (m_run[r] + log2f(l_run[r])) * DOL_LN2;
     }
 #pragma unroll
     for (int dc = 0; dc < DCH; ++dc) {
@@ -468,13 +474,13 @@ __global__ void __launch_bounds__(512, 2) fa_bwd_dkv_kernel(
         for (int cb = 0; cb < 4; ++cb) {
             const int qpos = qs + cb * 16 + lr;
             const bool qok = qpos < L;
-            float lsev = qok ? lse[(int64_t)h * T_total + s0 + qpos] : 0.f;
+            float lsev = qok ? lse[(int64_t)h * T_total + s0 + qpos] * DOL_LOG2E : 0.f;
             float delv = qok ? delta[(int64_t)h * T_total + s0 + qpos] : 0.f;
 #pragma unroll
             for (int r = 0; r < 4; ++r) {
                 const int kpos = ks + wave * 16 + lg * 4 + r;
                 bool ok = full_tile || (qok && kpos < kend && kpos <= qpos);
-                float pv = ok ? __expf(st[cb][r] * scale - lsev) : 0.f;
+                float pv = ok ? exp2f(st[cb][r] * (scale * DOL_LOG2E) - lsev) : 0.f;
                 float ds = ok ? pv * (dpt[cb][r] - delv) * scale : 0.f;
                 PTl[(wave * 16 + lg * 4 + r) * ST + cb * 16 + lr] = (__bf16)pv;
                 dSTl[(wave * 16 + lg * 4 + r) * ST + cb * 16 + lr] = (__bf16)ds;
@@ -575,7 +581,7 @@ __global__ void __launch_bounds__(512) fa_bwd_dq_kernel(
     for (int r = 0; r < 4; ++r) {
         const int qp_ = qs + wave * 16 + lg * 4 + r;
         bool ok = qp_ < L;
-        lsev[r] = ok ? lse[(int64_t)h * T_total + s0 + qp_] : 0.f;
+        lsev[r] = ok ? lse[(int64_t)h * T_total + s0 + qp_] * DOL_LOG2E : 0.f;
         delv[r] = ok ? delta[(int64_t)h * T_total + s0 + qp_] : 0.f;
     }
 
@@ -633,7 +639,7 @@ __global__ void __launch_bounds__(512) fa_bwd_dq_kernel(
                 const int qpos = qs + wave * 16 + lg * 4 + r;
                 const int kpos = ks + cb * 16 + lr;
                 bool ok = full_tile || (qpos < L && kpos < kend_total && kpos <= qpos);
-                float pv = ok ? __expf(sc[cb][r] * scale - lsev[r]) : 0.f;
+                float pv = ok ? exp2f(sc[cb][r] * (scale * DOL_LOG2E) - lsev[r]) : 0.f;
                 float ds = ok ? pv * (dp[cb][r] - delv[r]) * scale : 0.f;
                 dSw[(lg * 4 + r) * ST + cb * 16 + lr] = (__bf16)ds;
             }
