@@ -177,17 +177,13 @@ __global__ void __launch_bounds__(512) fa_fwd_kernel(
         }
         __syncthreads();
 
-        // --- QK^T: 4 key-blocks of 16, accumulate over KCH chunks.
-        // kc is the OUTER loop: consecutive MFMAs then target different
-        // accumulators (dependent-accumulator latency 40 cyc vs 17 cyc
-        // issue — same-accumulator chains stall the matrix pipe) ---
+        // --- QK^T: 4 key-blocks of 16, accumulate over KCH chunks ---
         f32x4 sc[4];
 #pragma unroll
-        for (int cb = 0; cb < 4; ++cb) sc[cb] = {0.f, 0.f, 0.f, 0.f};
+        for (int cb = 0; cb < 4; ++cb) {
+            sc[cb] = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
-        for (int kc = 0; kc < KCH; ++kc) {
-#pragma unroll
-            for (int cb = 0; cb < 4; ++cb) {
+            for (int kc = 0; kc < KCH; ++kc) {
                 bf16x8 kf = *(const bf16x8*)&Klds[(cb * 16 + lr) * SK + kc * 32 + lg * 8];
                 sc[cb] = MFMA16(qf[kc], kf, sc[cb]);
             }
@@ -460,13 +456,9 @@ __global__ void __launch_bounds__(512, 2) fa_bwd_dkv_kernel(
         for (int cb = 0; cb < 4; ++cb) {
             st[cb] = {0.f, 0.f, 0.f, 0.f};
             dpt[cb] = {0.f, 0.f, 0.f, 0.f};
-        }
-        // kc outer: 8 independent accumulators between same-acc MFMAs
 #pragma unroll
-        for (int kc = 0; kc < KCH; ++kc) {
-            int d0 = kc * 32 + lg * 8;
-#pragma unroll
-            for (int cb = 0; cb < 4; ++cb) {
+            for (int kc = 0; kc < KCH; ++kc) {
+                int d0 = kc * 32 + lg * 8;
                 bf16x8 qb = *(const bf16x8*)&Qlds[(cb * 16 + lr) * SQ + d0];
                 st[cb] = MFMA16(kfr[kc], qb, st[cb]);
                 bf16x8 db = *(const bf16x8*)&dOl[(cb * 16 + lr) * SQ + d0];
@@ -627,13 +619,9 @@ __global__ void __launch_bounds__(512) fa_bwd_dq_kernel(
         for (int cb = 0; cb < 4; ++cb) {
             sc[cb] = {0.f, 0.f, 0.f, 0.f};
             dp[cb] = {0.f, 0.f, 0.f, 0.f};
-        }
-        // kc outer: 8 independent accumulators between same-acc MFMAs
 #pragma unroll
-        for (int kc = 0; kc < KCH; ++kc) {
-            int d0 = kc * 32 + lg * 8;
-#pragma unroll
-            for (int cb = 0; cb < 4; ++cb) {
+            for (int kc = 0; kc < KCH; ++kc) {
+                int d0 = kc * 32 + lg * 8;
                 bf16x8 kb = *(const bf16x8*)&Klds[(cb * 16 + lr) * SQ + d0];
                 sc[cb] = MFMA16(qfr[kc], kb, sc[cb]);
                 bf16x8 vb = *(const bf16x8*)&Vlds[(cb * 16 + lr) * SQ + d0];
