@@ -77,7 +77,9 @@ def evaluate(val_iter, model_wrapper, global_step: int, eval_steps: int) -> floa
         if dist.get_backend() != "nccl":
             loss_mean = loss_mean / dist.get_world_size()
     loss_mean = loss_mean.item()
-    log_rank_0(f"step = {global_step}, val_loss = {loss_mean:.5f}")
+    from .tracking import ExperimentsTracker, track_val_metrics
+
+    track_val_metrics(global_step, loss_mean, ExperimentsTracker())
     model_wrapper.train()
     return loss_mean
 
@@ -98,6 +100,14 @@ def train(args: TrainingArgs, model_wrapper, engine, lr_scheduler, train_loader,
         * ga
         * world
     )
+
+    from .tracking import ExperimentsTracker, RunningMean, track_train_metrics
+
+    tracker = ExperimentsTracker(
+        args.logging_args.experiments_tracker_name,
+        args.save_args.save_path if args.save_args is not None else None,
+    )
+    running = RunningMean()
 
     train_iter = iter(train_loader)
     model_wrapper.train()
@@ -123,11 +133,18 @@ def train(args: TrainingArgs, model_wrapper, engine, lr_scheduler, train_loader,
             torch.cuda.synchronize()
         dt = time.perf_counter() - t0
 
+        loss_mean = running.add_loss(loss)
         if global_step % args.logging_args.log_interval == 0:
-            log_rank_0(
-                f"step = {global_step}, loss_step = {loss:.5f}, grad_norm = {grad_norm:.3f}, "
-                f"learning_rate = {lr_scheduler.get_lr():.3e}, step time (sec) = {dt:.3f}, "
-                f"throughput = {tokens_per_step / dt:.1f} tokens/s, FLOPS = {tflops_per_step / dt:.1f} TFLOPs"
+            track_train_metrics(
+                global_step,
+                loss,
+                grad_norm,
+                lr_scheduler.get_lr(),
+                tracker,
+                loss_mean,
+                flops=tflops_per_step * 1e12 / dt,
+                billion_tokens_per_day=tokens_per_step / dt * 86400 / 1e9,
+                step_time=dt,
             )
 
         if (

@@ -351,3 +351,24 @@ def test_zero2_gloo_grad_accumulation_matches_single_process(tmp_path):
     sd = model.state_dict()
     for k, v in dist_result["state"].items():
         torch.testing.assert_close(sd[k], v, rtol=1e-5, atol=1e-6, msg=lambda m: f"{k}: {m}")
+
+
+def test_jsonl_tracker_writes_reference_metric_names(tmp_path):
+    import yaml
+
+    from dolomite_engine_amd import pretrain
+
+    cfg = json.loads(json.dumps(TINY_CONFIG))
+    cfg["training_parameters"]["num_training_steps"] = 2
+    cfg["logging_args"] = {"experiments_tracker_name": "jsonl"}
+    cfg["save_args"] = {"save_path": str(tmp_path / "run")}
+    p = tmp_path / "cfg.yml"
+    with open(p, "w") as f:
+        yaml.safe_dump(cfg, f)
+    pretrain.main(["--config", str(p)])
+    rows = [json.loads(l) for l in open(tmp_path / "run" / "metrics.jsonl")]
+    assert rows, "no metrics written"
+    # reference metric names (train_utils.py:119-179)
+    for key in ["loss_step", "loss_running_mean", "learning_rate", "grad_norm", "FLOPS",
+                "throughput (B tokens/day)", "step time (sec)"]:
+        assert key in rows[0], key
