@@ -44,6 +44,32 @@ MODEL_3B = dict(
     pad_token_id=0,
 )
 
+# GPTDolomite-350M (config #2 shape: h=1024, L=24, seq 2048)
+MODEL_350M = dict(
+    vocab_size=49152,
+    n_positions=2048,
+    n_embd=1024,
+    n_layer=24,
+    n_head=16,
+    num_key_value_heads=None,
+    attention_head_type="mqa",
+    n_inner=4096,
+    activation_function="gelu_pytorch_tanh",
+    normalization_function="rmsnorm",
+    position_embedding_type="rope",
+    rope_theta=10000,
+    resid_pdrop=0.0,
+    embd_pdrop=0.0,
+    attn_pdrop=0.0,
+    layer_norm_epsilon=1e-5,
+    tie_word_embeddings=True,
+    bos_token_id=0,
+    eos_token_id=0,
+    pad_token_id=0,
+)
+
+MODELS = {"3b": MODEL_3B, "350m": MODEL_350M}
+
 SEQ_LEN = 4096
 MICRO_BATCH = 16  # tokens per rank per step = 16 * 4096 = 65536 (fills HBM better; +5% vs B=8)
 
@@ -57,7 +83,7 @@ def build(args, device):
         micro_batch_size=args.micro_batch,
         sequence_length=args.seq_len,
         model_name=None,
-        pretrained_config=dict(MODEL_3B),
+        pretrained_config=dict(MODELS[args.model]),
         dtype="bf16" if device.type == "cuda" else "fp32",
         attention_implementation="flash_attention_2" if device.type == "cuda" else "eager",
         use_padding_free_transformer=device.type == "cuda",
@@ -158,10 +184,16 @@ def main():
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=6)
     ap.add_argument("--warmup", type=int, default=3)
-    ap.add_argument("--micro-batch", type=int, default=MICRO_BATCH)
-    ap.add_argument("--seq-len", type=int, default=SEQ_LEN)
+    ap.add_argument("--micro-batch", type=int, default=None)
+    ap.add_argument("--seq-len", type=int, default=None)
+    ap.add_argument("--model", choices=sorted(MODELS), default="3b",
+                    help="3b = BASELINE config #3 (the headline metric); 350m = config #2")
     ap.add_argument("--skip-cpu-baseline", action="store_true")
     args = ap.parse_args()
+    if args.seq_len is None:
+        args.seq_len = SEQ_LEN if args.model == "3b" else 2048
+    if args.micro_batch is None:
+        args.micro_batch = MICRO_BATCH if args.model == "3b" else 64
 
     import torch.distributed as dist
 
@@ -178,7 +210,7 @@ def main():
 
     wrapper, engine = build(args, device)
     scheduler = LRScheduler(1e-5, 0, 0, None, 10**9, "constant", 0.1)
-    vocab = MODEL_3B["vocab_size"]
+    vocab = MODELS[args.model]["vocab_size"]
 
     # warmup
     run_steps(wrapper, engine, scheduler, rank, args.micro_batch, args.seq_len, vocab, args.warmup)
@@ -212,8 +244,9 @@ def main():
     ms_per_step = elapsed / args.steps * 1000
 
     # roofline of the dominant hand-written kernel (per launch, algorithmic)
-    B, S, L = args.micro_batch, args.seq_len, MODEL_3B["n_layer"]
-    H, D, h = MODEL_3B["n_head"], MODEL_3B["n_embd"] // MODEL_3B["n_head"], MODEL_3B["n_embd"]
+    MC = MODELS[args.model]
+    B, S, L = args.micro_batch, args.seq_len, MC["n_layer"]
+    H, D, h = MC["n_head"], MC["n_embd"] // MC["n_head"], MC["n_embd"]
     fa_fwd_flops = 2.0 * H * D * B * S * S  # QK^T + PV, causal half x2 matmuls, per layer-launch
     fa_bwd_flops = 2.5 * fa_fwd_flops  # 5 contractions vs 2 in fwd
     flops_per_launch = {"fa_varlen_fwd": fa_fwd_flops, "fa_varlen_bwd": fa_bwd_flops}
@@ -268,11 +301,15 @@ def main():
             }
 
     cpu = None
-    if not args.skip_cpu_baseline and world == 1:
+    if not args.skip_cpu_baseline and world == 1 and args.model == "3b":
         cpu = cpu_baseline(args)
 
     result = {
-        "metric": "tokens/sec/node GPTDolomite-3B bf16 seq4096 padding-free",
+        "metric": (
+            "tokens/sec/node GPTDolomite-3B bf16 seq4096 padding-free"
+            if args.model == "3b"
+            else f"tokens/sec/node GPTDolomite-{args.model} bf16 seq{args.seq_len} padding-free"
+        ),
         "value": value,
         "unit": "tokens/s",
         "n_gpus": world,
@@ -285,8 +322,8 @@ def main():
         "dtype": "bf16",
         "data": "synthetic",
         "config": {
-            "workload": "gptdolomite-3b-bf16-seq4096-paddingfree-pretrain",
-            "model": "GPTDolomite-3B",
+            "workload": f"gptdolomite-{args.model}-bf16-seq{args.seq_len}-paddingfree-pretrain",
+            "model": "GPTDolomite-3B" if args.model == "3b" else "GPTDolomite-350M",
             "global_batch": args.micro_batch * world,
             "seq_len": args.seq_len,
             "parallelism": f"dp{world}",
