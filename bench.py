@@ -306,9 +306,8 @@ def main():
 
     result = {
         "metric": (
-            "tokens/sec/node GPTDolomite-3B bf16 seq4096 padding-free"
-            if args.model == "3b"
-            else f"tokens/sec/node GPTDolomite-{args.model} bf16 seq{args.seq_len} padding-free"
+            f"tokens/sec/node GPTDolomite-{'3B' if args.model == '3b' else '350M'} "
+            f"bf16 seq{args.seq_len} padding-free"
         ),
         "value": value,
         "unit": "tokens/s",
