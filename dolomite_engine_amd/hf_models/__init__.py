@@ -10,6 +10,14 @@ from .modeling import (
     ParameterizedEmbedding,
     ParameterizedLinear,
 )
+from .moe import (
+    MoEDolomiteBlock,
+    MoEDolomiteConfig,
+    MoEDolomiteForCausalLM,
+    MoEDolomiteModel,
+    ParameterizedExperts,
+    SparseMoE,
+)
 
 
 def register_model_classes() -> None:
@@ -19,3 +27,6 @@ def register_model_classes() -> None:
     AutoConfig.register("gpt_dolomite", GPTDolomiteConfig, exist_ok=True)
     AutoModel.register(GPTDolomiteConfig, GPTDolomiteModel, exist_ok=True)
     AutoModelForCausalLM.register(GPTDolomiteConfig, GPTDolomiteForCausalLM, exist_ok=True)
+    AutoConfig.register("moe_dolomite", MoEDolomiteConfig, exist_ok=True)
+    AutoModel.register(MoEDolomiteConfig, MoEDolomiteModel, exist_ok=True)
+    AutoModelForCausalLM.register(MoEDolomiteConfig, MoEDolomiteForCausalLM, exist_ok=True)

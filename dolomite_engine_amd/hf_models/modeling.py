@@ -438,6 +438,8 @@ class GPTDolomitePreTrainedModel(PreTrainedModel):
 
 
 class GPTDolomiteModel(GPTDolomitePreTrainedModel):
+    block_class = GPTDolomiteBlock  # swapped by model families (MoE)
+
     def __init__(self, config: GPTDolomiteConfig, **kwargs):
         super().__init__(config, **kwargs)
         self.embed_dim = config.n_embd
@@ -447,7 +449,7 @@ class GPTDolomiteModel(GPTDolomitePreTrainedModel):
         self.drop = nn.Identity() if config.embd_pdrop == 0 else nn.Dropout(config.embd_pdrop)
         self.h = nn.ModuleList(
             [
-                GPTDolomiteBlock(config, self._use_padding_free_transformer, layer_idx=i)
+                self.block_class(config, self._use_padding_free_transformer, layer_idx=i)
                 for i in range(config.n_layer)
             ]
         )
@@ -578,10 +580,11 @@ class GPTDolomiteModel(GPTDolomitePreTrainedModel):
 
 class GPTDolomiteForCausalLM(GPTDolomitePreTrainedModel):
     _tied_weights_keys = {"lm_head.weight": "transformer.wte.weight"}
+    model_class = GPTDolomiteModel  # swapped by model families (MoE)
 
     def __init__(self, config: GPTDolomiteConfig, **kwargs):
         super().__init__(config, **kwargs)
-        self.transformer = GPTDolomiteModel(config, use_padding_free_transformer=self._use_padding_free_transformer)
+        self.transformer = self.model_class(config, use_padding_free_transformer=self._use_padding_free_transformer)
         if not self._tied_word_embeddings:
             self.lm_head = ParameterizedLinear(
                 config.n_embd, config.vocab_size, bias=False, std=config.initializer_range

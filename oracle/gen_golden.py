@@ -124,6 +124,66 @@ def gen_model_cases():
         print(f"model_{name}: loss={out.loss.item():.6f}")
 
 
+MOE_GRAD_KEYS = [
+    "transformer.wte.weight",
+    "transformer.h.0.attn.c_attn.weight",
+    "transformer.h.0.mlp.gate.weight",
+    "transformer.h.0.mlp.c_fc.weight",
+    "transformer.h.1.mlp.c_proj.weight",
+    "transformer.ln_f.weight",
+    "lm_head.weight",
+]
+
+
+def gen_moe_case():
+    """Tiny MoE fixture from the reference MoEDolomiteForCausalLM (eager moe,
+    eager attention, fp32). Two runs: CLM-only loss, and CLM + mixtral
+    load-balancing aux loss with grads (pins the router gradient path)."""
+    from oracle.ref_shim import make_reference_moe_config, make_reference_moe_model
+
+    torch.manual_seed(4321)
+    cfg_kwargs = dict(TINY)
+    cfg_kwargs.update(
+        attention_head_type="mqa",
+        position_embedding_type="rope",
+        normalization_function="rmsnorm",
+        activation_function="swiglu",
+        add_bias=False,
+        num_experts=4,
+        num_experts_per_tok=2,
+        router_aux_loss_coef=0.01,
+    )
+    cfg = make_reference_moe_config(**cfg_kwargs)
+    model = make_reference_moe_model(cfg, "eager")
+    model.eval()
+
+    B, S = 2, 12
+    input_ids = torch.randint(0, cfg_kwargs["vocab_size"], (B, S), generator=torch.Generator().manual_seed(99))
+    labels = input_ids.clone()
+
+    out_plain = model(input_ids=input_ids, labels=labels, output_router_logits=False)
+    out = model(input_ids=input_ids, labels=labels, output_router_logits=True)
+    out.loss.backward()
+    params = dict(model.named_parameters())
+    grads = {k: params[k].grad.detach().clone() for k in MOE_GRAD_KEYS if k in params}
+
+    torch.save(
+        dict(
+            config=cfg_kwargs,
+            state_dict={k: v.detach().clone() for k, v in model.state_dict().items()},
+            input_ids=input_ids,
+            labels=labels,
+            logits=out.logits.detach().clone(),
+            loss=out_plain.loss.detach().clone(),
+            loss_with_aux=out.loss.detach().clone(),
+            aux_loss=out.aux_loss.detach().clone(),
+            grads=grads,
+        ),
+        GOLDEN_DIR / "model_moe_mqa_rope_rmsnorm_swiglu.pt",
+    )
+    print(f"model_moe: loss={out_plain.loss.item():.6f} aux={out.aux_loss.item():.6f}")
+
+
 def gen_ops():
     from oracle.ref_shim import import_reference_hf_models
 
@@ -252,6 +312,12 @@ def gen_scheduler():
 
 if __name__ == "__main__":
     GOLDEN_DIR.mkdir(parents=True, exist_ok=True)
-    gen_model_cases()
-    gen_ops()
-    gen_scheduler()
+    import sys as _sys
+
+    if len(_sys.argv) > 1 and _sys.argv[1] == "moe":
+        gen_moe_case()
+    else:
+        gen_model_cases()
+        gen_moe_case()
+        gen_ops()
+        gen_scheduler()

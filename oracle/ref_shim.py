@@ -108,3 +108,46 @@ def make_reference_model(config, attn_implementation="eager", **kwargs):
     config._attn_implementation = attn_implementation
     model = hf_models.GPTDolomiteForCausalLM(config, **kwargs)
     return model
+
+
+# MoE family (reference moe_dolomite/config.py adds four fields on top of
+# CommonConfig; same transformers-5 attribute re-injection applies).
+_MOE_EXTRA_DEFAULTS = dict(
+    num_experts=8,
+    num_experts_per_tok=2,
+    output_router_logits=False,
+    router_aux_loss_coef=0.001,
+)
+
+
+def make_reference_moe_config(**kwargs):
+    hf_models = import_reference_hf_models()
+
+    kwargs.setdefault("tie_word_embeddings", False)
+    config = hf_models.MoEDolomiteConfig(**kwargs)
+
+    fields = dict(_CONFIG_DEFAULTS)
+    fields.update(_MOE_EXTRA_DEFAULTS)
+    for k in fields:
+        if k in kwargs:
+            fields[k] = kwargs[k]
+
+    if fields["n_inner"] is None:
+        fields["n_inner"] = 4 * fields["n_embd"]
+    aht = fields["attention_head_type"]
+    if aht == "mha" and fields["num_key_value_heads"] is None:
+        fields["num_key_value_heads"] = fields["n_head"]
+    elif aht == "mqa" and fields["num_key_value_heads"] is None:
+        fields["num_key_value_heads"] = 1
+    fields["multi_query"] = aht == "mqa"
+
+    for k, v in fields.items():
+        object.__setattr__(config, k, v)
+
+    return config
+
+
+def make_reference_moe_model(config, attn_implementation="eager", **kwargs):
+    hf_models = import_reference_hf_models()
+    config._attn_implementation = attn_implementation
+    return hf_models.MoEDolomiteForCausalLM(config, moe_implementation="eager", **kwargs)
