@@ -161,11 +161,16 @@ def gen_moe_case():
     input_ids = torch.randint(0, cfg_kwargs["vocab_size"], (B, S), generator=torch.Generator().manual_seed(99))
     labels = input_ids.clone()
 
+    params = dict(model.named_parameters())
+
     out_plain = model(input_ids=input_ids, labels=labels, output_router_logits=False)
+    out_plain.loss.backward()
+    grads = {k: params[k].grad.detach().clone() for k in MOE_GRAD_KEYS if k in params}
+    model.zero_grad()
+
     out = model(input_ids=input_ids, labels=labels, output_router_logits=True)
     out.loss.backward()
-    params = dict(model.named_parameters())
-    grads = {k: params[k].grad.detach().clone() for k in MOE_GRAD_KEYS if k in params}
+    grads_with_aux = {k: params[k].grad.detach().clone() for k in MOE_GRAD_KEYS if k in params}
 
     torch.save(
         dict(
@@ -178,6 +183,7 @@ def gen_moe_case():
             loss_with_aux=out.loss.detach().clone(),
             aux_loss=out.aux_loss.detach().clone(),
             grads=grads,
+            grads_with_aux=grads_with_aux,
         ),
         GOLDEN_DIR / "model_moe_mqa_rope_rmsnorm_swiglu.pt",
     )
