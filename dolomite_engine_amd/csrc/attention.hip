@@ -102,8 +102,17 @@ __global__ void __launch_bounds__(512) fa_fwd_kernel(
     int64_t o_ts, int64_t T_total, float scale) {
     constexpr int KCH = DPAD / 32;   // contraction chunks for QK^T
     constexpr int DCH = DPAD / 16;   // output col blocks for PV
-    constexpr int SK = DPAD + 8;     // K LDS row stride (elems), +16B pad
-    constexpr int SV = 64 + 8;       // V^T / P LDS row stride
+    // Strides chosen with tools_lds_sim.py (exact gfx950 bank model):
+    //   row-major K image: stride DPAD+16 makes every b128 B-frag read
+    //     conflict-free (the b128 lane groups mix two lg half-rows offset by
+    //     4 dwords; +8 padding left them 2-way conflicted);
+    //   transposed V image: stride 96 + block swizzle (col>>3)^(3*(row>>3)&7)
+    //     -> conflict-free b128 reads;
+    //   P strips: stride 72 (b16 writes conflict-free; reads 2-way — the
+    //     reverse trade costs more write cycles than it saves).
+    constexpr int SK = DPAD + 16;    // K LDS row stride (elems)
+    constexpr int SV = 64 + 8;       // P strip row stride
+    constexpr int TST = 96;          // transposed V image row stride
 
     int tile_id = blockIdx.x, b = blockIdx.y, h = blockIdx.z;
     xcd_remap_tile_bh(tile_id, b, h);
@@ -125,13 +134,13 @@ __global__ void __launch_bounds__(512) fa_fwd_kernel(
 
     extern __shared__ char smem_raw[];
     __bf16* Klds = (__bf16*)smem_raw;              // [64][SK]
-    __bf16* Vlds = Klds + 64 * SK;                 // [DPAD][SV] (transposed, swizzled)
-    __bf16* Plds = Vlds + DPAD * SV;               // [8 waves][16][SV]
+    __bf16* Vlds = Klds + 64 * SK;                 // [DPAD][TST] (transposed, swizzled)
+    __bf16* Plds = Vlds + DPAD * TST;              // [8 waves][16][SV]
     __bf16* Pw = Plds + wave * 16 * SV;
 
     // XOR block swizzle for the transposed V image (see fa_bwd_kernel)
-#define VSWZ(row, col) ((row) * SV + (((((col) >> 3) ^ (((row) >> 3) & 7)) << 3) | ((col) & 7)))
-#define VSWZ8(row, col0) ((row) * SV + (((((col0) >> 3) ^ (((row) >> 3) & 7)) << 3)))
+#define VSWZ(row, col) ((row) * TST + (((((col) >> 3) ^ ((3 * ((row) >> 3)) & 7)) << 3) | ((col) & 7)))
+#define VSWZ8(row, col0) ((row) * TST + (((((col0) >> 3) ^ ((3 * ((row) >> 3)) & 7)) << 3)))
 
     const int kvh = h / G;
     const int64_t q_hoff = (int64_t)(h / G) * q_gs + (int64_t)(h % G) * D;
@@ -273,7 +282,7 @@ static int launch_fa_fwd(hipStream_t stream, const __bf16* q, const __bf16* k, c
                          int64_t q_ts, int64_t q_gs, int64_t k_ts, int64_t k_hs,
                          int64_t v_ts, int64_t v_hs, int max_tiles, float scale) {
     dim3 grid(max_tiles, batch, H), block(512);
-    size_t shmem = (64 * (DPAD + 8) + DPAD * 72 + 8 * 16 * 72) * sizeof(__bf16);
+    size_t shmem = (64 * (DPAD + 16) + DPAD * 96 + 8 * 16 * 72) * sizeof(__bf16);
     hipLaunchKernelGGL((fa_fwd_kernel<DPAD>), grid, block, shmem, stream,
                        q, k, v, o, lse, cu, H, Hkv, D, G,
                        q_ts, q_gs, k_ts, k_hs, v_ts, v_hs, (int64_t)H * D, T, scale);
@@ -365,7 +374,7 @@ __global__ void __launch_bounds__(512, 2) fa_bwd_dkv_kernel(
     int64_t do_ts, int64_t T_total, float scale) {
     constexpr int KCH = DPAD / 32;
     constexpr int DCH = DPAD / 16;
-    constexpr int ST = 64 + 8;
+    constexpr int ST = 64 + 8;                 // strip stride (see fa_fwd note)
 
     int tile_id = blockIdx.x, b = blockIdx.y, h = blockIdx.z;
     xcd_remap_tile_bh(tile_id, b, h);
@@ -381,17 +390,18 @@ __global__ void __launch_bounds__(512, 2) fa_bwd_dkv_kernel(
     const int lr = lane & 15;
     const int lg = lane >> 4;
 
-    constexpr int SQ = DPAD + 8;               // row-major image stride
+    constexpr int SQ = DPAD + 16;              // row-major image stride
+    constexpr int TST = 96;                    // transposed image stride
     extern __shared__ char smem_raw[];
-    __bf16* QTl = (__bf16*)smem_raw;           // [DPAD][ST]    (Q^T, swizzled)
-    __bf16* dOTl = QTl + DPAD * ST;            // [DPAD][ST]    (dO^T, swizzled)
-    __bf16* dSTl = dOTl + DPAD * ST;           // [128 key][ST] (dS^T, [key][q])
+    __bf16* QTl = (__bf16*)smem_raw;           // [DPAD][TST]   (Q^T, swizzled)
+    __bf16* dOTl = QTl + DPAD * TST;           // [DPAD][TST]   (dO^T, swizzled)
+    __bf16* dSTl = dOTl + DPAD * TST;          // [128 key][ST] (dS^T, [key][q])
     __bf16* PTl = dSTl + 128 * ST;             // [128 key][ST] (P^T, [key][q])
     __bf16* Qlds = PTl + 128 * ST;             // [64 q][SQ]    (row-major)
     __bf16* dOl = Qlds + 64 * SQ;              // [64 q][SQ]    (row-major)
 
-#define SWZ(row, col) ((row) * ST + (((((col) >> 3) ^ (((row) >> 3) & 7)) << 3) | ((col) & 7)))
-#define SWZ8(row, col0) ((row) * ST + (((((col0) >> 3) ^ (((row) >> 3) & 7)) << 3)))
+#define SWZ(row, col) ((row) * TST + (((((col) >> 3) ^ ((3 * ((row) >> 3)) & 7)) << 3) | ((col) & 7)))
+#define SWZ8(row, col0) ((row) * TST + (((((col0) >> 3) ^ ((3 * ((row) >> 3)) & 7)) << 3)))
 
     const int kend = min(L, ks + 128);
 
@@ -530,7 +540,7 @@ __global__ void __launch_bounds__(512) fa_bwd_dq_kernel(
     int64_t do_ts, int64_t T_total, float scale) {
     constexpr int KCH = DPAD / 32;
     constexpr int DCH = DPAD / 16;
-    constexpr int ST = 64 + 8;
+    constexpr int ST = 64 + 8;                 // strip stride
 
     int tile_id = blockIdx.x, b = blockIdx.y, h = blockIdx.z;
     xcd_remap_tile_bh(tile_id, b, h);
@@ -547,16 +557,17 @@ __global__ void __launch_bounds__(512) fa_bwd_dq_kernel(
     const int lr = lane & 15;
     const int lg = lane >> 4;
 
-    constexpr int SQ = DPAD + 8;
+    constexpr int SQ = DPAD + 16;              // row-major image stride
+    constexpr int TST = 96;                    // transposed image stride
     extern __shared__ char smem_raw[];
-    __bf16* KTl = (__bf16*)smem_raw;           // [DPAD][ST] (K^T, swizzled)
-    __bf16* Klds = KTl + DPAD * ST;            // [64 key][SQ] (row-major)
+    __bf16* KTl = (__bf16*)smem_raw;           // [DPAD][TST] (K^T, swizzled)
+    __bf16* Klds = KTl + DPAD * TST;           // [64 key][SQ] (row-major)
     __bf16* Vlds = Klds + 64 * SQ;             // [64 key][SQ] (row-major)
     __bf16* dSl = Vlds + 64 * SQ;              // [8 waves][16 q][ST] (dS strips)
     __bf16* dSw = dSl + wave * 16 * ST;
 
-#define SWZ(row, col) ((row) * ST + (((((col) >> 3) ^ (((row) >> 3) & 7)) << 3) | ((col) & 7)))
-#define SWZ8(row, col0) ((row) * ST + (((((col0) >> 3) ^ (((row) >> 3) & 7)) << 3)))
+#define SWZ(row, col) ((row) * TST + (((((col) >> 3) ^ ((3 * ((row) >> 3)) & 7)) << 3) | ((col) & 7)))
+#define SWZ8(row, col0) ((row) * TST + (((((col0) >> 3) ^ ((3 * ((row) >> 3)) & 7)) << 3)))
 
     const int64_t q_hoff = (int64_t)(h / G) * q_gs + (int64_t)(h % G) * D;
     const int64_t do_hoff = (int64_t)h * D;
@@ -683,14 +694,15 @@ static int launch_fa_bwd(hipStream_t stream, const __bf16* q, const __bf16* k, c
     constexpr int ST = 64 + 8;
     dim3 block(512);
     dim3 grid(max_tiles, batch, H);
-    constexpr int SQ = DPAD + 8;
-    size_t shmem_dkv = (size_t)(DPAD * ST * 2 + 128 * ST * 2 + 64 * SQ * 2) * sizeof(__bf16);
+    constexpr int SQ = DPAD + 16;
+    constexpr int TST = 96;
+    size_t shmem_dkv = (size_t)(DPAD * TST * 2 + 128 * ST * 2 + 64 * SQ * 2) * sizeof(__bf16);
     hipLaunchKernelGGL((fa_bwd_dkv_kernel<DPAD>), grid, block, shmem_dkv, stream,
                        q, k, v, dout, lse, delta, dk_acc, dv_acc, cu, H, Hkv, D, G,
                        q_ts, q_gs, k_ts, k_hs, v_ts, v_hs, do_ts, T, scale);
     int err = dol_last_error();
     if (err) return err;
-    size_t shmem_dq = (size_t)(DPAD * ST + 64 * SQ * 2 + 8 * 16 * ST) * sizeof(__bf16);
+    size_t shmem_dq = (size_t)(DPAD * TST + 64 * SQ * 2 + 8 * 16 * ST) * sizeof(__bf16);
     hipLaunchKernelGGL((fa_bwd_dq_kernel<DPAD>), grid, block, shmem_dq, stream,
                        q, k, v, dout, lse, delta, dqkv_q, cu, H, Hkv, D, G,
                        q_ts, q_gs, k_ts, k_hs, v_ts, v_hs, do_ts, T, scale);
