@@ -15,7 +15,7 @@ PKG = CSRC.parent
 REPO = PKG.parent
 SO_PATH = PKG / "libdolomite_hip.so"
 
-SOURCES = [CSRC / "elementwise.hip", CSRC / "attention.hip", CSRC / "data_index.cpp"]
+SOURCES = [CSRC / "elementwise.hip", CSRC / "attention.hip", CSRC / "moe_gemm.hip", CSRC / "data_index.cpp"]
 
 
 def _needs_rebuild() -> bool:

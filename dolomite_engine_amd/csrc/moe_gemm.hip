@@ -1,0 +1,296 @@
+// dolomite_hip — grouped expert GEMMs for the MoE family (gfx950).
+//
+// Replaces the reference's scattermoe/triton grouped path
+// (hf_models/models/moe_dolomite/moe/scatter.py:109-138) for the expert
+// matmuls of SparseMoE (moe/base.py:137-156): rows are expert-sorted by the
+// router (bincount/sort in Python), an (E+1) offsets array delimits each
+// expert's row group, and one kernel launch computes every expert's GEMM.
+//
+//   fwd   : Y[t,n]   = sum_k  X[t,k]  * W[e(t),n,k]   (+ bias[e,n])
+//   dgrad : dX[t,k]  = sum_n dY[t,n]  * W[e(t),n,k]
+//   wgrad : dW[e,n,k]= sum_t dY[t,n]  * X[t,k]        (t in expert e's group)
+//
+// bf16 in / fp32 MFMA accumulate / bf16 out. K % 8 == 0 required (the
+// Python layer falls back to the eager per-expert loop otherwise).
+//
+// Tiling: 64x64 output tile per 4-wave workgroup; v_mfma_f32_16x16x32_bf16;
+// contraction staged through LDS in 64-wide chunks. LDS strides follow the
+// bank model in tools_lds_sim.py (same rules as attention.hip): row-major
+// images stride cols+16 (conflict-free b128 A/B-frag reads), transposed
+// images stride 96 with the (col>>3)^(3*(row>>3)&7) block swizzle.
+
+#include "common.h"
+
+typedef __bf16 bf16x8m __attribute__((ext_vector_type(8)));
+typedef float f32x4m __attribute__((ext_vector_type(4)));
+#define MFMA16M(a, b, c) __builtin_amdgcn_mfma_f32_16x16x32_bf16((a), (b), (c), 0, 0, 0)
+
+// rm stride for 64-col images; TST/swizzle for transposed images
+#define MOE_RS 80
+#define MOE_TS 96
+#define MSWZ(row, col) ((row) * MOE_TS + (((((col) >> 3) ^ ((3 * ((row) >> 3)) & 7)) << 3) | ((col) & 7)))
+#define MSWZ8(row, col0) ((row) * MOE_TS + (((((col0) >> 3) ^ ((3 * ((row) >> 3)) & 7)) << 3)))
+
+__device__ __forceinline__ bf16x8m moe_load8(const __bf16* p, bool valid) {
+    bf16x8m r = *(const bf16x8m*)(valid ? p : p);  // caller clamps pointer
+    if (!valid)
+#pragma unroll
+        for (int e = 0; e < 8; ++e) r[e] = (__bf16)0.f;
+    return r;
+}
+
+// ---------------------------------------------------------------------------
+// fwd: grid(x = m-tile, y = expert, z = n-tile), 256 threads.
+//   A = X rows (row-major LDS image), B = W[e] rows (row-major LDS image:
+//   B[c=k][j=n] wants image [j=n][c=k] = W's own row-major layout).
+// ---------------------------------------------------------------------------
+__global__ void __launch_bounds__(256) moe_gemm_fwd_kernel(
+    const __bf16* __restrict__ x, const __bf16* __restrict__ w, const __bf16* __restrict__ bias,
+    __bf16* __restrict__ y, const int32_t* __restrict__ offsets,
+    int E, int N, int K) {
+    const int e = blockIdx.y;
+    const int r0 = offsets[e], r1 = offsets[e + 1];
+    const int m0 = r0 + blockIdx.x * 64;
+    if (m0 >= r1) return;
+    const int n0 = blockIdx.z * 64;
+
+    const int lane = threadIdx.x & 63;
+    const int wave = threadIdx.x >> 6;
+    const int lr = lane & 15;
+    const int lg = lane >> 4;
+
+    extern __shared__ char smem_raw[];
+    __bf16* Xl = (__bf16*)smem_raw;      // [64 m][MOE_RS]
+    __bf16* Wl = Xl + 64 * MOE_RS;       // [64 n][MOE_RS]
+
+    const __bf16* we = w + (int64_t)e * N * K;
+
+    f32x4m acc[4];
+#pragma unroll
+    for (int cb = 0; cb < 4; ++cb) acc[cb] = {0.f, 0.f, 0.f, 0.f};
+
+    for (int k0 = 0; k0 < K; k0 += 64) {
+        __syncthreads();
+        // stage: 64 rows x 64 k of X and W (8-elem pieces, 512 of them)
+        for (int pidx = threadIdx.x; pidx < 512; pidx += 256) {
+            const int row = pidx / 8;
+            const int kk = (pidx % 8) * 8;
+            const bool kin = (k0 + kk) < K;
+            {
+                const bool v = kin && (m0 + row) < r1;
+                const __bf16* p = x + (int64_t)(v ? m0 + row : r0) * K + (v ? k0 + kk : 0);
+                *(bf16x8m*)&Xl[row * MOE_RS + kk] = moe_load8(p, v);
+            }
+            {
+                const bool v = kin && (n0 + row) < N;
+                const __bf16* p = we + (int64_t)(v ? n0 + row : 0) * K + (v ? k0 + kk : 0);
+                *(bf16x8m*)&Wl[row * MOE_RS + kk] = moe_load8(p, v);
+            }
+        }
+        __syncthreads();
+#pragma unroll
+        for (int kc = 0; kc < 2; ++kc) {
+            bf16x8m af = *(const bf16x8m*)&Xl[(wave * 16 + lr) * MOE_RS + kc * 32 + lg * 8];
+#pragma unroll
+            for (int cb = 0; cb < 4; ++cb) {
+                bf16x8m bf = *(const bf16x8m*)&Wl[(cb * 16 + lr) * MOE_RS + kc * 32 + lg * 8];
+                acc[cb] = MFMA16M(af, bf, acc[cb]);
+            }
+        }
+    }
+
+#pragma unroll
+    for (int cb = 0; cb < 4; ++cb)
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+            const int m = m0 + wave * 16 + lg * 4 + r;
+            const int n = n0 + cb * 16 + lr;
+            if (m < r1 && n < N) {
+                float v = acc[cb][r];
+                if (bias) v += (float)bias[(int64_t)e * N + n];
+                y[(int64_t)m * N + n] = (__bf16)v;
+            }
+        }
+}
+
+// ---------------------------------------------------------------------------
+// dgrad: dX[t,k] = sum_n dY[t,n] * W[e,n,k].
+//   A = dY rows (row-major image), B[c=n][j=k] wants image [k][n] =
+//   transposed W tile (swizzled scatter on stage).
+// grid(x = m-tile, y = expert, z = k-tile)
+// ---------------------------------------------------------------------------
+__global__ void __launch_bounds__(256) moe_gemm_dgrad_kernel(
+    const __bf16* __restrict__ dy, const __bf16* __restrict__ w, __bf16* __restrict__ dx,
+    const int32_t* __restrict__ offsets, int E, int N, int K) {
+    const int e = blockIdx.y;
+    const int r0 = offsets[e], r1 = offsets[e + 1];
+    const int m0 = r0 + blockIdx.x * 64;
+    if (m0 >= r1) return;
+    const int k0 = blockIdx.z * 64;
+
+    const int lane = threadIdx.x & 63;
+    const int wave = threadIdx.x >> 6;
+    const int lr = lane & 15;
+    const int lg = lane >> 4;
+
+    extern __shared__ char smem_raw[];
+    __bf16* Dl = (__bf16*)smem_raw;      // [64 t][MOE_RS]  (dY chunk, row-major)
+    __bf16* WT = Dl + 64 * MOE_RS;       // [64 k][MOE_TS]  (W^T image, swizzled)
+
+    const __bf16* we = w + (int64_t)e * N * K;
+
+    f32x4m acc[4];
+#pragma unroll
+    for (int cb = 0; cb < 4; ++cb) acc[cb] = {0.f, 0.f, 0.f, 0.f};
+
+    for (int nn0 = 0; nn0 < N; nn0 += 64) {
+        __syncthreads();
+        for (int pidx = threadIdx.x; pidx < 512; pidx += 256) {
+            const int row = pidx / 8;
+            const int c8 = (pidx % 8) * 8;
+            {
+                const bool v = (m0 + row) < r1 && (nn0 + c8) < N;
+                const __bf16* p = dy + (int64_t)(v ? m0 + row : r0) * N + (v ? nn0 + c8 : 0);
+                *(bf16x8m*)&Dl[row * MOE_RS + c8] = moe_load8(p, v);
+            }
+            {
+                // W rows n = nn0+row, k chunk c8: scatter to [k][n] image
+                const bool v = (nn0 + row) < N && (k0 + c8) < K;
+                const __bf16* p = we + (int64_t)(v ? nn0 + row : 0) * K + (v ? k0 + c8 : 0);
+                bf16x8m w8 = moe_load8(p, v);
+#pragma unroll
+                for (int t = 0; t < 8; ++t) WT[MSWZ(c8 + t, row)] = w8[t];
+            }
+        }
+        __syncthreads();
+#pragma unroll
+        for (int nc = 0; nc < 2; ++nc) {
+            bf16x8m af = *(const bf16x8m*)&Dl[(wave * 16 + lr) * MOE_RS + nc * 32 + lg * 8];
+#pragma unroll
+            for (int cb = 0; cb < 4; ++cb) {
+                bf16x8m bf = *(const bf16x8m*)&WT[MSWZ8(cb * 16 + lr, nc * 32 + lg * 8)];
+                acc[cb] = MFMA16M(af, bf, acc[cb]);
+            }
+        }
+    }
+
+#pragma unroll
+    for (int cb = 0; cb < 4; ++cb)
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+            const int m = m0 + wave * 16 + lg * 4 + r;
+            const int k = k0 + cb * 16 + lr;
+            if (m < r1 && k < K) dx[(int64_t)m * K + k] = (__bf16)acc[cb][r];
+        }
+}
+
+// ---------------------------------------------------------------------------
+// wgrad: dW[e,n,k] = sum_t dY[t,n] * X[t,k] over expert e's rows.
+//   A[i=n][c=t] -> dY^T image (swizzled), B[c=t][j=k] -> X^T image [k][t]
+//   (swizzled). grid(x = n-tile, y = expert, z = k-tile); loop over t.
+// ---------------------------------------------------------------------------
+__global__ void __launch_bounds__(256) moe_gemm_wgrad_kernel(
+    const __bf16* __restrict__ dy, const __bf16* __restrict__ x, __bf16* __restrict__ dw,
+    const int32_t* __restrict__ offsets, int E, int N, int K) {
+    const int e = blockIdx.y;
+    const int r0 = offsets[e], r1 = offsets[e + 1];
+    const int n0 = blockIdx.x * 64;
+    const int k0 = blockIdx.z * 64;
+    if (n0 >= N) return;
+
+    const int lane = threadIdx.x & 63;
+    const int wave = threadIdx.x >> 6;
+    const int lr = lane & 15;
+    const int lg = lane >> 4;
+
+    extern __shared__ char smem_raw[];
+    __bf16* DT = (__bf16*)smem_raw;      // [64 n][MOE_TS] (dY^T image, swizzled)
+    __bf16* XT = DT + 64 * MOE_TS;       // [64 k][MOE_TS] (X^T image, swizzled)
+
+    f32x4m acc[4];
+#pragma unroll
+    for (int cb = 0; cb < 4; ++cb) acc[cb] = {0.f, 0.f, 0.f, 0.f};
+
+    for (int t0 = r0; t0 < r1; t0 += 64) {
+        __syncthreads();
+        for (int pidx = threadIdx.x; pidx < 512; pidx += 256) {
+            const int trow = pidx / 8;   // t offset within the chunk
+            const int c8 = (pidx % 8) * 8;
+            const bool tin = (t0 + trow) < r1;
+            {
+                const bool v = tin && (n0 + c8) < N;
+                const __bf16* p = dy + (int64_t)(v ? t0 + trow : r0) * N + (v ? n0 + c8 : 0);
+                bf16x8m d8 = moe_load8(p, v);
+#pragma unroll
+                for (int t = 0; t < 8; ++t) DT[MSWZ(c8 + t, trow)] = d8[t];
+            }
+            {
+                const bool v = tin && (k0 + c8) < K;
+                const __bf16* p = x + (int64_t)(v ? t0 + trow : r0) * K + (v ? k0 + c8 : 0);
+                bf16x8m x8 = moe_load8(p, v);
+#pragma unroll
+                for (int t = 0; t < 8; ++t) XT[MSWZ(c8 + t, trow)] = x8[t];
+            }
+        }
+        __syncthreads();
+#pragma unroll
+        for (int tc = 0; tc < 2; ++tc) {
+            bf16x8m af = *(const bf16x8m*)&DT[MSWZ8(wave * 16 + lr, tc * 32 + lg * 8)];
+#pragma unroll
+            for (int cb = 0; cb < 4; ++cb) {
+                bf16x8m bf = *(const bf16x8m*)&XT[MSWZ8(cb * 16 + lr, tc * 32 + lg * 8)];
+                acc[cb] = MFMA16M(af, bf, acc[cb]);
+            }
+        }
+    }
+
+#pragma unroll
+    for (int cb = 0; cb < 4; ++cb)
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+            const int n = n0 + wave * 16 + lg * 4 + r;
+            const int k = k0 + cb * 16 + lr;
+            if (n < N && k < K)
+                dw[((int64_t)e * N + n) * K + k] = (__bf16)acc[cb][r];
+        }
+}
+
+// ---------------------------------------------------------------------------
+// C-ABI
+// ---------------------------------------------------------------------------
+extern "C" int dolomite_moe_gemm_fwd(dolomite_stream_t stream, const void* x, const void* w,
+                                     const void* bias, void* y, const int32_t* offsets,
+                                     int E, int max_rows, int N, int K, int dtype) {
+    if (dtype != DOLOMITE_BF16) return 9010;
+    if (K % 8 != 0 || N % 8 != 0) return 9020;
+    dim3 grid((max_rows + 63) / 64, E, (N + 63) / 64), block(256);
+    size_t shmem = (size_t)(64 * MOE_RS * 2) * sizeof(__bf16);
+    hipLaunchKernelGGL(moe_gemm_fwd_kernel, grid, block, shmem, (hipStream_t)stream,
+                       (const __bf16*)x, (const __bf16*)w, (const __bf16*)bias, (__bf16*)y,
+                       offsets, E, N, K);
+    return dol_last_error();
+}
+
+extern "C" int dolomite_moe_gemm_dgrad(dolomite_stream_t stream, const void* dy, const void* w,
+                                       void* dx, const int32_t* offsets,
+                                       int E, int max_rows, int N, int K, int dtype) {
+    if (dtype != DOLOMITE_BF16) return 9010;
+    if (K % 8 != 0 || N % 8 != 0) return 9020;
+    dim3 grid((max_rows + 63) / 64, E, (K + 63) / 64), block(256);
+    size_t shmem = (size_t)(64 * MOE_RS + 64 * MOE_TS) * sizeof(__bf16);
+    hipLaunchKernelGGL(moe_gemm_dgrad_kernel, grid, block, shmem, (hipStream_t)stream,
+                       (const __bf16*)dy, (const __bf16*)w, (__bf16*)dx, offsets, E, N, K);
+    return dol_last_error();
+}
+
+extern "C" int dolomite_moe_gemm_wgrad(dolomite_stream_t stream, const void* dy, const void* x,
+                                       void* dw, const int32_t* offsets,
+                                       int E, int N, int K, int dtype) {
+    if (dtype != DOLOMITE_BF16) return 9010;
+    if (K % 8 != 0 || N % 8 != 0) return 9020;
+    dim3 grid((N + 63) / 64, E, (K + 63) / 64), block(256);
+    size_t shmem = (size_t)(64 * MOE_TS * 2) * sizeof(__bf16);
+    hipLaunchKernelGGL(moe_gemm_wgrad_kernel, grid, block, shmem, (hipStream_t)stream,
+                       (const __bf16*)dy, (const __bf16*)x, (__bf16*)dw, offsets, E, N, K);
+    return dol_last_error();
+}

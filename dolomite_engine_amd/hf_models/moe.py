@@ -67,6 +67,15 @@ class ParameterizedExperts(nn.Module):
             self.bias.zero_()
 
     def forward(self, input: torch.Tensor, num_tokens_per_expert: torch.Tensor) -> torch.Tensor:
+        if input.is_cuda:
+            # one grouped-GEMM HIP launch over all experts (csrc/moe_gemm.hip,
+            # the scattermoe-path replacement); None only for unsupported
+            # dtype/dims, where the eager per-expert loop below still applies
+            from ..ops import grouped_expert_gemm
+
+            y = grouped_expert_gemm(input, self.weight, self.bias, num_tokens_per_expert)
+            if y is not None:
+                return y
         pieces = input.split(num_tokens_per_expert.tolist(), dim=0)
         outs = [
             F.linear(pieces[i], self.weight[i], None if self.bias is None else self.bias[i])

@@ -6,6 +6,7 @@ from .functional import (
     fused_cross_entropy,
     fused_layernorm,
     fused_rmsnorm,
+    grouped_expert_gemm,
     rope_packed_qkv,
     varlen_attention,
 )

@@ -201,6 +201,29 @@ int dolomite_scale_inplace(dolomite_stream_t stream, void* buf, int64_t n,
                            float scale, int dtype);
 
 /* ------------------------------------------------------------------------
+ * Grouped expert GEMMs for the MoE family — replaces the reference's
+ * scattermoe/triton grouped path (moe_dolomite/moe/scatter.py:109-138) for
+ * the SparseMoE expert matmuls (moe/base.py:137-156). Rows are
+ * expert-sorted; offsets is the (E+1) int32 cumulative row count per
+ * expert (device pointer). bf16 in / fp32 accumulate / bf16 out.
+ * Requires K % 8 == 0 and N % 8 == 0 (error 9020 otherwise; the Python
+ * layer falls back to the eager per-expert loop).
+ *   fwd   : y[t,n]    = sum_k x[t,k] * w[e(t),n,k] (+ bias[e,n], opt NULL)
+ *   dgrad : dx[t,k]   = sum_n dy[t,n] * w[e(t),n,k]
+ *   wgrad : dw[e,n,k] = sum_{t in group e} dy[t,n] * x[t,k]
+ * max_rows = largest group size (host int, for grid sizing).
+ * ---------------------------------------------------------------------- */
+int dolomite_moe_gemm_fwd(dolomite_stream_t stream, const void* x, const void* w,
+                          const void* bias, void* y, const int32_t* offsets,
+                          int E, int max_rows, int N, int K, int dtype);
+int dolomite_moe_gemm_dgrad(dolomite_stream_t stream, const void* dy, const void* w,
+                            void* dx, const int32_t* offsets,
+                            int E, int max_rows, int N, int K, int dtype);
+int dolomite_moe_gemm_wgrad(dolomite_stream_t stream, const void* dy, const void* x,
+                            void* dw, const int32_t* offsets,
+                            int E, int N, int K, int dtype);
+
+/* ------------------------------------------------------------------------
  * Host-side dataset index builders (CPU; the reference's only native
  * component, data/megatron/utils/helpers.cpp).
  * build_sample_idx: pack epoch-replicated documents into seq_length+1 token

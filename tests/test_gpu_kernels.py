@@ -250,3 +250,78 @@ def test_adamw_matches_torch_on_gpu(golden_dir):
     torch.testing.assert_close(m.cpu(), fx["exp_avg"], rtol=1e-5, atol=1e-7)
     torch.testing.assert_close(v.cpu(), fx["exp_avg_sq"], rtol=1e-5, atol=1e-7)
     torch.testing.assert_close(pout.cpu(), fx["p_final"].to(torch.bfloat16), rtol=0, atol=0)
+
+
+class TestMoEGroupedGemm:
+    """Grouped expert GEMM (csrc/moe_gemm.hip) vs the eager per-expert loop
+    (reference moe/base.py:12-50 semantics) — ragged groups, empty groups,
+    bias, fwd + both grads."""
+
+    @pytest.mark.parametrize(
+        "E,N,K,counts,bias",
+        [
+            (4, 128, 64, [37, 0, 91, 12], True),
+            (8, 512, 256, [64, 128, 1, 0, 200, 17, 63, 39], False),
+            (2, 72, 96, [130, 70], True),   # N not multiple of 64
+            (4, 256, 520, [65, 64, 63, 8], False),  # K % 64 != 0 (tail chunk)
+        ],
+    )
+    def test_matches_eager(self, E, N, K, counts, bias):
+        torch.manual_seed(17)
+        T = sum(counts)
+        x = (torch.randn(T, K) * 0.5).bfloat16().cuda().requires_grad_(True)
+        w = (torch.randn(E, N, K) * 0.05).bfloat16().cuda().requires_grad_(True)
+        b = (torch.randn(E, N) * 0.1).bfloat16().cuda().requires_grad_(True) if bias else None
+        num = torch.tensor(counts, device="cuda")
+
+        from dolomite_engine_amd.ops import grouped_expert_gemm
+
+        y = grouped_expert_gemm(x, w, b, num)
+        assert y is not None, "HIP grouped path did not engage"
+        dy = torch.randn_like(y) * 0.3
+        y.backward(dy)
+
+        # eager reference (fp32 accumulated per-expert linear)
+        xr = x.detach().float().requires_grad_(True)
+        wr = w.detach().float().requires_grad_(True)
+        br = b.detach().float().requires_grad_(True) if bias else None
+        pieces = xr.split(counts, dim=0)
+        ref = torch.cat(
+            [torch.nn.functional.linear(pieces[i], wr[i], None if br is None else br[i]) for i in range(E)]
+        )
+        ref.backward(dy.float())
+
+        torch.testing.assert_close(y.float(), ref.detach(), rtol=2e-2, atol=2e-2)
+        torch.testing.assert_close(x.grad.float(), xr.grad, rtol=3e-2, atol=3e-2)
+        torch.testing.assert_close(w.grad.float(), wr.grad, rtol=3e-2, atol=3e-2)
+        if bias:
+            torch.testing.assert_close(b.grad.float(), br.grad, rtol=3e-2, atol=3e-2)
+
+    def test_moe_model_grouped_path_runs(self, golden_dir):
+        """Full MoE model on GPU routes experts through the grouped kernel
+        (moe_gemm_fwd appears in the op profile)."""
+        from dolomite_engine_amd.ops import hip as hip_mod
+        from tests.test_moe_cpu import FIXTURE, build_model as build_moe
+
+        p = golden_dir / FIXTURE
+        if not p.exists():
+            pytest.skip("moe golden missing")
+        fx = torch.load(p, weights_only=False)
+        model = build_moe(fx, "flash_attention_2", padding_free=True, dtype=torch.bfloat16).cuda()
+        B, S = fx["input_ids"].shape
+        hip_mod.enable_profiling()
+        try:
+            out = model(
+                input_ids=fx["input_ids"].reshape(-1).cuda(),
+                position_ids=torch.arange(S).repeat(B).cuda(),
+                cu_seqlens=torch.arange(0, B * S + 1, S, dtype=torch.int32).cuda(),
+                max_seqlen=S,
+                labels=fx["labels"].reshape(-1).cuda(),
+            )
+            out.loss.backward()
+            torch.cuda.synchronize()
+            prof = hip_mod.collect_profile()
+        finally:
+            hip_mod.disable_profiling()
+        assert any("moe_gemm_fwd" in k for k in prof), sorted(prof)
+        torch.testing.assert_close(out.loss.float().cpu(), fx["loss"], rtol=2e-2, atol=2e-2)
