@@ -38,6 +38,128 @@ typedef float f32x4 __attribute__((ext_vector_type(4)));
 #define DOL_LOG2E 1.44269504088896340736f
 #define DOL_LN2 0.69314718055994530942f
 
+// ---------------------------------------------------------------------------
+// ds_read_b64_tr_b16 — gfx950 hardware transpose read (semantics pinned on
+// HW by dolomite_tr16_probe): each lane reads 64b at its own 8B-aligned LDS
+// address; within a 16-lane group, output lane l slot j receives element
+// (l&3) of the word read by lane (l&48) + 4j + ((l>>2)&3).
+//
+// Feeding an MFMA B-fragment B[k=(l>>4)*8+e][j=l&15] for a 32-row
+// contraction chunk at `rowbase`, 16-col block at `d0`, from a ROW-MAJOR
+// image with stride S: lane l passes the address of
+//   row = rowbase + 8*(l>>4) + 4h + ((l>>2)&3),  col = d0 + 4*(l&3)
+// and read h delivers fragment elements e = 4h..4h+3.
+//
+// Conflicts (tools_lds_sim.py, validated against SQ_LDS_BANK_CONFLICT):
+// with S=112 the natural row order 2-way conflicts (rows 0 and 8 alias);
+// storing image rows with bits 2 and 3 SWAPPED makes both the tr reads and
+// the b128 row-major B-frag reads conflict-free. All images consumed by tr
+// reads therefore use PI23 row placement (writers and readers alike).
+// ---------------------------------------------------------------------------
+#define PI23(q) (((q) & ~12) | ((((q) >> 2) & 1) << 3) | ((((q) >> 3) & 1) << 2))
+
+typedef __bf16 bf16x4 __attribute__((ext_vector_type(4)));
+
+// The reads and their waitcnt MUST live in one asm block: the compiler
+// assumes an asm statement completes synchronously, so with a separate wait
+// it is free to copy/shuffle the "ready" result registers before the data
+// actually lands (observed as silent corruption of one of two fragments).
+// Earlyclobber outputs keep the dest registers disjoint from the address
+// operands of later reads in the same block.
+
+// B-fragment from a PI23 row-major image: rowbase = 32-row chunk start
+// (multiple of 32), d0 = 16-col block start.
+__device__ __forceinline__ bf16x8 lds_tr16_bfrag(const __bf16* img, int rowbase, int S, int d0, int lane) {
+    const int g8 = (lane >> 4) * 8;
+    const int jj = (lane >> 2) & 3;
+    const int cc = (lane & 3) * 4;
+    unsigned a0 = (unsigned)(size_t)(img + PI23(rowbase + g8 + jj) * S + d0 + cc);
+    unsigned a1 = (unsigned)(size_t)(img + PI23(rowbase + g8 + 4 + jj) * S + d0 + cc);
+    bf16x4 lo, hi;
+    asm volatile(
+        "ds_read_b64_tr_b16 %0, %2\n\t"
+        "ds_read_b64_tr_b16 %1, %3\n\t"
+        "s_waitcnt lgkmcnt(0)"
+        : "=&v"(lo), "=&v"(hi)
+        : "v"(a0), "v"(a1)
+        : "memory");
+    bf16x8 out;
+#pragma unroll
+    for (int t = 0; t < 4; ++t) { out[t] = lo[t]; out[4 + t] = hi[t]; }
+    return out;
+}
+
+// Two B-fragments (same chunk, two images) in one asm block — the four
+// reads pipeline in the LDS queue behind a single drain.
+__device__ __forceinline__ void lds_tr16_bfrag2(const __bf16* imgA, const __bf16* imgB, int rowbase,
+                                                int S, int d0, int lane, bf16x8* fA, bf16x8* fB) {
+    const int g8 = (lane >> 4) * 8;
+    const int jj = (lane >> 2) & 3;
+    const int cc = (lane & 3) * 4;
+    const int off0 = PI23(rowbase + g8 + jj) * S + d0 + cc;
+    const int off1 = PI23(rowbase + g8 + 4 + jj) * S + d0 + cc;
+    unsigned a0 = (unsigned)(size_t)(imgA + off0);
+    unsigned a1 = (unsigned)(size_t)(imgA + off1);
+    unsigned b0 = (unsigned)(size_t)(imgB + off0);
+    unsigned b1 = (unsigned)(size_t)(imgB + off1);
+    bf16x4 alo, ahi, blo, bhi;
+    asm volatile(
+        "ds_read_b64_tr_b16 %0, %4\n\t"
+        "ds_read_b64_tr_b16 %1, %5\n\t"
+        "ds_read_b64_tr_b16 %2, %6\n\t"
+        "ds_read_b64_tr_b16 %3, %7\n\t"
+        "s_waitcnt lgkmcnt(0)"
+        : "=&v"(alo), "=&v"(ahi), "=&v"(blo), "=&v"(bhi)
+        : "v"(a0), "v"(a1), "v"(b0), "v"(b1)
+        : "memory");
+#pragma unroll
+    for (int t = 0; t < 4; ++t) {
+        (*fA)[t] = alo[t]; (*fA)[4 + t] = ahi[t];
+        (*fB)[t] = blo[t]; (*fB)[4 + t] = bhi[t];
+    }
+}
+
+// Four B-fragments (two col-blocks x two images) behind ONE drain — halves
+// the waitcnt stalls of the dkv inner loop vs per-dc drains.
+__device__ __forceinline__ void lds_tr16_bfrag4(const __bf16* imgA, const __bf16* imgB, int rowbase,
+                                                int S, int d0a, int d0b, int lane,
+                                                bf16x8* fA0, bf16x8* fB0, bf16x8* fA1, bf16x8* fB1) {
+    const int g8 = (lane >> 4) * 8;
+    const int jj = (lane >> 2) & 3;
+    const int cc = (lane & 3) * 4;
+    const int r0 = PI23(rowbase + g8 + jj) * S + cc;
+    const int r1 = PI23(rowbase + g8 + 4 + jj) * S + cc;
+    unsigned a0 = (unsigned)(size_t)(imgA + r0 + d0a);
+    unsigned a1 = (unsigned)(size_t)(imgA + r1 + d0a);
+    unsigned b0 = (unsigned)(size_t)(imgB + r0 + d0a);
+    unsigned b1 = (unsigned)(size_t)(imgB + r1 + d0a);
+    unsigned a2 = (unsigned)(size_t)(imgA + r0 + d0b);
+    unsigned a3 = (unsigned)(size_t)(imgA + r1 + d0b);
+    unsigned b2 = (unsigned)(size_t)(imgB + r0 + d0b);
+    unsigned b3 = (unsigned)(size_t)(imgB + r1 + d0b);
+    bf16x4 o0, o1, o2, o3, o4, o5, o6, o7;
+    asm volatile(
+        "ds_read_b64_tr_b16 %0, %8\n\t"
+        "ds_read_b64_tr_b16 %1, %9\n\t"
+        "ds_read_b64_tr_b16 %2, %10\n\t"
+        "ds_read_b64_tr_b16 %3, %11\n\t"
+        "ds_read_b64_tr_b16 %4, %12\n\t"
+        "ds_read_b64_tr_b16 %5, %13\n\t"
+        "ds_read_b64_tr_b16 %6, %14\n\t"
+        "ds_read_b64_tr_b16 %7, %15\n\t"
+        "s_waitcnt lgkmcnt(0)"
+        : "=&v"(o0), "=&v"(o1), "=&v"(o2), "=&v"(o3), "=&v"(o4), "=&v"(o5), "=&v"(o6), "=&v"(o7)
+        : "v"(a0), "v"(a1), "v"(b0), "v"(b1), "v"(a2), "v"(a3), "v"(b2), "v"(b3)
+        : "memory");
+#pragma unroll
+    for (int t = 0; t < 4; ++t) {
+        (*fA0)[t] = o0[t]; (*fA0)[4 + t] = o1[t];
+        (*fB0)[t] = o2[t]; (*fB0)[4 + t] = o3[t];
+        (*fA1)[t] = o4[t]; (*fA1)[4 + t] = o5[t];
+        (*fB1)[t] = o6[t]; (*fB1)[4 + t] = o7[t];
+    }
+}
+
 // Branchless guarded 16B load. Requires D % 8 == 0 (every head dim on this
 // path): a chunk is then fully inside [0, D) or fully in the pad, so the
 // guard reduces to ONE wave-divergent-free vector load from a clamped
@@ -110,9 +232,8 @@ __global__ void __launch_bounds__(512) fa_fwd_kernel(
     //     -> conflict-free b128 reads;
     //   P strips: stride 72 (b16 writes conflict-free; reads 2-way — the
     //     reverse trade costs more write cycles than it saves).
-    constexpr int SK = DPAD + 16;    // K LDS row stride (elems)
+    constexpr int SK = DPAD + 16;    // K/V LDS row stride (elems)
     constexpr int SV = 64 + 8;       // P strip row stride
-    constexpr int TST = 96;          // transposed V image row stride
 
     int tile_id = blockIdx.x, b = blockIdx.y, h = blockIdx.z;
     xcd_remap_tile_bh(tile_id, b, h);
@@ -134,13 +255,11 @@ __global__ void __launch_bounds__(512) fa_fwd_kernel(
 
     extern __shared__ char smem_raw[];
     __bf16* Klds = (__bf16*)smem_raw;              // [64][SK]
-    __bf16* Vlds = Klds + 64 * SK;                 // [DPAD][TST] (transposed, swizzled)
-    __bf16* Plds = Vlds + DPAD * TST;              // [8 waves][16][SV]
+    __bf16* Vlds = Klds + 64 * SK;                 // [64 key][SK] (row-major, PI23 rows)
+    __bf16* Plds = Vlds + 64 * SK;                 // [8 waves][16][SV]
     __bf16* Pw = Plds + wave * 16 * SV;
-
-    // XOR block swizzle for the transposed V image (see fa_bwd_kernel)
-#define VSWZ(row, col) ((row) * TST + (((((col) >> 3) ^ ((3 * ((row) >> 3)) & 7)) << 3) | ((col) & 7)))
-#define VSWZ8(row, col0) ((row) * TST + (((((col0) >> 3) ^ ((3 * ((row) >> 3)) & 7)) << 3)))
+    // V is stored row-major (coalesced staging) and consumed as the PV
+    // B-fragment via ds_read_b64_tr_b16 — no transposed image, no scatter.
 
     const int kvh = h / G;
     const int64_t q_hoff = (int64_t)(h / G) * q_gs + (int64_t)(h % G) * D;
@@ -179,9 +298,7 @@ __global__ void __launch_bounds__(512) fa_fwd_kernel(
                 bf16x8 kk = load_bf16x8_guard(kp, d0, D, kv_valid);
                 *(bf16x8*)&Klds[key * SK + d0] = kk;
                 const __bf16* vp = v + (int64_t)(s0 + (kv_valid ? ks + key : 0)) * v_ts + (int64_t)kvh * v_hs + d0;
-                bf16x8 vv = load_bf16x8_guard(vp, d0, D, kv_valid);
-#pragma unroll
-                for (int e = 0; e < 8; ++e) Vlds[VSWZ(d0 + e, key)] = vv[e];
+                *(bf16x8*)&Vlds[PI23(key) * SK + d0] = load_bf16x8_guard(vp, d0, D, kv_valid);
             }
         }
         __syncthreads();
@@ -239,13 +356,13 @@ __global__ void __launch_bounds__(512) fa_fwd_kernel(
 #pragma unroll
             for (int r = 0; r < 4; ++r) Pw[(lg * 4 + r) * SV + cb * 16 + lr] = (__bf16)p_val[cb][r];
 
-        // --- PV: A = P (this wave's rows), B = V^T image ---
+        // --- PV: A = P (this wave's rows), B via tr16 reads of row-major V ---
 #pragma unroll
         for (int kc2 = 0; kc2 < 2; ++kc2) {
             bf16x8 pf = *(const bf16x8*)&Pw[lr * SV + kc2 * 32 + lg * 8];
 #pragma unroll
             for (int dc = 0; dc < DCH; ++dc) {
-                bf16x8 vf = *(const bf16x8*)&Vlds[VSWZ8(dc * 16 + lr, kc2 * 32 + lg * 8)];
+                bf16x8 vf = lds_tr16_bfrag(Vlds, kc2 * 32, SK, dc * 16, lane);
                 o_acc[dc] = MFMA16(pf, vf, o_acc[dc]);
             }
         }
@@ -271,8 +388,6 @@ __global__ void __launch_bounds__(512) fa_fwd_kernel(
                 o[(int64_t)(s0 + qpos) * o_ts + (int64_t)h * D + d] = (__bf16)(o_acc[dc][r] * inv_l[r]);
         }
     }
-#undef VSWZ
-#undef VSWZ8
 }
 
 template <int DPAD>
@@ -282,7 +397,7 @@ static int launch_fa_fwd(hipStream_t stream, const __bf16* q, const __bf16* k, c
                          int64_t q_ts, int64_t q_gs, int64_t k_ts, int64_t k_hs,
                          int64_t v_ts, int64_t v_hs, int max_tiles, float scale) {
     dim3 grid(max_tiles, batch, H), block(512);
-    size_t shmem = (64 * (DPAD + 16) + DPAD * 96 + 8 * 16 * 72) * sizeof(__bf16);
+    size_t shmem = (64 * (DPAD + 16) * 2 + 8 * 16 * 72) * sizeof(__bf16);
     hipLaunchKernelGGL((fa_fwd_kernel<DPAD>), grid, block, shmem, stream,
                        q, k, v, o, lse, cu, H, Hkv, D, G,
                        q_ts, q_gs, k_ts, k_hs, v_ts, v_hs, (int64_t)H * D, T, scale);
@@ -781,6 +896,67 @@ extern "C" int dolomite_fa_grad_finalize(dolomite_stream_t stream,
 // ===========================================================================
 // MFMA fragment-layout self-test (see header comment).
 // ===========================================================================
+
+// ds_read_b64_tr_b16 semantics probe: LDS holds bf16 values = element index
+// (0..255, exact in bf16). Each variant issues the transpose-read with a
+// different per-lane address convention; the host inspects which elements
+// landed in which lane/slot to pin the gather pattern (guide T10 says
+// lane l elem j <- lds[(l&15) + j*16 + (l>>4)*64] but is not a spec).
+typedef __bf16 bf16x4p __attribute__((ext_vector_type(4)));
+__global__ void __launch_bounds__(64) tr16_probe_kernel(float* out) {
+    __shared__ __bf16 s[256];
+    const int lane = threadIdx.x & 63;
+    for (int i = threadIdx.x; i < 256; i += 64) s[i] = (__bf16)(float)i;
+    __syncthreads();
+    const unsigned base = (unsigned)(size_t)(__bf16*)s;  // LDS byte address of s[0]
+    unsigned addrs[3] = {
+        base + (unsigned)((lane >> 4) * 128),                        // group base only
+        base + (unsigned)(lane * 8),                                 // lane-linear b64
+        base + (unsigned)(((lane & 15) * 2) + ((lane >> 4) * 128)),  // +column offset
+    };
+#pragma unroll
+    for (int v = 0; v < 3; ++v) {
+        bf16x4p r;
+        asm volatile("ds_read_b64_tr_b16 %0, %1\n\ts_waitcnt lgkmcnt(0)"
+                     : "=v"(r)
+                     : "v"(addrs[v])
+                     : "memory");
+#pragma unroll
+        for (int j = 0; j < 4; ++j) out[(v * 64 + lane) * 4 + j] = (float)r[j];
+    }
+}
+
+extern "C" int dolomite_tr16_probe(dolomite_stream_t stream, float* out) {
+    hipLaunchKernelGGL(tr16_probe_kernel, dim3(1), dim3(64), 0, (hipStream_t)stream, out);
+    return dol_last_error();
+}
+
+// End-to-end check of lds_tr16_bfrag under kernel conditions (dynamic LDS,
+// PI23-staged row-major image): src is a row-major [64][cols] bf16 matrix;
+// out[l][e] (float) = the B-fragment element e lane l received for the
+// chunk (rowbase, d0). Host asserts out[l][e] == src[rowbase+(l>>4)*8+e][d0+(l&15)].
+__global__ void __launch_bounds__(64) tr16_bfrag_probe_kernel(
+    const __bf16* src, float* out, int cols, int S, int rowbase, int d0) {
+    extern __shared__ char smem_raw[];
+    __bf16* img = (__bf16*)smem_raw;
+    const int lane = threadIdx.x & 63;
+    for (int i = threadIdx.x; i < 64 * (cols / 8); i += 64) {
+        int row = i / (cols / 8), c8 = (i % (cols / 8)) * 8;
+        *(bf16x8*)&img[PI23(row) * S + c8] = *(const bf16x8*)&src[row * cols + c8];
+    }
+    __syncthreads();
+    bf16x8 f = lds_tr16_bfrag(img, rowbase, S, d0, lane);
+#pragma unroll
+    for (int e = 0; e < 8; ++e) out[lane * 8 + e] = (float)f[e];
+}
+
+extern "C" int dolomite_tr16_bfrag_probe(dolomite_stream_t stream, const void* src, float* out,
+                                         int cols, int S, int rowbase, int d0) {
+    size_t shmem = (size_t)64 * S * sizeof(__bf16);
+    hipLaunchKernelGGL(tr16_bfrag_probe_kernel, dim3(1), dim3(64), shmem, (hipStream_t)stream,
+                       (const __bf16*)src, out, cols, S, rowbase, d0);
+    return dol_last_error();
+}
 
 __global__ void __launch_bounds__(64) mfma_probe_kernel(const __bf16* A, const __bf16* B, float* C) {
     int lane = threadIdx.x & 63;

@@ -55,6 +55,8 @@ _SIGNATURES = {
     "dolomite_moe_gemm_dgrad": ([_p, _p, _p, _p, _p, _i32, _i32, _i32, _i32, _i32], _i32),
     "dolomite_moe_gemm_wgrad": ([_p, _p, _p, _p, _p, _i32, _i32, _i32, _i32], _i32),
     "dolomite_mfma_probe": ([_p, _p, _p, _p], _i32),
+    "dolomite_tr16_probe": ([_p, _p], _i32),
+    "dolomite_tr16_bfrag_probe": ([_p, _p, _p, _i32, _i32, _i32, _i32], _i32),
     "dolomite_build_sample_idx_i32": ([_p, _p, _i32, _i32, _i64, _p, _i64], _i32),
     "dolomite_build_sample_idx_i64": ([_p, _p, _i32, _i32, _i64, _p, _i64], _i32),
     "dolomite_build_blending_indices": ([_p, _p, _p, _i32, _i64], _i32),
