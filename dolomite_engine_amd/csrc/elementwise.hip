@@ -850,115 +850,3 @@ extern "C" int dolomite_scale_inplace(dolomite_stream_t stream, void* buf, int64
 }
 
 extern "C" int dolomite_hip_abi_version(void) { return 1; }
-
-// ===========================================================================
-// Fused tanh-GELU (torch gelu approximate="tanh"; gpt_dolomite's
-// gelu_pytorch_tanh activation, mlp.py:45-50). at::native's elementwise
-// kernels run ~4.3 TB/s on bf16; these vec8 versions run at the norm
-// kernels' ~5.7 TB/s and keep the math in fp32.
-// ===========================================================================
-
-__device__ __forceinline__ float gelu_tanh_f32(float x) {
-    const float k = 0.7978845608028654f;   // sqrt(2/pi)
-    const float c = 0.044715f;
-    float t = tanhf(k * (x + c * x * x * x));
-    return 0.5f * x * (1.f + t);
-}
-
-__device__ __forceinline__ float gelu_tanh_grad_f32(float x) {
-    const float k = 0.7978845608028654f;
-    const float c = 0.044715f;
-    float x2 = x * x;
-    float t = tanhf(k * (x + c * x2 * x));
-    return 0.5f * (1.f + t) + 0.5f * x * (1.f - t * t) * k * (1.f + 3.f * c * x2);
-}
-
-__global__ void __launch_bounds__(256) gelu_fwd_kernel(const uint16_t* __restrict__ x,
-                                                       uint16_t* __restrict__ y, int64_t n) {
-    int64_t i0 = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) * 8;
-    if (i0 + 8 <= n) {
-        float v[8];
-        VecIO<uint16_t, 8>::load(x + i0, v);
-#pragma unroll
-        for (int e = 0; e < 8; ++e) v[e] = gelu_tanh_f32(v[e]);
-        VecIO<uint16_t, 8>::store(y + i0, v);
-    } else {
-        for (int64_t i = i0; i < n; ++i) y[i] = f32_to_bf16(gelu_tanh_f32(bf16_to_f32(x[i])));
-    }
-}
-
-__global__ void __launch_bounds__(256) gelu_bwd_kernel(const uint16_t* __restrict__ x,
-                                                       const uint16_t* __restrict__ dy,
-                                                       uint16_t* __restrict__ dx, int64_t n) {
-    int64_t i0 = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) * 8;
-    if (i0 + 8 <= n) {
-        float xv[8], dv[8];
-        VecIO<uint16_t, 8>::load(x + i0, xv);
-        VecIO<uint16_t, 8>::load(dy + i0, dv);
-#pragma unroll
-        for (int e = 0; e < 8; ++e) dv[e] *= gelu_tanh_grad_f32(xv[e]);
-        VecIO<uint16_t, 8>::store(dx + i0, dv);
-    } else {
-        for (int64_t i = i0; i < n; ++i)
-            dx[i] = f32_to_bf16(bf16_to_f32(dy[i]) * gelu_tanh_grad_f32(bf16_to_f32(x[i])));
-    }
-}
-
-extern "C" int dolomite_gelu_fwd(dolomite_stream_t stream, const void* x, void* y, int64_t n, int dtype) {
-    if (dtype != DOLOMITE_BF16) return 9010;
-    dim3 grid((uint32_t)((n / 8 + 255) / 256 + 1)), block(256);
-    hipLaunchKernelGGL(gelu_fwd_kernel, grid, block, 0, (hipStream_t)stream,
-                       (const uint16_t*)x, (uint16_t*)y, n);
-    return dol_last_error();
-}
-
-extern "C" int dolomite_gelu_bwd(dolomite_stream_t stream, const void* x, const void* dy, void* dx,
-                                 int64_t n, int dtype) {
-    if (dtype != DOLOMITE_BF16) return 9010;
-    dim3 grid((uint32_t)((n / 8 + 255) / 256 + 1)), block(256);
-    hipLaunchKernelGGL(gelu_bwd_kernel, grid, block, 0, (hipStream_t)stream,
-                       (const uint16_t*)x, (const uint16_t*)dy, (uint16_t*)dx, n);
-    return dol_last_error();
-}
-
-// ===========================================================================
-// Bias-gradient column sum: out[n] = sum_t dy[t][n], fp32 accumulation
-// (replaces at::native::reduce_kernel on the projection bias grads).
-// grid.y splits rows; fp32 partials joined with atomics (out must be zeroed).
-// ===========================================================================
-
-__global__ void __launch_bounds__(256) colsum_kernel(const uint16_t* __restrict__ dy,
-                                                     float* __restrict__ out,
-                                                     int64_t T, int64_t N, int64_t rows_per_blk) {
-    int64_t c0 = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) * 8;
-    if (c0 >= N) return;
-    int64_t r0 = (int64_t)blockIdx.y * rows_per_blk;
-    int64_t r1 = min(r0 + rows_per_blk, T);
-    float acc[8] = {0, 0, 0, 0, 0, 0, 0, 0};
-    if (c0 + 8 <= N) {
-        for (int64_t r = r0; r < r1; ++r) {
-            float v[8];
-            VecIO<uint16_t, 8>::load(dy + r * N + c0, v);
-#pragma unroll
-            for (int e = 0; e < 8; ++e) acc[e] += v[e];
-        }
-#pragma unroll
-        for (int e = 0; e < 8; ++e) atomicAdd(&out[c0 + e], acc[e]);
-    } else {
-        for (int64_t r = r0; r < r1; ++r)
-            for (int64_t c = c0; c < N; ++c) acc[c - c0] += bf16_to_f32(dy[r * N + c]);
-        for (int64_t c = c0; c < N; ++c) atomicAdd(&out[c], acc[c - c0]);
-    }
-}
-
-extern "C" int dolomite_bias_colsum(dolomite_stream_t stream, const void* dy, float* out,
-                                    int64_t T, int64_t N, int dtype) {
-    if (dtype != DOLOMITE_BF16) return 9010;
-    // ~1024 rows per block slice; enough blocks to fill 256 CUs
-    int64_t rows_per_blk = 1024;
-    int64_t nyc = (T + rows_per_blk - 1) / rows_per_blk;
-    dim3 grid((uint32_t)((N / 8 + 255) / 256 + 1), (uint32_t)nyc), block(256);
-    hipLaunchKernelGGL(colsum_kernel, grid, block, 0, (hipStream_t)stream,
-                       (const uint16_t*)dy, out, T, N, rows_per_blk);
-    return dol_last_error();
-}

@@ -24,10 +24,8 @@ from transformers.modeling_outputs import BaseModelOutputWithPast, CausalLMOutpu
 from ..ops import (
     QKVLayout,
     fused_cross_entropy,
-    fused_gelu_tanh,
     fused_layernorm,
     fused_rmsnorm,
-    proj_linear,
     rope_packed_qkv,
     varlen_attention,
 )
@@ -52,11 +50,6 @@ class ParameterizedLinear(nn.Linear):
             nn.init.normal_(self.weight, mean=0, std=self.std)
             if getattr(self, "bias", None) is not None:
                 self.bias.zero_()
-
-    def forward(self, input):
-        # GPU 2-D bf16 with bias: same rocBLAS GEMMs, fused bias-grad
-        # column sum in backward (ops/functional.py ProjLinear)
-        return proj_linear(input, self.weight, self.bias)
 
 
 class ParameterizedEmbedding(nn.Embedding):
@@ -291,7 +284,7 @@ def _activation(name: str):
         if base is None:
             raise NotImplementedError(f"activation {name}")
     if base in ("gelu_pytorch_tanh", "gelu_tanh"):
-        fn = fused_gelu_tanh  # vec8 HIP kernel on GPU; F.gelu(tanh) on CPU
+        fn = lambda x: F.gelu(x, approximate="tanh")
     elif base == "gelu":
         fn = F.gelu
     elif base in ("silu", "swish"):

@@ -5,10 +5,8 @@ from .functional import (
     adamw_step_flat,
     fused_cross_entropy,
     fused_layernorm,
-    fused_gelu_tanh,
     fused_rmsnorm,
     grouped_expert_gemm,
-    proj_linear,
     rope_packed_qkv,
     varlen_attention,
 )

@@ -659,78 +659,3 @@ def grouped_expert_gemm(x, weight, bias, num_tokens_per_expert):
     if max_rows == 0:
         return torch.zeros(x.shape[0], N, dtype=x.dtype, device=x.device)
     return GroupedExpertGemm.apply(x.contiguous(), weight, bias, offsets, max_rows)
-
-
-# ---------------------------------------------------------------------------
-# Fused tanh-GELU and bias-aware projection linear (GPU-only fast paths;
-# the model falls back to plain torch ops on CPU so CPU parity stays
-# bit-exact with the reference eager path).
-# ---------------------------------------------------------------------------
-
-
-class FusedGeluTanh(torch.autograd.Function):
-    """F.gelu(x, approximate="tanh") with vec8 bf16 kernels (fp32 math)."""
-
-    @staticmethod
-    def forward(ctx, x):
-        x = x.contiguous()
-        y = torch.empty_like(x)
-        with hip.prof("gelu_fwd"):
-            hip.check(hip.lib().dolomite_gelu_fwd(hip.stream(), hip.ptr(x), hip.ptr(y), x.numel(), hip.dt(x)),
-                      "gelu_fwd")
-        ctx.save_for_backward(x)
-        return y
-
-    @staticmethod
-    def backward(ctx, dy):
-        (x,) = ctx.saved_tensors
-        dy = dy.contiguous()
-        dx = torch.empty_like(x)
-        with hip.prof("gelu_bwd"):
-            hip.check(hip.lib().dolomite_gelu_bwd(hip.stream(), hip.ptr(x), hip.ptr(dy), hip.ptr(dx),
-                                                  x.numel(), hip.dt(x)),
-                      "gelu_bwd")
-        return dx
-
-
-def fused_gelu_tanh(x):
-    if x.is_cuda and x.dtype == torch.bfloat16:
-        return FusedGeluTanh.apply(x)
-    return torch.nn.functional.gelu(x, approximate="tanh")
-
-
-class ProjLinear(torch.autograd.Function):
-    """F.linear with the SAME rocBLAS GEMMs autograd would use, but the bias
-    gradient through the fused fp32 column-sum kernel instead of
-    at::native::reduce (measured 2x on the (T, N) bf16 reductions)."""
-
-    @staticmethod
-    def forward(ctx, x, weight, bias):
-        ctx.save_for_backward(x, weight)
-        ctx.bias_dtype = bias.dtype
-        return torch.nn.functional.linear(x, weight, bias)
-
-    @staticmethod
-    def backward(ctx, dy):
-        x, weight = ctx.saved_tensors
-        dy = dy.contiguous()
-        dx = dy @ weight if ctx.needs_input_grad[0] else None
-        dw = dy.transpose(-1, -2) @ x if ctx.needs_input_grad[1] else None
-        db = None
-        if ctx.needs_input_grad[2]:
-            t2d = dy.reshape(-1, dy.shape[-1])
-            acc = torch.zeros(t2d.shape[1], dtype=torch.float32, device=dy.device)
-            with hip.prof("bias_colsum"):
-                hip.check(hip.lib().dolomite_bias_colsum(hip.stream(), hip.ptr(t2d), hip.ptr(acc),
-                                                         t2d.shape[0], t2d.shape[1], hip.dt(t2d)),
-                          "bias_colsum")
-            db = acc.to(ctx.bias_dtype)
-        return dx, dw, db
-
-
-def proj_linear(x, weight, bias):
-    """Dispatch: fused-bias-grad path on CUDA bf16 2-D inputs with bias."""
-    if (bias is not None and x.is_cuda and x.dtype == torch.bfloat16
-            and x.dim() == 2 and x.shape[1] % 8 == 0 and weight.shape[0] % 8 == 0):
-        return ProjLinear.apply(x, weight, bias)
-    return torch.nn.functional.linear(x, weight, bias)

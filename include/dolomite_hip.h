@@ -201,17 +201,6 @@ int dolomite_scale_inplace(dolomite_stream_t stream, void* buf, int64_t n,
                            float scale, int dtype);
 
 /* ------------------------------------------------------------------------
- * Fused tanh-GELU (gelu_pytorch_tanh activation, gpt_dolomite/mlp.py:45-50;
- * fp32 math, bf16 I/O) and the projection-bias gradient column sum
- * (out[n] = sum_t dy[t][n], fp32 accumulation; out must be zeroed).
- * ---------------------------------------------------------------------- */
-int dolomite_gelu_fwd(dolomite_stream_t stream, const void* x, void* y, int64_t n, int dtype);
-int dolomite_gelu_bwd(dolomite_stream_t stream, const void* x, const void* dy, void* dx,
-                      int64_t n, int dtype);
-int dolomite_bias_colsum(dolomite_stream_t stream, const void* dy, float* out,
-                         int64_t T, int64_t N, int dtype);
-
-/* ------------------------------------------------------------------------
  * Grouped expert GEMMs for the MoE family — replaces the reference's
  * scattermoe/triton grouped path (moe_dolomite/moe/scatter.py:109-138) for
  * the SparseMoE expert matmuls (moe/base.py:137-156). Rows are
