@@ -15,7 +15,7 @@ PKG = CSRC.parent
 REPO = PKG.parent
 SO_PATH = PKG / "libdolomite_hip.so"
 
-SOURCES = [CSRC / "elementwise.hip", CSRC / "attention.hip", CSRC / "moe_gemm.hip", CSRC / "data_index.cpp", CSRC / "blaslt.cpp"]
+SOURCES = [CSRC / "elementwise.hip", CSRC / "attention.hip", CSRC / "moe_gemm.hip", CSRC / "data_index.cpp"]
 
 
 def _needs_rebuild() -> bool:
@@ -37,7 +37,6 @@ def build(force: bool = False) -> Path:
         "-fPIC",
         "-shared",
         *[str(s) for s in SOURCES],
-        "-lhipblaslt",
         "-o",
         str(SO_PATH),
     ]

@@ -54,8 +54,6 @@ _SIGNATURES = {
     "dolomite_moe_gemm_fwd": ([_p, _p, _p, _p, _p, _p, _i32, _i32, _i32, _i32, _i32], _i32),
     "dolomite_moe_gemm_dgrad": ([_p, _p, _p, _p, _p, _i32, _i32, _i32, _i32, _i32], _i32),
     "dolomite_moe_gemm_wgrad": ([_p, _p, _p, _p, _p, _i32, _i32, _i32, _i32], _i32),
-    "dolomite_mlp_fc_gelu_fwd": ([_p, _p, _p, _p, _p, _p, _i64, _i64, _i64], _i32),
-    "dolomite_mlp_dgelu_dgrad": ([_p, _p, _p, _p, _p, _p, _i64, _i64, _i64], _i32),
     "dolomite_mfma_probe": ([_p, _p, _p, _p], _i32),
     "dolomite_tr16_probe": ([_p, _p], _i32),
     "dolomite_tr16_bfrag_probe": ([_p, _p, _p, _i32, _i32, _i32, _i32], _i32),
