@@ -68,7 +68,36 @@ MODEL_350M = dict(
     pad_token_id=0,
 )
 
-MODELS = {"3b": MODEL_3B, "350m": MODEL_350M}
+# MoE-1.6B (moe_dolomite family, §8f.3): 32 experts top-2, swiglu —
+# exercises the grouped expert GEMM path end-to-end (~250M active params)
+MODEL_MOE = dict(
+    model_type="moe_dolomite",
+    vocab_size=49152,
+    n_positions=4096,
+    n_embd=1024,
+    n_layer=24,
+    n_head=16,
+    num_key_value_heads=None,
+    attention_head_type="mqa",
+    n_inner=1024,
+    activation_function="swiglu",
+    normalization_function="rmsnorm",
+    position_embedding_type="rope",
+    rope_theta=10000,
+    resid_pdrop=0.0,
+    embd_pdrop=0.0,
+    attn_pdrop=0.0,
+    layer_norm_epsilon=1e-5,
+    tie_word_embeddings=True,
+    add_bias=False,
+    num_experts=32,
+    num_experts_per_tok=2,
+    bos_token_id=0,
+    eos_token_id=0,
+    pad_token_id=0,
+)
+
+MODELS = {"3b": MODEL_3B, "350m": MODEL_350M, "moe": MODEL_MOE}
 
 SEQ_LEN = 4096
 MICRO_BATCH = 16  # tokens per rank per step = 16 * 4096 = 65536 (fills HBM better; +5% vs B=8)
@@ -191,9 +220,10 @@ def main():
     ap.add_argument("--skip-cpu-baseline", action="store_true")
     args = ap.parse_args()
     if args.seq_len is None:
-        args.seq_len = SEQ_LEN if args.model == "3b" else 2048
+        args.seq_len = 2048 if args.model == "350m" else SEQ_LEN
     if args.micro_batch is None:
-        args.micro_batch = MICRO_BATCH if args.model == "3b" else 64
+        args.micro_batch = {"3b": MICRO_BATCH, "350m": 64, "moe": 16}[args.model]
+    fam = "moedolomite" if args.model == "moe" else "gptdolomite"
 
     import torch.distributed as dist
 
@@ -306,7 +336,8 @@ def main():
 
     result = {
         "metric": (
-            f"tokens/sec/node GPTDolomite-{'3B' if args.model == '3b' else '350M'} "
+            f"tokens/sec/node "
+            f"{ {'3b': 'GPTDolomite-3B', '350m': 'GPTDolomite-350M', 'moe': 'MoEDolomite-1.6B'}[args.model] } "
             f"bf16 seq{args.seq_len} padding-free"
         ),
         "value": value,
@@ -321,8 +352,9 @@ def main():
         "dtype": "bf16",
         "data": "synthetic",
         "config": {
-            "workload": f"gptdolomite-{args.model}-bf16-seq{args.seq_len}-paddingfree-pretrain",
-            "model": "GPTDolomite-3B" if args.model == "3b" else "GPTDolomite-350M",
+            "workload": f"{fam}-{args.model}-bf16-seq{args.seq_len}-paddingfree-pretrain",
+            "model": {"3b": "GPTDolomite-3B", "350m": "GPTDolomite-350M",
+                      "moe": "MoEDolomite-1.6B-A0.25B"}[args.model],
             "global_batch": args.micro_batch * world,
             "seq_len": args.seq_len,
             "parallelism": f"dp{world}",

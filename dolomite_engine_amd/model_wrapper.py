@@ -4,9 +4,20 @@ reimplemented for the MI355X engine (single-node data parallel, no TP)."""
 import torch
 import torch.nn as nn
 
-from .hf_models import GPTDolomiteConfig, GPTDolomiteForCausalLM
+from .hf_models import (
+    GPTDolomiteConfig,
+    GPTDolomiteForCausalLM,
+    MoEDolomiteConfig,
+    MoEDolomiteForCausalLM,
+)
 from .ops import fused_cross_entropy
 from .utils import string_to_torch_dtype
+
+# model family registry (reference resolves via AutoConfig/model_type)
+_MODEL_CLASSES = {
+    "gpt_dolomite": (GPTDolomiteConfig, GPTDolomiteForCausalLM),
+    "moe_dolomite": (MoEDolomiteConfig, MoEDolomiteForCausalLM),
+}
 
 
 class ModelWrapper(nn.Module):
@@ -37,19 +48,26 @@ class ModelWrapper(nn.Module):
             )
 
         if model_name is None:
-            config = GPTDolomiteConfig(**pretrained_config)
+            cfg_kwargs = dict(pretrained_config)
+            model_type = cfg_kwargs.pop("model_type", "gpt_dolomite")
+            config_class, model_class = _MODEL_CLASSES[model_type]
+            config = config_class(**cfg_kwargs)
             config._attn_implementation = attention_implementation
             self.config = config
             import contextlib
 
             dev_ctx = torch.device(self._init_device) if self._init_device is not None else contextlib.nullcontext()
             with dev_ctx:
-                self.model = GPTDolomiteForCausalLM(config, use_padding_free_transformer=use_padding_free_transformer)
+                self.model = model_class(config, use_padding_free_transformer=use_padding_free_transformer)
         else:
-            config = GPTDolomiteConfig.from_pretrained(model_name)
+            from transformers import AutoConfig
+
+            model_type = AutoConfig.from_pretrained(model_name).model_type
+            config_class, model_class = _MODEL_CLASSES[model_type]
+            config = config_class.from_pretrained(model_name)
             config._attn_implementation = attention_implementation
             self.config = config
-            self.model = GPTDolomiteForCausalLM.from_pretrained(
+            self.model = model_class.from_pretrained(
                 model_name, config=config, use_padding_free_transformer=use_padding_free_transformer
             )
         self.model = self.model.to(dtype)
