@@ -6,12 +6,12 @@
 // layouts addressed in place on the fused c_attn projection output
 // (attention/base.py:72-81) via (t_stride, group_stride) addressing.
 //
-// Design (CDNA4-first, correctness-first round 1):
+// Design (CDNA4-first):
 //   - MFMA v_mfma_f32_16x16x32_bf16 for QK^T and PV; 64-wide waves.
-//   - workgroup = 4 waves = 256 threads; q-tile 64 rows (16 per wave),
-//     k-tile 64 keys streamed through LDS ([key][d] image for QK^T B-frags,
-//     transposed [d][key] image for PV B-frags — both read with 16-byte
-//     ds_read_b128, rows padded +16B so conflict groups hit distinct banks).
+//   - forward: workgroup = 8 waves = 512 threads; q-tile 128 rows (16 per
+//     wave), k-tile 64 keys streamed through LDS as row-major [key][d]
+//     images (stride DPAD+16, conflict-free b128 B-frag reads); the PV
+//     B-fragment comes from the PI23-rowed V image via ds_read_b64_tr_b16.
 //   - online softmax in fp32 VGPRs, m/l carried per row, row reductions via
 //     16-lane __shfl_xor (the C-fragment's 16 columns of one row).
 //   - P routed through a per-wave LDS tile to re-shape C-layout -> A-layout.
