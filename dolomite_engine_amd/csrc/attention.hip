@@ -89,77 +89,6 @@ __device__ __forceinline__ bf16x8 lds_tr16_bfrag(const __bf16* img, int rowbase,
     return out;
 }
 
-// Two B-fragments (same chunk, two images) in one asm block — the four
-// reads pipeline in the LDS queue behind a single drain.
-__device__ __forceinline__ void lds_tr16_bfrag2(const __bf16* imgA, const __bf16* imgB, int rowbase,
-                                                int S, int d0, int lane, bf16x8* fA, bf16x8* fB) {
-    const int g8 = (lane >> 4) * 8;
-    const int jj = (lane >> 2) & 3;
-    const int cc = (lane & 3) * 4;
-    const int off0 = PI23(rowbase + g8 + jj) * S + d0 + cc;
-    const int off1 = PI23(rowbase + g8 + 4 + jj) * S + d0 + cc;
-    unsigned a0 = (unsigned)(size_t)(imgA + off0);
-    unsigned a1 = (unsigned)(size_t)(imgA + off1);
-    unsigned b0 = (unsigned)(size_t)(imgB + off0);
-    unsigned b1 = (unsigned)(size_t)(imgB + off1);
-    bf16x4 alo, ahi, blo, bhi;
-    asm volatile(
-        "ds_read_b64_tr_b16 %0, %4\n\t"
-        "ds_read_b64_tr_b16 %1, %5\n\t"
-        "ds_read_b64_tr_b16 %2, %6\n\t"
-        "ds_read_b64_tr_b16 %3, %7\n\t"
-        "s_waitcnt lgkmcnt(0)"
-        : "=&v"(alo), "=&v"(ahi), "=&v"(blo), "=&v"(bhi)
-        : "v"(a0), "v"(a1), "v"(b0), "v"(b1)
-        : "memory");
-#pragma unroll
-    for (int t = 0; t < 4; ++t) {
-        (*fA)[t] = alo[t]; (*fA)[4 + t] = ahi[t];
-        (*fB)[t] = blo[t]; (*fB)[4 + t] = bhi[t];
-    }
-}
-
-// Four B-fragments (two col-blocks x two images) behind ONE drain — halves
-// the waitcnt stalls of the dkv inner loop vs per-dc drains.
-__device__ __forceinline__ void lds_tr16_bfrag4(const __bf16* imgA, const __bf16* imgB, int rowbase,
-                                                int S, int d0a, int d0b, int lane,
-                                                bf16x8* fA0, bf16x8* fB0, bf16x8* fA1, bf16x8* fB1) {
-    const int g8 = (lane >> 4) * 8;
-    const int jj = (lane >> 2) & 3;
-    const int cc = (lane & 3) * 4;
-    const int r0 = PI23(rowbase + g8 + jj) * S + cc;
-    const int r1 = PI23(rowbase + g8 + 4 + jj) * S + cc;
-    unsigned a0 = (unsigned)(size_t)(imgA + r0 + d0a);
-    unsigned a1 = (unsigned)(size_t)(imgA + r1 + d0a);
-    unsigned b0 = (unsigned)(size_t)(imgB + r0 + d0a);
-    unsigned b1 = (unsigned)(size_t)(imgB + r1 + d0a);
-    unsigned a2 = (unsigned)(size_t)(imgA + r0 + d0b);
-    unsigned a3 = (unsigned)(size_t)(imgA + r1 + d0b);
-    unsigned b2 = (unsigned)(size_t)(imgB + r0 + d0b);
-    unsigned b3 = (unsigned)(size_t)(imgB + r1 + d0b);
-    bf16x4 o0, o1, o2, o3, o4, o5, o6, o7;
-    asm volatile(
-        "ds_read_b64_tr_b16 %0, %8\n\t"
-        "ds_read_b64_tr_b16 %1, %9\n\t"
-        "ds_read_b64_tr_b16 %2, %10\n\t"
-        "ds_read_b64_tr_b16 %3, %11\n\t"
-        "ds_read_b64_tr_b16 %4, %12\n\t"
-        "ds_read_b64_tr_b16 %5, %13\n\t"
-        "ds_read_b64_tr_b16 %6, %14\n\t"
-        "ds_read_b64_tr_b16 %7, %15\n\t"
-        "s_waitcnt lgkmcnt(0)"
-        : "=&v"(o0), "=&v"(o1), "=&v"(o2), "=&v"(o3), "=&v"(o4), "=&v"(o5), "=&v"(o6), "=&v"(o7)
-        : "v"(a0), "v"(a1), "v"(b0), "v"(b1), "v"(a2), "v"(a3), "v"(b2), "v"(b3)
-        : "memory");
-#pragma unroll
-    for (int t = 0; t < 4; ++t) {
-        (*fA0)[t] = o0[t]; (*fA0)[4 + t] = o1[t];
-        (*fB0)[t] = o2[t]; (*fB0)[4 + t] = o3[t];
-        (*fA1)[t] = o4[t]; (*fA1)[4 + t] = o5[t];
-        (*fB1)[t] = o6[t]; (*fB1)[4 + t] = o7[t];
-    }
-}
-
 // Branchless guarded 16B load. Requires D % 8 == 0 (every head dim on this
 // path): a chunk is then fully inside [0, D) or fully in the pad, so the
 // guard reduces to ONE wave-divergent-free vector load from a clamped
