@@ -310,3 +310,26 @@ def test_gpu_pretrain_cli_end_to_end(tmp_path):
     assert final_a.keys() == final_b.keys()
     for k in final_a:
         torch.testing.assert_close(final_b[k], final_a[k], rtol=0, atol=0, msg=lambda m: f"{k}: {m}")
+
+
+def test_gpu_finetune_cli_end_to_end(tmp_path):
+    """The `dolomite_engine_amd.finetune --config` path on hardware: bf16
+    padding-free finetune (list inputs, loss-internal) through the HIP
+    kernels, with a checkpoint written at the end."""
+    import json
+
+    import yaml
+
+    from dolomite_engine_amd import finetune
+    from tests.test_finetune_cpu import TINY_FT_CONFIG
+
+    cfg = json.loads(json.dumps(TINY_FT_CONFIG))
+    cfg["model_args"]["attention_implementation"] = "flash_attention_2"
+    cfg["model_args"]["use_padding_free_transformer"] = True
+    cfg["mixed_precision_args"] = {"dtype": "bf16"}
+    cfg["training_parameters"]["num_training_steps"] = 4
+    cfg["save_args"] = {"save_path": str(tmp_path / "ckpt"), "save_interval": 4}
+    p = tmp_path / "cfg.yml"
+    p.write_text(yaml.safe_dump(cfg))
+    finetune.main(["--config", str(p)])
+    assert (tmp_path / "ckpt" / "global_step4" / "model").exists()
