@@ -97,7 +97,33 @@ MODEL_MOE = dict(
     pad_token_id=0,
 )
 
-MODELS = {"3b": MODEL_3B, "350m": MODEL_350M, "moe": MODEL_MOE}
+# Llama-3-8B shape (config #4 class: GQA d_head=128, swiglu, seq 8192 —
+# the import_from_huggingface finetune target measured at micro-batch 1)
+MODEL_LLAMA8B = dict(
+    vocab_size=128256,
+    n_positions=8192,
+    n_embd=4096,
+    n_layer=32,
+    n_head=32,
+    num_key_value_heads=8,
+    attention_head_type="gqa",
+    n_inner=14336,
+    activation_function="swiglu",
+    normalization_function="rmsnorm",
+    position_embedding_type="rope",
+    rope_theta=500000,
+    resid_pdrop=0.0,
+    embd_pdrop=0.0,
+    attn_pdrop=0.0,
+    layer_norm_epsilon=1e-5,
+    add_bias=False,
+    tie_word_embeddings=False,
+    bos_token_id=0,
+    eos_token_id=0,
+    pad_token_id=0,
+)
+
+MODELS = {"3b": MODEL_3B, "350m": MODEL_350M, "moe": MODEL_MOE, "llama8b": MODEL_LLAMA8B}
 
 SEQ_LEN = 4096
 MICRO_BATCH = 16  # tokens per rank per step = 16 * 4096 = 65536 (fills HBM better; +5% vs B=8)
@@ -220,9 +246,9 @@ def main():
     ap.add_argument("--skip-cpu-baseline", action="store_true")
     args = ap.parse_args()
     if args.seq_len is None:
-        args.seq_len = 2048 if args.model == "350m" else SEQ_LEN
+        args.seq_len = {"350m": 2048, "llama8b": 8192}.get(args.model, SEQ_LEN)
     if args.micro_batch is None:
-        args.micro_batch = {"3b": MICRO_BATCH, "350m": 64, "moe": 16}[args.model]
+        args.micro_batch = {"3b": MICRO_BATCH, "350m": 64, "moe": 16, "llama8b": 2}[args.model]
     fam = "moedolomite" if args.model == "moe" else "gptdolomite"
 
     import torch.distributed as dist
@@ -337,7 +363,7 @@ def main():
     result = {
         "metric": (
             f"tokens/sec/node "
-            f"{ {'3b': 'GPTDolomite-3B', '350m': 'GPTDolomite-350M', 'moe': 'MoEDolomite-1.6B'}[args.model] } "
+            f"{ {'3b': 'GPTDolomite-3B', '350m': 'GPTDolomite-350M', 'moe': 'MoEDolomite-1.6B', 'llama8b': 'Llama-3-8B-shape'}[args.model] } "
             f"bf16 seq{args.seq_len} padding-free"
         ),
         "value": value,
@@ -354,7 +380,7 @@ def main():
         "config": {
             "workload": f"{fam}-{args.model}-bf16-seq{args.seq_len}-paddingfree-pretrain",
             "model": {"3b": "GPTDolomite-3B", "350m": "GPTDolomite-350M",
-                      "moe": "MoEDolomite-1.6B-A0.25B"}[args.model],
+                      "moe": "MoEDolomite-1.6B-A0.25B", "llama8b": "Llama-3-8B-shape"}[args.model],
             "global_batch": args.micro_batch * world,
             "seq_len": args.seq_len,
             "parallelism": f"dp{world}",
