@@ -170,6 +170,7 @@ __global__ void __launch_bounds__(256) norm_bwd_kernel(
     const T* __restrict__ dy, const T* __restrict__ s, const T* __restrict__ w,
     const float* __restrict__ mean, const float* __restrict__ rstd,
     T* __restrict__ dx, float* __restrict__ dw_partial,
+    const T* __restrict__ dres,  // optional residual-stream grad: dx += dres
     int64_t T_rows, int H) {
     const int lane = threadIdx.x & 63;
     const int64_t wslot = (int64_t)blockIdx.x * 4 + (threadIdx.x >> 6);
@@ -222,11 +223,14 @@ __global__ void __launch_bounds__(256) norm_bwd_kernel(
         float m1 = IS_LN ? wave_reduce_sum(d1) / (float)H : 0.f;
 
         T* dxr = dx + row * (int64_t)H;
+        const T* drr = dres ? dres + row * (int64_t)H : nullptr;
 #pragma unroll
         for (int i = 0; i < ITMAX; ++i) {
             int c0 = (i * 64 + lane) * V;
             if (c0 < H) {
                 float dxv[V];
+                float drv[V];
+                if (drr) VecIO<T, V>::load(drr + c0, drv);
 #pragma unroll
                 for (int kk = 0; kk < V; ++kk) {
                     if (IS_LN) {
@@ -242,6 +246,9 @@ __global__ void __launch_bounds__(256) norm_bwd_kernel(
                         dw_acc[i][kk] += dyv[i][kk] * load_as_f32(&tmp);
                     }
                 }
+                if (drr)
+#pragma unroll
+                    for (int kk = 0; kk < V; ++kk) dxv[kk] += drv[kk];
                 VecIO<T, V>::store(dxr + c0, dxv);
             }
         }
@@ -265,26 +272,26 @@ __global__ void __launch_bounds__(256) norm_bwd_kernel(
 extern "C" int dolomite_rmsnorm_bwd(dolomite_stream_t stream,
                                     const void* dy, const void* s, const void* w,
                                     const float* rstd, void* dx, float* dw_partial,
-                                    int64_t T_rows, int64_t H, int dtype) {
+                                    const void* dres, int64_t T_rows, int64_t H, int dtype) {
     if (T_rows == 0) return 0;
     hipStream_t stream_ = (hipStream_t)stream;
     dim3 grid(RMS_BWD_BLOCKS), block(256);
     if (dtype == DOLOMITE_BF16 && H % 8 == 0)
         DOL_NORM_DISPATCH(norm_bwd_kernel, uint16_t, 8, false,
                           (const uint16_t*)dy, (const uint16_t*)s, (const uint16_t*)w,
-                          nullptr, rstd, (uint16_t*)dx, dw_partial, T_rows, (int)H);
+                          nullptr, rstd, (uint16_t*)dx, dw_partial, (const uint16_t*)dres, T_rows, (int)H);
     else if (dtype == DOLOMITE_BF16)
         DOL_NORM_DISPATCH(norm_bwd_kernel, uint16_t, 1, false,
                           (const uint16_t*)dy, (const uint16_t*)s, (const uint16_t*)w,
-                          nullptr, rstd, (uint16_t*)dx, dw_partial, T_rows, (int)H);
+                          nullptr, rstd, (uint16_t*)dx, dw_partial, (const uint16_t*)dres, T_rows, (int)H);
     else if (H % 4 == 0)
         DOL_NORM_DISPATCH(norm_bwd_kernel, float, 4, false,
                           (const float*)dy, (const float*)s, (const float*)w,
-                          nullptr, rstd, (float*)dx, dw_partial, T_rows, (int)H);
+                          nullptr, rstd, (float*)dx, dw_partial, (const float*)dres, T_rows, (int)H);
     else
         DOL_NORM_DISPATCH(norm_bwd_kernel, float, 1, false,
                           (const float*)dy, (const float*)s, (const float*)w,
-                          nullptr, rstd, (float*)dx, dw_partial, T_rows, (int)H);
+                          nullptr, rstd, (float*)dx, dw_partial, (const float*)dres, T_rows, (int)H);
     return dol_last_error();
 }
 
@@ -415,26 +422,26 @@ extern "C" int dolomite_layernorm_bwd(dolomite_stream_t stream,
                                       const void* dy, const void* s, const void* w,
                                       const float* mean, const float* rstd,
                                       void* dx, float* dwdb_partial,
-                                      int64_t T_rows, int64_t H, int dtype) {
+                                      const void* dres, int64_t T_rows, int64_t H, int dtype) {
     if (T_rows == 0) return 0;
     hipStream_t stream_ = (hipStream_t)stream;
     dim3 grid(RMS_BWD_BLOCKS), block(256);
     if (dtype == DOLOMITE_BF16 && H % 8 == 0)
         DOL_NORM_DISPATCH(norm_bwd_kernel, uint16_t, 8, true,
                           (const uint16_t*)dy, (const uint16_t*)s, (const uint16_t*)w,
-                          mean, rstd, (uint16_t*)dx, dwdb_partial, T_rows, (int)H);
+                          mean, rstd, (uint16_t*)dx, dwdb_partial, (const uint16_t*)dres, T_rows, (int)H);
     else if (dtype == DOLOMITE_BF16)
         DOL_NORM_DISPATCH(norm_bwd_kernel, uint16_t, 1, true,
                           (const uint16_t*)dy, (const uint16_t*)s, (const uint16_t*)w,
-                          mean, rstd, (uint16_t*)dx, dwdb_partial, T_rows, (int)H);
+                          mean, rstd, (uint16_t*)dx, dwdb_partial, (const uint16_t*)dres, T_rows, (int)H);
     else if (H % 4 == 0)
         DOL_NORM_DISPATCH(norm_bwd_kernel, float, 4, true,
                           (const float*)dy, (const float*)s, (const float*)w,
-                          mean, rstd, (float*)dx, dwdb_partial, T_rows, (int)H);
+                          mean, rstd, (float*)dx, dwdb_partial, (const float*)dres, T_rows, (int)H);
     else
         DOL_NORM_DISPATCH(norm_bwd_kernel, float, 1, true,
                           (const float*)dy, (const float*)s, (const float*)w,
-                          mean, rstd, (float*)dx, dwdb_partial, T_rows, (int)H);
+                          mean, rstd, (float*)dx, dwdb_partial, (const float*)dres, T_rows, (int)H);
     return dol_last_error();
 }
 
@@ -849,4 +856,4 @@ extern "C" int dolomite_scale_inplace(dolomite_stream_t stream, void* buf, int64
     return dol_last_error();
 }
 
-extern "C" int dolomite_hip_abi_version(void) { return 1; }
+extern "C" int dolomite_hip_abi_version(void) { return 2; }

@@ -135,6 +135,7 @@ class FusedRMSNorm(torch.autograd.Function):
         dy2 = dy.reshape(-1, H)
         if s.is_cuda:
             dy2 = dy2.contiguous()
+            ds2 = ds.reshape(-1, H).contiguous() if ds is not None else None
             dx = torch.empty_like(s)
             nb = hip.lib().dolomite_rmsnorm_bwd_nblocks(s.shape[0])
             dw_partial = torch.empty(nb, H, dtype=torch.float32, device=s.device)
@@ -143,6 +144,7 @@ class FusedRMSNorm(torch.autograd.Function):
                 hip.lib().dolomite_rmsnorm_bwd(
                     hip.stream(), hip.ptr(dy2), hip.ptr(s), hip.ptr(weight),
                     hip.ptr(rstd), hip.ptr(dx), hip.ptr(dw_partial),
+                    hip.ptr(ds2),  # residual-stream grad folded into dx in-kernel
                     s.shape[0], H, hip.dt(s),
                 ),
                 "rmsnorm_bwd",
@@ -153,11 +155,12 @@ class FusedRMSNorm(torch.autograd.Function):
                 "reduce_partials",
             )
             dw = dw32.to(weight.dtype)
+            dx = dx.view(ctx.shape)
         else:
             dx, dw = _rmsnorm_bwd_cpu(dy2, s, weight, rstd, dy.dtype)
-        dx = dx.view(ctx.shape)
-        if ds is not None:
-            dx = dx + ds
+            dx = dx.view(ctx.shape)
+            if ds is not None:
+                dx = dx + ds
         dres = dx if ctx.has_residual else None
         return dx, dw, None, dres
 
@@ -217,6 +220,7 @@ class FusedLayerNorm(torch.autograd.Function):
         dy2 = dy.reshape(-1, H)
         if s.is_cuda:
             dy2 = dy2.contiguous()
+            ds2 = ds.reshape(-1, H).contiguous() if ds is not None else None
             dx = torch.empty_like(s)
             nb = hip.lib().dolomite_rmsnorm_bwd_nblocks(s.shape[0])
             dwdb = torch.empty(2 * nb, H, dtype=torch.float32, device=s.device)
@@ -224,7 +228,7 @@ class FusedLayerNorm(torch.autograd.Function):
                 hip.lib().dolomite_layernorm_bwd(
                     hip.stream(), hip.ptr(dy2), hip.ptr(s), hip.ptr(weight),
                     hip.ptr(mean), hip.ptr(rstd), hip.ptr(dx), hip.ptr(dwdb),
-                    s.shape[0], H, hip.dt(s),
+                    hip.ptr(ds2), s.shape[0], H, hip.dt(s),
                 ),
                 "layernorm_bwd",
             )
@@ -238,6 +242,7 @@ class FusedLayerNorm(torch.autograd.Function):
             )
             dw = dw32.to(weight.dtype)
             db = db32.to(weight.dtype)
+            dx = dx.view(ctx.shape)
         else:
             s32 = s.float()
             xhat = (s32 - mean.unsqueeze(-1)) * rstd.unsqueeze(-1)
@@ -247,9 +252,9 @@ class FusedLayerNorm(torch.autograd.Function):
             dx = (rstd.unsqueeze(-1) * (wdy - m1 - xhat * m2)).to(dy.dtype)
             dw = (dy2.float() * xhat).sum(0).to(weight.dtype)
             db = dy2.float().sum(0).to(weight.dtype)
-        dx = dx.view(ctx.shape)
-        if ds is not None:
-            dx = dx + ds
+            dx = dx.view(ctx.shape)
+            if ds is not None:
+                dx = dx + ds
         dres = dx if ctx.has_residual else None
         return dx, dw, db, None, dres
 

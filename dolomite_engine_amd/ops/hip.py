@@ -12,7 +12,7 @@ from pathlib import Path
 import torch
 
 _SO_PATH = Path(__file__).resolve().parent.parent / "libdolomite_hip.so"
-_ABI_VERSION = 1
+_ABI_VERSION = 2
 
 _lib = None
 
@@ -30,9 +30,9 @@ _SIGNATURES = {
     "dolomite_hip_abi_version": ([], _i32),
     "dolomite_rmsnorm_fwd": ([_p, _p, _p, _p, _p, _p, _p, _i64, _i64, _f32, _i32], _i32),
     "dolomite_rmsnorm_bwd_nblocks": ([_i64], _i32),
-    "dolomite_rmsnorm_bwd": ([_p, _p, _p, _p, _p, _p, _p, _i64, _i64, _i32], _i32),
+    "dolomite_rmsnorm_bwd": ([_p, _p, _p, _p, _p, _p, _p, _p, _i64, _i64, _i32], _i32),
     "dolomite_layernorm_fwd": ([_p, _p, _p, _p, _p, _p, _p, _p, _p, _i64, _i64, _f32, _i32], _i32),
-    "dolomite_layernorm_bwd": ([_p, _p, _p, _p, _p, _p, _p, _p, _i64, _i64, _i32], _i32),
+    "dolomite_layernorm_bwd": ([_p, _p, _p, _p, _p, _p, _p, _p, _p, _i64, _i64, _i32], _i32),
     "dolomite_reduce_partials": ([_p, _p, _p, _i64, _i64], _i32),
     "dolomite_rope_qkv": ([_p, _p, _p, _p, _p, _i64, _i64, _i32, _i32, _i32, _i32, _i64, _i64, _i64, _i32, _i32, _i32], _i32),
     "dolomite_fa_varlen_fwd": (

@@ -48,7 +48,9 @@ int dolomite_rmsnorm_fwd(dolomite_stream_t stream,
 
 /* Backward: given dy and the saved pre-norm input s (= res_out of fwd, or x
  * when no residual) and rstd, computes
- *   dx  = r*(w*dy - s_hat * mean(w*dy*s_hat))  with s_hat = s_fp32*rstd
+ *   dx  = r*(w*dy - s_hat * mean(w*dy*s_hat)) [+ dres_in]  (s_hat = s_fp32*rstd;
+ *         dres_in, when non-NULL, is the residual-stream gradient folded into
+ *         dx in the same pass instead of a separate elementwise add)
  *   dw  = sum_rows(dy * cast_to_dtype(s_hat))        (fp32 accumulation)
  * dw_partial: (nblocks, H) fp32 scratch written by the kernel;
  * dolomite_reduce_partials sums it into dw (H,) fp32.
@@ -57,7 +59,7 @@ int dolomite_rmsnorm_bwd_nblocks(int64_t T);
 int dolomite_rmsnorm_bwd(dolomite_stream_t stream,
                          const void* dy, const void* s, const void* w,
                          const float* rstd, void* dx, float* dw_partial,
-                         int64_t T, int64_t H, int dtype);
+                         const void* dres_in, int64_t T, int64_t H, int dtype);
 
 /* LayerNorm, same calling pattern (replaces the reference 'layernorm'/'torch'
  * nn.LayerNorm path; mean/rstd cached fp32). db shares dw_partial layout:
@@ -70,7 +72,7 @@ int dolomite_layernorm_bwd(dolomite_stream_t stream,
                            const void* dy, const void* s, const void* w,
                            const float* mean, const float* rstd,
                            void* dx, float* dwdb_partial,
-                           int64_t T, int64_t H, int dtype);
+                           const void* dres_in, int64_t T, int64_t H, int dtype);
 
 /* Sum partials: out[h] = sum_i partial[i*H + h]; out fp32 (H,). */
 int dolomite_reduce_partials(dolomite_stream_t stream,

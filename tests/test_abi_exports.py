@@ -49,7 +49,7 @@ def test_library_exports_every_declared_symbol(solib):
 def test_abi_version(solib):
     fn = solib.dolomite_hip_abi_version
     fn.restype = ctypes.c_int32
-    assert fn() == 1
+    assert fn() == 2
 
 
 def test_loader_signatures_cover_header():
