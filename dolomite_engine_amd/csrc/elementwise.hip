@@ -165,13 +165,15 @@ extern "C" int dolomite_rmsnorm_bwd_nblocks(int64_t T_rows) {
     return RMS_BWD_BLOCKS * 4;
 }
 
-template <typename T, int V, int ITMAX, bool IS_LN>
+template <typename T, int V, int ITMAX, int MODE>  // MODE: bit0 = layernorm, bit1 = fused dres add
 __global__ void __launch_bounds__(256) norm_bwd_kernel(
     const T* __restrict__ dy, const T* __restrict__ s, const T* __restrict__ w,
     const float* __restrict__ mean, const float* __restrict__ rstd,
     T* __restrict__ dx, float* __restrict__ dw_partial,
-    const T* __restrict__ dres,  // optional residual-stream grad: dx += dres
+    const T* __restrict__ dres,  // residual-stream grad: dx += dres (iff HAS_DRES)
     int64_t T_rows, int H) {
+    constexpr bool IS_LN = (MODE & 1) != 0;
+    constexpr bool HAS_DRES = (MODE & 2) != 0;
     const int lane = threadIdx.x & 63;
     const int64_t wslot = (int64_t)blockIdx.x * 4 + (threadIdx.x >> 6);
     const int64_t nslots = (int64_t)gridDim.x * 4;
@@ -223,14 +225,14 @@ __global__ void __launch_bounds__(256) norm_bwd_kernel(
         float m1 = IS_LN ? wave_reduce_sum(d1) / (float)H : 0.f;
 
         T* dxr = dx + row * (int64_t)H;
-        const T* drr = dres ? dres + row * (int64_t)H : nullptr;
+        const T* drr = HAS_DRES ? dres + row * (int64_t)H : nullptr;
 #pragma unroll
         for (int i = 0; i < ITMAX; ++i) {
             int c0 = (i * 64 + lane) * V;
             if (c0 < H) {
                 float dxv[V];
                 float drv[V];
-                if (drr) VecIO<T, V>::load(drr + c0, drv);
+                if (HAS_DRES) VecIO<T, V>::load(drr + c0, drv);
 #pragma unroll
                 for (int kk = 0; kk < V; ++kk) {
                     if (IS_LN) {
@@ -246,7 +248,7 @@ __global__ void __launch_bounds__(256) norm_bwd_kernel(
                         dw_acc[i][kk] += dyv[i][kk] * load_as_f32(&tmp);
                     }
                 }
-                if (drr)
+                if (HAS_DRES)
 #pragma unroll
                     for (int kk = 0; kk < V; ++kk) dxv[kk] += drv[kk];
                 VecIO<T, V>::store(dxr + c0, dxv);
@@ -276,20 +278,39 @@ extern "C" int dolomite_rmsnorm_bwd(dolomite_stream_t stream,
     if (T_rows == 0) return 0;
     hipStream_t stream_ = (hipStream_t)stream;
     dim3 grid(RMS_BWD_BLOCKS), block(256);
+    if (dres == nullptr) {
+        if (dtype == DOLOMITE_BF16 && H % 8 == 0)
+            DOL_NORM_DISPATCH(norm_bwd_kernel, uint16_t, 8, 0,
+                              (const uint16_t*)dy, (const uint16_t*)s, (const uint16_t*)w,
+                              nullptr, rstd, (uint16_t*)dx, dw_partial, (const uint16_t*)dres, T_rows, (int)H);
+        else if (dtype == DOLOMITE_BF16)
+            DOL_NORM_DISPATCH(norm_bwd_kernel, uint16_t, 1, 0,
+                              (const uint16_t*)dy, (const uint16_t*)s, (const uint16_t*)w,
+                              nullptr, rstd, (uint16_t*)dx, dw_partial, (const uint16_t*)dres, T_rows, (int)H);
+        else if (H % 4 == 0)
+            DOL_NORM_DISPATCH(norm_bwd_kernel, float, 4, 0,
+                              (const float*)dy, (const float*)s, (const float*)w,
+                              nullptr, rstd, (float*)dx, dw_partial, (const float*)dres, T_rows, (int)H);
+        else
+            DOL_NORM_DISPATCH(norm_bwd_kernel, float, 1, 0,
+                              (const float*)dy, (const float*)s, (const float*)w,
+                              nullptr, rstd, (float*)dx, dw_partial, (const float*)dres, T_rows, (int)H);
+        return dol_last_error();
+    }
     if (dtype == DOLOMITE_BF16 && H % 8 == 0)
-        DOL_NORM_DISPATCH(norm_bwd_kernel, uint16_t, 8, false,
+        DOL_NORM_DISPATCH(norm_bwd_kernel, uint16_t, 8, 2,
                           (const uint16_t*)dy, (const uint16_t*)s, (const uint16_t*)w,
                           nullptr, rstd, (uint16_t*)dx, dw_partial, (const uint16_t*)dres, T_rows, (int)H);
     else if (dtype == DOLOMITE_BF16)
-        DOL_NORM_DISPATCH(norm_bwd_kernel, uint16_t, 1, false,
+        DOL_NORM_DISPATCH(norm_bwd_kernel, uint16_t, 1, 2,
                           (const uint16_t*)dy, (const uint16_t*)s, (const uint16_t*)w,
                           nullptr, rstd, (uint16_t*)dx, dw_partial, (const uint16_t*)dres, T_rows, (int)H);
     else if (H % 4 == 0)
-        DOL_NORM_DISPATCH(norm_bwd_kernel, float, 4, false,
+        DOL_NORM_DISPATCH(norm_bwd_kernel, float, 4, 2,
                           (const float*)dy, (const float*)s, (const float*)w,
                           nullptr, rstd, (float*)dx, dw_partial, (const float*)dres, T_rows, (int)H);
     else
-        DOL_NORM_DISPATCH(norm_bwd_kernel, float, 1, false,
+        DOL_NORM_DISPATCH(norm_bwd_kernel, float, 1, 2,
                           (const float*)dy, (const float*)s, (const float*)w,
                           nullptr, rstd, (float*)dx, dw_partial, (const float*)dres, T_rows, (int)H);
     return dol_last_error();
@@ -426,20 +447,39 @@ extern "C" int dolomite_layernorm_bwd(dolomite_stream_t stream,
     if (T_rows == 0) return 0;
     hipStream_t stream_ = (hipStream_t)stream;
     dim3 grid(RMS_BWD_BLOCKS), block(256);
+    if (dres == nullptr) {
+        if (dtype == DOLOMITE_BF16 && H % 8 == 0)
+            DOL_NORM_DISPATCH(norm_bwd_kernel, uint16_t, 8, 1,
+                              (const uint16_t*)dy, (const uint16_t*)s, (const uint16_t*)w,
+                              mean, rstd, (uint16_t*)dx, dwdb_partial, (const uint16_t*)dres, T_rows, (int)H);
+        else if (dtype == DOLOMITE_BF16)
+            DOL_NORM_DISPATCH(norm_bwd_kernel, uint16_t, 1, 1,
+                              (const uint16_t*)dy, (const uint16_t*)s, (const uint16_t*)w,
+                              mean, rstd, (uint16_t*)dx, dwdb_partial, (const uint16_t*)dres, T_rows, (int)H);
+        else if (H % 4 == 0)
+            DOL_NORM_DISPATCH(norm_bwd_kernel, float, 4, 1,
+                              (const float*)dy, (const float*)s, (const float*)w,
+                              mean, rstd, (float*)dx, dwdb_partial, (const float*)dres, T_rows, (int)H);
+        else
+            DOL_NORM_DISPATCH(norm_bwd_kernel, float, 1, 1,
+                              (const float*)dy, (const float*)s, (const float*)w,
+                              mean, rstd, (float*)dx, dwdb_partial, (const float*)dres, T_rows, (int)H);
+        return dol_last_error();
+    }
     if (dtype == DOLOMITE_BF16 && H % 8 == 0)
-        DOL_NORM_DISPATCH(norm_bwd_kernel, uint16_t, 8, true,
+        DOL_NORM_DISPATCH(norm_bwd_kernel, uint16_t, 8, 3,
                           (const uint16_t*)dy, (const uint16_t*)s, (const uint16_t*)w,
                           mean, rstd, (uint16_t*)dx, dwdb_partial, (const uint16_t*)dres, T_rows, (int)H);
     else if (dtype == DOLOMITE_BF16)
-        DOL_NORM_DISPATCH(norm_bwd_kernel, uint16_t, 1, true,
+        DOL_NORM_DISPATCH(norm_bwd_kernel, uint16_t, 1, 3,
                           (const uint16_t*)dy, (const uint16_t*)s, (const uint16_t*)w,
                           mean, rstd, (uint16_t*)dx, dwdb_partial, (const uint16_t*)dres, T_rows, (int)H);
     else if (H % 4 == 0)
-        DOL_NORM_DISPATCH(norm_bwd_kernel, float, 4, true,
+        DOL_NORM_DISPATCH(norm_bwd_kernel, float, 4, 3,
                           (const float*)dy, (const float*)s, (const float*)w,
                           mean, rstd, (float*)dx, dwdb_partial, (const float*)dres, T_rows, (int)H);
     else
-        DOL_NORM_DISPATCH(norm_bwd_kernel, float, 1, true,
+        DOL_NORM_DISPATCH(norm_bwd_kernel, float, 1, 3,
                           (const float*)dy, (const float*)s, (const float*)w,
                           mean, rstd, (float*)dx, dwdb_partial, (const float*)dres, T_rows, (int)H);
     return dol_last_error();
