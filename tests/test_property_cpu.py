@@ -99,3 +99,88 @@ def test_document_boundary_reset_property(tokens, eos):
     seg_lens = [b - a for a, b in zip(ref_cu, ref_cu[1:])]
     assert max_len == max(seg_lens)
     assert pos.tolist() == [i for n in seg_lens for i in range(n)]
+
+
+@given(
+    st.integers(1, 6),       # rows
+    st.sampled_from([8, 64, 96, 2560]),
+    st.sampled_from(["fp32", "bf16"]),
+)
+def test_rmsnorm_cpu_branch_matches_torch(T, H, dtype):
+    """The CPU branch of FusedRMSNorm (the branch the gloo training tests
+    run) vs torch's own rms_norm at random shapes, fwd and bwd."""
+    dt = torch.float32 if dtype == "fp32" else torch.bfloat16
+    g = torch.Generator().manual_seed(T * H)
+    x = torch.randn(T, H, generator=g).to(dt).requires_grad_(True)
+    w = (torch.randn(H, generator=g) * 0.1 + 1.0).to(dt).requires_grad_(True)
+
+    from dolomite_engine_amd.ops import fused_rmsnorm
+
+    y, _ = fused_rmsnorm(x, w, 1e-6)
+    dy = torch.randn(T, H, generator=g).to(dt)
+    y.backward(dy)
+
+    xr = x.detach().clone().requires_grad_(True)
+    wr = w.detach().clone().requires_grad_(True)
+    # reference semantics (rmsnorm/base.py:18-25): fp32 accum, cast BEFORE w
+    x32 = xr.float()
+    yr = wr * (x32 * torch.rsqrt(x32.pow(2).mean(-1, keepdim=True) + 1e-6)).to(dt)
+    yr.backward(dy)
+
+    tol = dict(rtol=1e-6, atol=1e-6) if dtype == "fp32" else dict(rtol=2e-2, atol=2e-2)
+    torch.testing.assert_close(y, yr, **tol)
+    torch.testing.assert_close(x.grad, xr.grad, **tol)
+    torch.testing.assert_close(w.grad, wr.grad, **tol)
+
+
+@given(head_configs(), st.integers(2, 12))
+def test_rope_forward_inverse_roundtrip(cfg, T):
+    """RoPE packed apply then inverse apply is the identity (the backward
+    rotation used by RoPEPackedQKV.backward is exact: R^T = -R, duplicated
+    table halves commute)."""
+    from dolomite_engine_amd.ops.functional import RoPEPackedQKV
+
+    head_type, H, Hkv, D = cfg
+    if D % 2 != 0:
+        return
+    lo = QKVLayout.make(H, Hkv, D, head_type)
+    g = torch.Generator().manual_seed(T)
+    qkv = torch.randn(T, lo.row_len, generator=g)
+    pos = torch.arange(T, dtype=torch.float32)
+    inv = 1.0 / (10000.0 ** (torch.arange(0, D, 2, dtype=torch.float32) / D))
+    fr = torch.outer(pos, inv)
+    emb = torch.cat((fr, fr), dim=-1)
+    cos, sin = emb.cos(), emb.sin()
+
+    rotated = RoPEPackedQKV.apply(qkv.clone(), cos, sin, lo)
+    # inverse rotation = apply with negated sin
+    restored = RoPEPackedQKV.apply(rotated.clone(), cos, -sin, lo)
+    torch.testing.assert_close(restored, qkv, rtol=1e-5, atol=1e-5)
+    # v slots must be untouched by the forward rotation
+    _, _, v0 = lo.unpack_cpu(qkv)
+    _, _, v1 = lo.unpack_cpu(rotated)
+    torch.testing.assert_close(v0, v1, rtol=0, atol=0)
+
+
+@given(st.integers(2, 40), st.sampled_from([17, 512, 1000]), st.integers(0, 3))
+def test_fused_ce_cpu_matches_torch(T, V, n_ignore):
+    """CPU branch of FusedCrossEntropy vs F.cross_entropy(mean, -100),
+    forward and dlogits, at random shapes with ignored labels."""
+    import torch.nn.functional as F
+
+    from dolomite_engine_amd.ops import fused_cross_entropy
+
+    g = torch.Generator().manual_seed(T * V + n_ignore)
+    logits = torch.randn(T, V, generator=g).requires_grad_(True)
+    labels = torch.randint(0, V, (T,), generator=g)
+    labels[:n_ignore] = -100
+    if (labels != -100).sum() == 0:
+        return
+    loss = fused_cross_entropy(logits, labels)
+    loss.backward()
+
+    lr = logits.detach().clone().requires_grad_(True)
+    ref = F.cross_entropy(lr, labels, ignore_index=-100)
+    ref.backward()
+    torch.testing.assert_close(loss, ref, rtol=1e-6, atol=1e-6)
+    torch.testing.assert_close(logits.grad, lr.grad, rtol=1e-5, atol=1e-7)
