@@ -89,6 +89,62 @@ __device__ __forceinline__ bf16x8 lds_tr16_bfrag(const __bf16* img, int rowbase,
     return out;
 }
 
+// ---------------------------------------------------------------------------
+// Pipelined tr16 ladder (guide §5.7 forms (i)/(ii) + T3/T4 counted waits).
+//
+// Round 1 measured that per-fragment `lgkmcnt(0)` drains serialize the MFMA
+// stream (each tr16 read's 50-cycle latency lands on the critical path).
+// The ladder splits issue and wait: fragment i+1's reads are issued BEFORE
+// waiting on fragment i with a counted `lgkmcnt(N)` (N = the just-issued
+// reads), so LDS latency hides under the current fragment's MFMAs and the
+// counter never drains to 0 inside the loop.
+//
+// Count discipline: the ladder's waits are exact only if no OTHER lgkm op
+// is outstanding, so each region (a) opens with a full drain that names the
+// already-loaded operands (strip A-fragments), and (b) is fenced with
+// sched_barrier(0) so the compiler cannot move its own ds/smem ops inside.
+// ---------------------------------------------------------------------------
+
+// Issue one B-fragment's two transpose reads (no wait): OFF = byte offset of
+// the (rowbase, d0) chunk inside the image, a0/a1 = the lane's two PI23 row
+// base addresses (computed once per kernel).
+template <int OFF>
+__device__ __forceinline__ void tr16_issue(unsigned a0, unsigned a1, bf16x4& lo, bf16x4& hi) {
+    asm volatile(
+        "ds_read_b64_tr_b16 %0, %2 offset:%c4\n\t"
+        "ds_read_b64_tr_b16 %1, %3 offset:%c4"
+        : "=&v"(lo), "=&v"(hi)
+        : "v"(a0), "v"(a1), "i"(OFF)
+        : "memory");
+}
+
+// Counted wait naming the registers it guarantees (guide §5.7 form (ii)).
+template <int N>
+__device__ __forceinline__ void lgkm_wait2(bf16x4& a, bf16x4& b) {
+    asm volatile("s_waitcnt lgkmcnt(%c2)" : "+v"(a), "+v"(b) : "i"(N));
+}
+template <int N>
+__device__ __forceinline__ void lgkm_wait4(bf16x4& a, bf16x4& b, bf16x4& c, bf16x4& d) {
+    asm volatile("s_waitcnt lgkmcnt(%c4)" : "+v"(a), "+v"(b), "+v"(c), "+v"(d) : "i"(N));
+}
+
+typedef __bf16 bf16x8v __attribute__((ext_vector_type(8)));
+__device__ __forceinline__ bf16x8 tr16_join8(bf16x4 lo, bf16x4 hi) {
+    bf16x8 out;
+#pragma unroll
+    for (int t = 0; t < 4; ++t) { out[t] = lo[t]; out[4 + t] = hi[t]; }
+    return out;
+}
+
+// Full drain opening a ladder region; names the strip fragments the region's
+// MFMAs consume so their (compiler-issued) reads are complete and counted out.
+__device__ __forceinline__ void lgkm_drain2x8(bf16x8& a, bf16x8& b) {
+    asm volatile("s_waitcnt lgkmcnt(0)" : "+v"(a), "+v"(b));
+}
+__device__ __forceinline__ void lgkm_drain4x8(bf16x8& a, bf16x8& b, bf16x8& c, bf16x8& d) {
+    asm volatile("s_waitcnt lgkmcnt(0)" : "+v"(a), "+v"(b), "+v"(c), "+v"(d));
+}
+
 // Branchless guarded 16B load. Requires D % 8 == 0 (every head dim on this
 // path): a chunk is then fully inside [0, D) or fully in the pad, so the
 // guard reduces to ONE wave-divergent-free vector load from a clamped
@@ -211,6 +267,14 @@ __global__ void __launch_bounds__(512) fa_fwd_kernel(
 #pragma unroll
     for (int dc = 0; dc < DCH; ++dc) o_acc[dc] = {0.f, 0.f, 0.f, 0.f};
 
+    // lane base addresses for the PV tr16 ladder (PI23 row placement; the
+    // per-step (rowbase, d0) displacement goes in the ds_read immediate)
+    const int tr_jj = (lane >> 2) & 3;
+    const int tr_g8 = (lane >> 4) * 8;
+    const int tr_cc = (lane & 3) * 4;
+    const unsigned aV0 = (unsigned)(size_t)(Vlds + PI23(tr_g8 + tr_jj) * SK + tr_cc);
+    const unsigned aV1 = (unsigned)(size_t)(Vlds + PI23(tr_g8 + 4 + tr_jj) * SK + tr_cc);
+
     const int kend = min(L, qs + 128);
     const int ntiles = (kend + 63) / 64;
 
@@ -285,15 +349,34 @@ __global__ void __launch_bounds__(512) fa_fwd_kernel(
 #pragma unroll
             for (int r = 0; r < 4; ++r) Pw[(lg * 4 + r) * SV + cb * 16 + lr] = (__bf16)p_val[cb][r];
 
-        // --- PV: A = P (this wave's rows), B via tr16 reads of row-major V ---
-#pragma unroll
-        for (int kc2 = 0; kc2 < 2; ++kc2) {
-            bf16x8 pf = *(const bf16x8*)&Pw[lr * SV + kc2 * 32 + lg * 8];
-#pragma unroll
-            for (int dc = 0; dc < DCH; ++dc) {
-                bf16x8 vf = lds_tr16_bfrag(Vlds, kc2 * 32, SK, dc * 16, lane);
-                o_acc[dc] = MFMA16(pf, vf, o_acc[dc]);
-            }
+        // --- PV: A = P (this wave's rows), B via the pipelined tr16 ladder
+        // over the row-major PI23 V image (issue frag i+1, counted-wait
+        // frag i, MFMA — LDS latency hides under the matrix pipe) ---
+        {
+            bf16x8 pf0 = *(const bf16x8*)&Pw[lr * SV + lg * 8];
+            bf16x8 pf1 = *(const bf16x8*)&Pw[lr * SV + 32 + lg * 8];
+            lgkm_drain2x8(pf0, pf1);
+            __builtin_amdgcn_sched_barrier(0);
+            bf16x4 vlo[2], vhi[2];
+            tr16_issue<0>(aV0, aV1, vlo[0], vhi[0]);
+#define FWD_PV_STEP(i)                                                                                  \
+    if constexpr ((i) < 2 * DCH) {                                                                      \
+        constexpr int kc2_ = (i) / DCH, dc_ = (i) % DCH;                                                \
+        if constexpr ((i) + 1 < 2 * DCH) {                                                              \
+            constexpr int kn_ = ((i) + 1) / DCH, dn_ = ((i) + 1) % DCH;                                 \
+            tr16_issue<(kn_ * 32 * SK + dn_ * 16) * 2>(aV0, aV1, vlo[((i) + 1) & 1], vhi[((i) + 1) & 1]); \
+            lgkm_wait2<2>(vlo[(i) & 1], vhi[(i) & 1]);                                                  \
+        } else {                                                                                        \
+            lgkm_wait2<0>(vlo[(i) & 1], vhi[(i) & 1]);                                                  \
+        }                                                                                               \
+        o_acc[dc_] = MFMA16(kc2_ ? pf1 : pf0, tr16_join8(vlo[(i) & 1], vhi[(i) & 1]), o_acc[dc_]);      \
+    }
+            FWD_PV_STEP(0) FWD_PV_STEP(1) FWD_PV_STEP(2) FWD_PV_STEP(3)
+            FWD_PV_STEP(4) FWD_PV_STEP(5) FWD_PV_STEP(6) FWD_PV_STEP(7)
+            FWD_PV_STEP(8) FWD_PV_STEP(9) FWD_PV_STEP(10) FWD_PV_STEP(11)
+            FWD_PV_STEP(12) FWD_PV_STEP(13) FWD_PV_STEP(14) FWD_PV_STEP(15)
+#undef FWD_PV_STEP
+            __builtin_amdgcn_sched_barrier(0);
         }
         __syncthreads();  // K/V LDS reused next tile
     }
@@ -408,8 +491,22 @@ extern "C" int dolomite_fa_bwd_preprocess(dolomite_stream_t stream,
 //     registers over the kv-tile loop — no atomics at all — and is stored
 //     once. K^T is the only LDS image (swizzled), staged per kv tile.
 
+// dkv round-2 structure: NO transposed LDS images. The Q/dO images are
+// stored row-major with PI23 row placement only, read two ways:
+//   - S^T/dP^T B-fragments: b128 row reads (row cb*16+PI23(lr), conflict-free
+//     at stride DPAD+16 per tools_lds_sim.py);
+//   - dK/dV B-fragments: ds_read_b64_tr_b16 ladder (same PI23 image).
+// This removes 2*DPAD*96 elements of LDS (102 KB -> 64 KB at DPAD=96) and
+// the 16 scalar b16 scatter-writes per staged piece. Together with the
+// __launch_bounds__(512, 4) register cap (<=128 VGPR) the kernel reaches
+// 2 workgroups/CU: while one workgroup stages Q/dO from HBM the co-resident
+// one runs its MFMA segments — the cross-workgroup latency hiding the
+// round-1 single-occupancy version lacked.
+// Occupancy bound: <=96 head-dim fits the 128-VGPR / 64-KB-LDS budget for
+// 2 workgroups/CU (measured: 128 VGPR, 3 prologue spills); the 128 variant
+// does not (43 spills into the hot loop at the cap) and keeps 1 WG/CU.
 template <int DPAD>
-__global__ void __launch_bounds__(512, 2) fa_bwd_dkv_kernel(
+__global__ void __launch_bounds__(512, DPAD <= 96 ? 4 : 2) fa_bwd_dkv_kernel(
     const __bf16* __restrict__ q, const __bf16* __restrict__ k, const __bf16* __restrict__ v,
     const __bf16* __restrict__ dout, const float* __restrict__ lse, const float* __restrict__ delta,
     float* __restrict__ dk_acc, float* __restrict__ dv_acc,
@@ -434,18 +531,12 @@ __global__ void __launch_bounds__(512, 2) fa_bwd_dkv_kernel(
     const int lr = lane & 15;
     const int lg = lane >> 4;
 
-    constexpr int SQ = DPAD + 16;              // row-major image stride
-    constexpr int TST = 96;                    // transposed image stride
+    constexpr int SQ = DPAD + 16;              // PI23-rowed image stride
     extern __shared__ char smem_raw[];
-    __bf16* QTl = (__bf16*)smem_raw;           // [DPAD][TST]   (Q^T, swizzled)
-    __bf16* dOTl = QTl + DPAD * TST;           // [DPAD][TST]   (dO^T, swizzled)
-    __bf16* dSTl = dOTl + DPAD * TST;          // [128 key][ST] (dS^T, [key][q])
+    __bf16* Qlds = (__bf16*)smem_raw;          // [64 q][SQ] (PI23 rows)
+    __bf16* dOl = Qlds + 64 * SQ;              // [64 q][SQ] (PI23 rows)
+    __bf16* dSTl = dOl + 64 * SQ;              // [128 key][ST] (dS^T, [key][q])
     __bf16* PTl = dSTl + 128 * ST;             // [128 key][ST] (P^T, [key][q])
-    __bf16* Qlds = PTl + 128 * ST;             // [64 q][SQ]    (row-major)
-    __bf16* dOl = Qlds + 64 * SQ;              // [64 q][SQ]    (row-major)
-
-#define SWZ(row, col) ((row) * TST + (((((col) >> 3) ^ ((3 * ((row) >> 3)) & 7)) << 3) | ((col) & 7)))
-#define SWZ8(row, col0) ((row) * TST + (((((col0) >> 3) ^ ((3 * ((row) >> 3)) & 7)) << 3)))
 
     const int kend = min(L, ks + 128);
 
@@ -468,9 +559,18 @@ __global__ void __launch_bounds__(512, 2) fa_bwd_dkv_kernel(
         dkr[dc] = {0.f, 0.f, 0.f, 0.f};
     }
 
-    // T5 static priority (guide §5.5): at 2 waves/SIMD the later-dispatched
-    // half of an 8-wave workgroup loses VALU arbitration to the older half;
-    // one setprio(1) for that half removes its start-of-segment penalty.
+    // lane base addresses for the dK/dV tr16 ladders
+    const int tr_jj = (lane >> 2) & 3;
+    const int tr_g8 = (lane >> 4) * 8;
+    const int tr_cc = (lane & 3) * 4;
+    const unsigned aQ0 = (unsigned)(size_t)(Qlds + PI23(tr_g8 + tr_jj) * SQ + tr_cc);
+    const unsigned aQ1 = (unsigned)(size_t)(Qlds + PI23(tr_g8 + 4 + tr_jj) * SQ + tr_cc);
+    const unsigned aO0 = (unsigned)(size_t)(dOl + PI23(tr_g8 + tr_jj) * SQ + tr_cc);
+    const unsigned aO1 = (unsigned)(size_t)(dOl + PI23(tr_g8 + 4 + tr_jj) * SQ + tr_cc);
+
+    // T5 static priority (guide §5.5): the later-dispatched half of an
+    // 8-wave workgroup loses VALU arbitration to the older half; one
+    // setprio(1) for that half removes its start-of-segment penalty.
     // The guard must be provably wave-uniform (readfirstlane) or s_setprio
     // is emitted unconditionally.
     if (__builtin_amdgcn_readfirstlane(threadIdx.x) >= 256) __builtin_amdgcn_s_setprio(1);
@@ -490,42 +590,30 @@ __global__ void __launch_bounds__(512, 2) fa_bwd_dkv_kernel(
                 int d0 = (pidx % (DPAD / 8)) * 8;
                 bool valid = (qs + qq) < L;
                 const __bf16* qp = q + (int64_t)(s0 + (valid ? qs + qq : 0)) * q_ts + q_hoff + d0;
-                bf16x8 qq8 = load_bf16x8_guard(qp, d0, D, valid);
-                *(bf16x8*)&Qlds[qq * SQ + d0] = qq8;
-#pragma unroll
-                for (int e = 0; e < 8; ++e) QTl[SWZ(d0 + e, qq)] = qq8[e];
+                *(bf16x8*)&Qlds[PI23(qq) * SQ + d0] = load_bf16x8_guard(qp, d0, D, valid);
                 const __bf16* dp = dout + (int64_t)(s0 + (valid ? qs + qq : 0)) * do_ts + do_hoff + d0;
-                bf16x8 dd8 = load_bf16x8_guard(dp, d0, D, valid);
-                *(bf16x8*)&dOl[qq * SQ + d0] = dd8;
-#pragma unroll
-                for (int e = 0; e < 8; ++e) dOTl[SWZ(d0 + e, qq)] = dd8[e];
+                *(bf16x8*)&dOl[PI23(qq) * SQ + d0] = load_bf16x8_guard(dp, d0, D, valid);
             }
         }
         __syncthreads();
 
-        // S^T = K*Q^T, dP^T = V*dO^T; B-frags (identical across the 4
-        // waves) come from the row-major LDS images: Q[q=lr][d0..d0+8)
-        f32x4 st[4], dpt[4];
-#pragma unroll
-        for (int cb = 0; cb < 4; ++cb) {
-            st[cb] = {0.f, 0.f, 0.f, 0.f};
-            dpt[cb] = {0.f, 0.f, 0.f, 0.f};
-#pragma unroll
-            for (int kc = 0; kc < KCH; ++kc) {
-                int d0 = kc * 32 + lg * 8;
-                bf16x8 qb = *(const bf16x8*)&Qlds[(cb * 16 + lr) * SQ + d0];
-                st[cb] = MFMA16(kfr[kc], qb, st[cb]);
-                bf16x8 db = *(const bf16x8*)&dOl[(cb * 16 + lr) * SQ + d0];
-                dpt[cb] = MFMA16(vfr[kc], db, dpt[cb]);
-            }
-        }
-
-        // P^T, dS^T -> own-wave LDS strips (no barrier: each wave reads only
-        // the 16-key rows it wrote)
-        // full tiles (qs entirely below this wave's key strip, no clipping)
+        // Per q-column block: S^T = K*Q^T and dP^T = V*dO^T MFMAs, then that
+        // block's softmax straight into the LDS strips. Merging the two
+        // loops keeps only ONE C-fragment pair (8 VGPRs) live instead of
+        // four (32) — the register headroom the (512,4) cap needs.
         const bool full_tile = (ks + wave * 16 + 16 <= qs + 1) && (kend == ks + 64) && (qs + 64 <= L);
 #pragma unroll
         for (int cb = 0; cb < 4; ++cb) {
+            f32x4 st = {0.f, 0.f, 0.f, 0.f};
+            f32x4 dpt = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+            for (int kc = 0; kc < KCH; ++kc) {
+                int d0 = kc * 32 + lg * 8;
+                bf16x8 qb = *(const bf16x8*)&Qlds[(cb * 16 + PI23(lr)) * SQ + d0];
+                st = MFMA16(kfr[kc], qb, st);
+                bf16x8 db = *(const bf16x8*)&dOl[(cb * 16 + PI23(lr)) * SQ + d0];
+                dpt = MFMA16(vfr[kc], db, dpt);
+            }
             const int qpos = qs + cb * 16 + lr;
             const bool qok = qpos < L;
             float lsev = qok ? lse[(int64_t)h * T_total + s0 + qpos] * DOL_LOG2E : 0.f;
@@ -534,25 +622,46 @@ __global__ void __launch_bounds__(512, 2) fa_bwd_dkv_kernel(
             for (int r = 0; r < 4; ++r) {
                 const int kpos = ks + wave * 16 + lg * 4 + r;
                 bool ok = full_tile || (qok && kpos < kend && kpos <= qpos);
-                float pv = ok ? exp2f(st[cb][r] * (scale * DOL_LOG2E) - lsev) : 0.f;
-                float ds = ok ? pv * (dpt[cb][r] - delv) * scale : 0.f;
+                float pv = ok ? exp2f(st[r] * (scale * DOL_LOG2E) - lsev) : 0.f;
+                float ds = ok ? pv * (dpt[r] - delv) * scale : 0.f;
                 PTl[(wave * 16 + lg * 4 + r) * ST + cb * 16 + lr] = (__bf16)pv;
                 dSTl[(wave * 16 + lg * 4 + r) * ST + cb * 16 + lr] = (__bf16)ds;
             }
         }
 
-        // dV += P^T*dO ; dK += dS^T*Q (contraction over q)
-#pragma unroll
-        for (int kc2 = 0; kc2 < 2; ++kc2) {
-            bf16x8 ptf = *(const bf16x8*)&PTl[(wave * 16 + lr) * ST + kc2 * 32 + lg * 8];
-            bf16x8 dstf = *(const bf16x8*)&dSTl[(wave * 16 + lr) * ST + kc2 * 32 + lg * 8];
-#pragma unroll
-            for (int dc = 0; dc < DCH; ++dc) {
-                bf16x8 dotb = *(const bf16x8*)&dOTl[SWZ8(dc * 16 + lr, kc2 * 32 + lg * 8)];
-                dvr[dc] = MFMA16(ptf, dotb, dvr[dc]);
-                bf16x8 qtb = *(const bf16x8*)&QTl[SWZ8(dc * 16 + lr, kc2 * 32 + lg * 8)];
-                dkr[dc] = MFMA16(dstf, qtb, dkr[dc]);
-            }
+        // dV += P^T*dO ; dK += dS^T*Q (contraction over q) — pipelined tr16
+        // ladder: issue pair i+1 (4 reads), counted-wait pair i, 2 MFMAs.
+        {
+            bf16x8 ptf0 = *(const bf16x8*)&PTl[(wave * 16 + lr) * ST + lg * 8];
+            bf16x8 dstf0 = *(const bf16x8*)&dSTl[(wave * 16 + lr) * ST + lg * 8];
+            bf16x8 ptf1 = *(const bf16x8*)&PTl[(wave * 16 + lr) * ST + 32 + lg * 8];
+            bf16x8 dstf1 = *(const bf16x8*)&dSTl[(wave * 16 + lr) * ST + 32 + lg * 8];
+            lgkm_drain4x8(ptf0, dstf0, ptf1, dstf1);
+            __builtin_amdgcn_sched_barrier(0);
+            bf16x4 olo[2], ohi[2], qlo[2], qhi[2];
+            tr16_issue<0>(aO0, aO1, olo[0], ohi[0]);
+            tr16_issue<0>(aQ0, aQ1, qlo[0], qhi[0]);
+#define DKV_STEP(i)                                                                                     \
+    if constexpr ((i) < 2 * DCH) {                                                                      \
+        constexpr int kc2_ = (i) / DCH, dc_ = (i) % DCH;                                                \
+        if constexpr ((i) + 1 < 2 * DCH) {                                                              \
+            constexpr int kn_ = ((i) + 1) / DCH, dn_ = ((i) + 1) % DCH;                                 \
+            constexpr int off_ = (kn_ * 32 * SQ + dn_ * 16) * 2;                                        \
+            tr16_issue<off_>(aO0, aO1, olo[((i) + 1) & 1], ohi[((i) + 1) & 1]);                         \
+            tr16_issue<off_>(aQ0, aQ1, qlo[((i) + 1) & 1], qhi[((i) + 1) & 1]);                         \
+            lgkm_wait4<4>(olo[(i) & 1], ohi[(i) & 1], qlo[(i) & 1], qhi[(i) & 1]);                      \
+        } else {                                                                                        \
+            lgkm_wait4<0>(olo[(i) & 1], ohi[(i) & 1], qlo[(i) & 1], qhi[(i) & 1]);                      \
+        }                                                                                               \
+        dvr[dc_] = MFMA16(kc2_ ? ptf1 : ptf0, tr16_join8(olo[(i) & 1], ohi[(i) & 1]), dvr[dc_]);        \
+        dkr[dc_] = MFMA16(kc2_ ? dstf1 : dstf0, tr16_join8(qlo[(i) & 1], qhi[(i) & 1]), dkr[dc_]);      \
+    }
+            DKV_STEP(0) DKV_STEP(1) DKV_STEP(2) DKV_STEP(3)
+            DKV_STEP(4) DKV_STEP(5) DKV_STEP(6) DKV_STEP(7)
+            DKV_STEP(8) DKV_STEP(9) DKV_STEP(10) DKV_STEP(11)
+            DKV_STEP(12) DKV_STEP(13) DKV_STEP(14) DKV_STEP(15)
+#undef DKV_STEP
+            __builtin_amdgcn_sched_barrier(0);
         }
     }
 
@@ -570,12 +679,10 @@ __global__ void __launch_bounds__(512, 2) fa_bwd_dkv_kernel(
             }
         }
     }
-#undef SWZ
-#undef SWZ8
 }
 
 template <int DPAD>
-__global__ void __launch_bounds__(512) fa_bwd_dq_kernel(
+__global__ void __launch_bounds__(512, 4) fa_bwd_dq_kernel(
     const __bf16* __restrict__ q, const __bf16* __restrict__ k, const __bf16* __restrict__ v,
     const __bf16* __restrict__ dout, const float* __restrict__ lse, const float* __restrict__ delta,
     __bf16* __restrict__ dqkv_q,
@@ -601,20 +708,25 @@ __global__ void __launch_bounds__(512) fa_bwd_dq_kernel(
     const int lr = lane & 15;
     const int lg = lane >> 4;
 
-    constexpr int SQ = DPAD + 16;              // row-major image stride
-    constexpr int TST = 96;                    // transposed image stride
+    // Round-2: the swizzled K^T image is gone — the dQ+=dS*K B-fragments
+    // come from the PI23-rowed row-major K image via the tr16 ladder, like
+    // dkv. Saves DPAD*96 LDS elements and the per-piece scatter writes.
+    constexpr int SQ = DPAD + 16;              // image stride
     extern __shared__ char smem_raw[];
-    __bf16* KTl = (__bf16*)smem_raw;           // [DPAD][TST] (K^T, swizzled)
-    __bf16* Klds = KTl + DPAD * TST;           // [64 key][SQ] (row-major)
-    __bf16* Vlds = Klds + 64 * SQ;             // [64 key][SQ] (row-major)
+    __bf16* Klds = (__bf16*)smem_raw;          // [64 key][SQ] (PI23 rows)
+    __bf16* Vlds = Klds + 64 * SQ;             // [64 key][SQ] (natural rows)
     __bf16* dSl = Vlds + 64 * SQ;              // [8 waves][16 q][ST] (dS strips)
     __bf16* dSw = dSl + wave * 16 * ST;
 
-#define SWZ(row, col) ((row) * TST + (((((col) >> 3) ^ ((3 * ((row) >> 3)) & 7)) << 3) | ((col) & 7)))
-#define SWZ8(row, col0) ((row) * TST + (((((col0) >> 3) ^ ((3 * ((row) >> 3)) & 7)) << 3)))
-
     const int64_t q_hoff = (int64_t)(h / G) * q_gs + (int64_t)(h % G) * D;
     const int64_t do_hoff = (int64_t)h * D;
+
+    // lane base addresses for the dQ tr16 ladder over Klds
+    const int tr_jj = (lane >> 2) & 3;
+    const int tr_g8 = (lane >> 4) * 8;
+    const int tr_cc = (lane & 3) * 4;
+    const unsigned aK0 = (unsigned)(size_t)(Klds + PI23(tr_g8 + tr_jj) * SQ + tr_cc);
+    const unsigned aK1 = (unsigned)(size_t)(Klds + PI23(tr_g8 + 4 + tr_jj) * SQ + tr_cc);
 
     // this wave's Q and dO fragments (A-layout: i = lr -> q row)
     const int qrow = qs + wave * 16 + lr;
@@ -628,8 +740,6 @@ __global__ void __launch_bounds__(512) fa_bwd_dq_kernel(
         const __bf16* dp = dout + (int64_t)(s0 + (qvalid ? qrow : 0)) * do_ts + do_hoff + d0;
         dfr[kc] = load_bf16x8_guard(dp, d0, D, qvalid);
     }
-    const int qpos_r[1] = {};  // (silence unused in some instantiations)
-    (void)qpos_r;
 
     float lsev[4], delv[4];
 #pragma unroll
@@ -649,8 +759,8 @@ __global__ void __launch_bounds__(512) fa_bwd_dq_kernel(
 
     for (int kt = 0; kt < nkt; ++kt) {
         const int ks = kt * 64;
-        __syncthreads();  // previous tile's K^T reads done
-        // stage K^T (swizzled) + row-major K/V images, cooperative
+        __syncthreads();  // previous tile's image reads done
+        // stage row-major K (PI23 rows) + V images, cooperative
         {
             const int pieces = 64 * DPAD / 8;
             for (int pidx = threadIdx.x; pidx < pieces; pidx += 512) {
@@ -658,57 +768,66 @@ __global__ void __launch_bounds__(512) fa_bwd_dq_kernel(
                 int d0 = (pidx % (DPAD / 8)) * 8;
                 bool valid = (ks + key) < kend_total;
                 const __bf16* kp = k + (int64_t)(s0 + (valid ? ks + key : 0)) * k_ts + (int64_t)kvh * k_hs + d0;
-                bf16x8 kk = load_bf16x8_guard(kp, d0, D, valid);
-                *(bf16x8*)&Klds[key * SQ + d0] = kk;
-#pragma unroll
-                for (int e = 0; e < 8; ++e) KTl[SWZ(d0 + e, key)] = kk[e];
+                *(bf16x8*)&Klds[PI23(key) * SQ + d0] = load_bf16x8_guard(kp, d0, D, valid);
                 const __bf16* vp = v + (int64_t)(s0 + (valid ? ks + key : 0)) * v_ts + (int64_t)kvh * v_hs + d0;
                 *(bf16x8*)&Vlds[key * SQ + d0] = load_bf16x8_guard(vp, d0, D, valid);
             }
         }
         __syncthreads();
 
-        // S = Q*K^T and dP = dO*V^T; B-frags are K/V rows from the LDS images
-        f32x4 sc[4], dp[4];
-#pragma unroll
-        for (int cb = 0; cb < 4; ++cb) {
-            sc[cb] = {0.f, 0.f, 0.f, 0.f};
-            dp[cb] = {0.f, 0.f, 0.f, 0.f};
-#pragma unroll
-            for (int kc = 0; kc < KCH; ++kc) {
-                int d0 = kc * 32 + lg * 8;
-                bf16x8 kb = *(const bf16x8*)&Klds[(cb * 16 + lr) * SQ + d0];
-                sc[cb] = MFMA16(qfr[kc], kb, sc[cb]);
-                bf16x8 vb = *(const bf16x8*)&Vlds[(cb * 16 + lr) * SQ + d0];
-                dp[cb] = MFMA16(dfr[kc], vb, dp[cb]);
-            }
-        }
-
-        // dS = P*(dP - delta)*scale, C-layout (row = q, col = key);
-        // stash into this wave's own dS strip (no cross-wave use)
+        // Per key-column block: S = Q*K^T and dP = dO*V^T MFMAs, then that
+        // block's dS straight into the wave's strip (one C-pair live).
         const bool full_tile = (ks + 64 <= qs + wave * 16 + 1) && (ks + 64 <= kend_total) && (qs + wave * 16 + 16 <= L);
 #pragma unroll
         for (int cb = 0; cb < 4; ++cb) {
+            f32x4 sc = {0.f, 0.f, 0.f, 0.f};
+            f32x4 dp = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+            for (int kc = 0; kc < KCH; ++kc) {
+                int d0 = kc * 32 + lg * 8;
+                bf16x8 kb = *(const bf16x8*)&Klds[(cb * 16 + PI23(lr)) * SQ + d0];
+                sc = MFMA16(qfr[kc], kb, sc);
+                bf16x8 vb = *(const bf16x8*)&Vlds[(cb * 16 + lr) * SQ + d0];
+                dp = MFMA16(dfr[kc], vb, dp);
+            }
 #pragma unroll
             for (int r = 0; r < 4; ++r) {
                 const int qpos = qs + wave * 16 + lg * 4 + r;
                 const int kpos = ks + cb * 16 + lr;
                 bool ok = full_tile || (qpos < L && kpos < kend_total && kpos <= qpos);
-                float pv = ok ? exp2f(sc[cb][r] * (scale * DOL_LOG2E) - lsev[r]) : 0.f;
-                float ds = ok ? pv * (dp[cb][r] - delv[r]) * scale : 0.f;
+                float pv = ok ? exp2f(sc[r] * (scale * DOL_LOG2E) - lsev[r]) : 0.f;
+                float ds = ok ? pv * (dp[r] - delv[r]) * scale : 0.f;
                 dSw[(lg * 4 + r) * ST + cb * 16 + lr] = (__bf16)ds;
             }
         }
 
-        // dQ += dS*K (contraction over this tile's keys)
-#pragma unroll
-        for (int kc2 = 0; kc2 < 2; ++kc2) {
-            bf16x8 dsf = *(const bf16x8*)&dSw[lr * ST + kc2 * 32 + lg * 8];
-#pragma unroll
-            for (int dc = 0; dc < DCH; ++dc) {
-                bf16x8 ktb = *(const bf16x8*)&KTl[SWZ8(dc * 16 + lr, kc2 * 32 + lg * 8)];
-                dq[dc] = MFMA16(dsf, ktb, dq[dc]);
-            }
+        // dQ += dS*K (contraction over this tile's keys) — pipelined tr16
+        // ladder over the PI23 K image
+        {
+            bf16x8 dsf0 = *(const bf16x8*)&dSw[lr * ST + lg * 8];
+            bf16x8 dsf1 = *(const bf16x8*)&dSw[lr * ST + 32 + lg * 8];
+            lgkm_drain2x8(dsf0, dsf1);
+            __builtin_amdgcn_sched_barrier(0);
+            bf16x4 klo[2], khi[2];
+            tr16_issue<0>(aK0, aK1, klo[0], khi[0]);
+#define DQ_STEP(i)                                                                                      \
+    if constexpr ((i) < 2 * DCH) {                                                                      \
+        constexpr int kc2_ = (i) / DCH, dc_ = (i) % DCH;                                                \
+        if constexpr ((i) + 1 < 2 * DCH) {                                                              \
+            constexpr int kn_ = ((i) + 1) / DCH, dn_ = ((i) + 1) % DCH;                                 \
+            tr16_issue<(kn_ * 32 * SQ + dn_ * 16) * 2>(aK0, aK1, klo[((i) + 1) & 1], khi[((i) + 1) & 1]); \
+            lgkm_wait2<2>(klo[(i) & 1], khi[(i) & 1]);                                                  \
+        } else {                                                                                        \
+            lgkm_wait2<0>(klo[(i) & 1], khi[(i) & 1]);                                                  \
+        }                                                                                               \
+        dq[dc_] = MFMA16(kc2_ ? dsf1 : dsf0, tr16_join8(klo[(i) & 1], khi[(i) & 1]), dq[dc_]);          \
+    }
+            DQ_STEP(0) DQ_STEP(1) DQ_STEP(2) DQ_STEP(3)
+            DQ_STEP(4) DQ_STEP(5) DQ_STEP(6) DQ_STEP(7)
+            DQ_STEP(8) DQ_STEP(9) DQ_STEP(10) DQ_STEP(11)
+            DQ_STEP(12) DQ_STEP(13) DQ_STEP(14) DQ_STEP(15)
+#undef DQ_STEP
+            __builtin_amdgcn_sched_barrier(0);
         }
     }
 
@@ -724,8 +843,6 @@ __global__ void __launch_bounds__(512) fa_bwd_dq_kernel(
                 dqkv_q[(int64_t)(s0 + qpos) * q_ts + q_hoff + d] = (__bf16)dq[dc][r];
         }
     }
-#undef SWZ
-#undef SWZ8
 }
 
 template <int DPAD>
@@ -739,14 +856,13 @@ static int launch_fa_bwd(hipStream_t stream, const __bf16* q, const __bf16* k, c
     dim3 block(512);
     dim3 grid(max_tiles, batch, H);
     constexpr int SQ = DPAD + 16;
-    constexpr int TST = 96;
-    size_t shmem_dkv = (size_t)(DPAD * TST * 2 + 128 * ST * 2 + 64 * SQ * 2) * sizeof(__bf16);
+    size_t shmem_dkv = (size_t)(128 * ST * 2 + 64 * SQ * 2) * sizeof(__bf16);
     hipLaunchKernelGGL((fa_bwd_dkv_kernel<DPAD>), grid, block, shmem_dkv, stream,
                        q, k, v, dout, lse, delta, dk_acc, dv_acc, cu, H, Hkv, D, G,
                        q_ts, q_gs, k_ts, k_hs, v_ts, v_hs, do_ts, T, scale);
     int err = dol_last_error();
     if (err) return err;
-    size_t shmem_dq = (size_t)(DPAD * TST + 64 * SQ * 2 + 8 * 16 * ST) * sizeof(__bf16);
+    size_t shmem_dq = (size_t)(64 * SQ * 2 + 8 * 16 * ST) * sizeof(__bf16);
     hipLaunchKernelGGL((fa_bwd_dq_kernel<DPAD>), grid, block, shmem_dq, stream,
                        q, k, v, dout, lse, delta, dqkv_q, cu, H, Hkv, D, G,
                        q_ts, q_gs, k_ts, k_hs, v_ts, v_hs, do_ts, T, scale);
