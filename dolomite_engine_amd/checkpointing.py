@@ -97,7 +97,17 @@ def load_checkpoint_for_training(
     d = _step_dir(load_path, it)
 
     sd = GPTDolomite_load_state_dict(d / "model")
-    model_wrapper.model.load_state_dict(sd, strict=False)
+    # Strict load except for tied weights the save path deduplicates
+    # (lm_head.weight aliases the embedding when tie_word_embeddings):
+    # resuming from a checkpoint with missing/renamed keys must fail loudly,
+    # not silently train from partially-initialized weights.
+    missing, unexpected = model_wrapper.model.load_state_dict(sd, strict=False)
+    tied_ok = {"lm_head.weight"} if getattr(model_wrapper.model.config, "tie_word_embeddings", True) else set()
+    bad_missing = [k for k in missing if k not in tied_ok]
+    if bad_missing or unexpected:
+        raise RuntimeError(
+            f"checkpoint/model mismatch loading {d}: missing={bad_missing} unexpected={list(unexpected)}"
+        )
     # repointed flat buffers: copy loaded values into the engine's storage
     if engine is not None:
         for b in engine.buckets:

@@ -278,6 +278,10 @@ __global__ void __launch_bounds__(512) fa_fwd_kernel(
     const int kend = min(L, qs + 128);
     const int ntiles = (kend + 63) / 64;
 
+    // T5 static priority: the later-dispatched half of an 8-wave workgroup
+    // loses VALU arbitration; one setprio for it, no per-cluster flips.
+    if (__builtin_amdgcn_readfirstlane(threadIdx.x) >= 256) __builtin_amdgcn_s_setprio(1);
+
     for (int kt = 0; kt < ntiles; ++kt) {
         const int ks = kt * 64;
         // --- cooperative staging: K -> [key][d], V -> transposed [d][key] ---
@@ -756,6 +760,9 @@ __global__ void __launch_bounds__(512, 4) fa_bwd_dq_kernel(
 
     const int kend_total = min(L, qs + 128);
     const int nkt = (kend_total + 63) / 64;
+
+    // T5 static priority for the younger workgroup half (see dkv note)
+    if (__builtin_amdgcn_readfirstlane(threadIdx.x) >= 256) __builtin_amdgcn_s_setprio(1);
 
     for (int kt = 0; kt < nkt; ++kt) {
         const int ks = kt * 64;

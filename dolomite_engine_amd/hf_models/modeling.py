@@ -695,9 +695,22 @@ def apply_gradient_checkpointing(model, gradient_checkpointing_method: str = "bl
     gradient_checkpointing/block.py:13-34: every `checkpoint_every`-th
     GPTDolomiteBlock recomputes its forward in backward)."""
     assert gradient_checkpointing_method == "block", gradient_checkpointing_method
-    idx = 0
+    # Resolve the block class from the model itself (reference block.py
+    # looks the block class up by name) so MoEDolomiteBlock and any future
+    # family is wrapped too, not just GPTDolomiteBlock.
+    block_class = None
     for module in model.modules():
-        if isinstance(module, GPTDolomiteBlock):
+        if getattr(module, "block_class", None) is not None:
+            block_class = module.block_class
+            break
+    if block_class is None:
+        block_class = GPTDolomiteBlock
+    idx = 0
+    marked = 0
+    for module in model.modules():
+        if isinstance(module, block_class):
             if idx % checkpoint_every == 0:
                 module._gradient_checkpointing = True
+                marked += 1
             idx += 1
+    assert marked > 0, f"no {block_class.__name__} blocks found to checkpoint"

@@ -145,6 +145,35 @@ def build_blending_indices(weights: numpy.ndarray, size: int) -> tuple[numpy.nda
     return dataset_index, dataset_sample_index
 
 
+def _build_document_index(documents, num_epochs, rng, separate_final_epoch: bool):
+    """Epochs-replicated shuffled document order (gpt_dataset.py:442-476).
+
+    With separate_final_epoch the last epoch's documents are shuffled on
+    their own and appended, so a partially-consumed final epoch samples a
+    contiguous (separately shuffled) tail instead of leaking into the
+    global shuffle."""
+    if not separate_final_epoch or num_epochs == 1:
+        doc_idx = numpy.tile(documents, num_epochs).astype(documents.dtype)
+        rng.shuffle(doc_idx)
+        return doc_idx
+    first = _build_document_index(documents, num_epochs - 1, rng, False)
+    last = _build_document_index(documents, 1, rng, False)
+    return numpy.concatenate((first, last))
+
+
+def _build_shuffle_index(num_samples: int, total_size: int, rng):
+    """[0, num_samples) and [num_samples, total_size) shuffled separately
+    (gpt_dataset.py:478-510)."""
+    dtype_ = numpy.uint32 if total_size < numpy.iinfo(numpy.uint32).max - 1 else numpy.int64
+    first = numpy.arange(0, num_samples, dtype=dtype_)
+    rng.shuffle(first)
+    if num_samples == total_size:
+        return first
+    last = numpy.arange(num_samples, total_size, dtype=dtype_)
+    rng.shuffle(last)
+    return numpy.concatenate((first, last))
+
+
 class GPTDataset:
     """Doc/sample/shuffle-index dataset over an MMapIndexedDataset
     (gpt_dataset.py:241-401; indices built in memory, not cached to disk)."""
@@ -168,18 +197,29 @@ class GPTDataset:
             if (tokens - 1) // seq_length >= num_samples:
                 break
 
+        # separate_final_epoch decision (gpt_dataset.py:300-317): shuffle the
+        # final epoch separately when the run consumes <80% of it
+        if num_epochs == 1:
+            separate_final_epoch = False
+            num_samples_sans_final_epoch = 0
+        else:
+            num_samples_sans_final_epoch = ((num_epochs - 1) * tokens_per_epoch - 1) // seq_length
+            num_samples_from_final_epoch = num_samples - num_samples_sans_final_epoch
+            num_samples_per_epoch = (tokens_per_epoch - 1) // seq_length
+            assert 0 <= num_samples_from_final_epoch <= num_samples_per_epoch + 1
+            separate_final_epoch = num_samples_from_final_epoch < int(0.80 * num_samples_per_epoch)
+
         rng = numpy.random.RandomState(seed)
-        # document index: epochs-replicated then shuffled (gpt_dataset.py:442-476)
-        doc_idx = numpy.tile(documents, num_epochs).astype(numpy.int32)
-        rng.shuffle(doc_idx)
+        doc_idx = _build_document_index(documents.astype(numpy.int32), num_epochs, rng, separate_final_epoch)
         self.doc_idx = doc_idx
         self.sample_idx = build_sample_idx(
             self.indexed.sequence_lengths, doc_idx, seq_length, num_epochs, tokens_per_epoch
         )
         total = self.sample_idx.shape[0] - 1
-        shuffle_idx = numpy.arange(total, dtype=numpy.uint32 if total < 2**32 - 2 else numpy.int64)
-        rng.shuffle(shuffle_idx)
-        self.shuffle_idx = shuffle_idx
+        if separate_final_epoch:
+            self.shuffle_idx = _build_shuffle_index(num_samples_sans_final_epoch, total, rng)
+        else:
+            self.shuffle_idx = _build_shuffle_index(total, total, rng)
         self.num_samples = num_samples
 
     def __len__(self):
@@ -204,6 +244,113 @@ class GPTDataset:
             text = numpy.concatenate(parts)
         assert len(text) == self.seq_length + 1, (len(text), self.seq_length + 1)
         return {"text": torch.from_numpy(text.astype(numpy.int64))}
+
+
+def parse_and_normalize_split(split: str) -> list:
+    """"949,50,1" -> [0.949, 0.05, 0.001] (blended_megatron_dataset_config.py:98-115)."""
+    import re
+
+    vals = list(map(float, re.findall(r"[.0-9]+", split)))
+    vals = vals + [0.0] * (3 - len(vals))
+    assert len(vals) == 3 and all(v >= 0.0 for v in vals)
+    total = sum(vals)
+    return [v / total for v in vals]
+
+
+def get_split_indices(split: list, num_elements: int) -> list:
+    """Document-index bounds per split (blended_megatron_dataset_builder.py:376-398)."""
+    bounds = [0]
+    for pct in split:
+        bounds.append(bounds[-1] + int(round(pct * float(num_elements))))
+    bounds[1:] = [b - (bounds[-1] - num_elements) for b in bounds[1:]]
+    assert bounds[-1] == num_elements
+    return bounds
+
+
+class BlendedDataset:
+    """Weighted interleave of GPTDatasets (blended_dataset.py:22-120): the
+    greedy max-error assignment comes from the native builder
+    (helpers.cpp:17-69 semantics via csrc)."""
+
+    def __init__(self, datasets: list, weights: list, size: int):
+        assert len(datasets) == len(weights)
+        total = sum(weights)
+        self.weights = [w / total for w in weights]
+        self.datasets = datasets
+        self.size = size
+        self.dataset_index, self.dataset_sample_index = build_blending_indices(
+            numpy.array(self.weights, dtype=numpy.float64), size
+        )
+
+    def __len__(self):
+        return self.size
+
+    def __getitem__(self, idx: int) -> dict:
+        if idx >= self.size:
+            raise IndexError(idx)
+        return self.datasets[int(self.dataset_index[idx])][int(self.dataset_sample_index[idx])]
+
+
+def train_val_test_samples(num_training_steps: int, micro_batch_size: int, gradient_accumulation_steps: int,
+                           eval_interval: int | None, eval_steps: int, world_size: int) -> tuple:
+    """Per-split sample targets (data/megatron/__init__.py:215-234)."""
+    per_step = micro_batch_size * gradient_accumulation_steps * world_size
+    train = num_training_steps * per_step
+    val = (num_training_steps // eval_interval + 1) * eval_steps * per_step if eval_interval else 0
+    test = eval_steps * per_step
+    return train, val, test
+
+
+def build_train_val_test_datasets(data_path, split: str, sizes: tuple, seq_length: int, seed: int):
+    """Build the train/val/test GPTDatasets, blended across weighted paths
+    (data/megatron/__init__.py:18-110 options 1 and 2).
+
+    data_path: "prefix" | ["prefix"] | [w0, "prefix0", w1, "prefix1", ...]
+    split: "949,50,1"-style ratio string over each dataset's documents.
+    sizes: (train_samples, val_samples, test_samples).
+    """
+    if isinstance(data_path, (str, Path)):
+        blend = [1.0, str(data_path)]
+    elif len(data_path) == 1:
+        blend = [1.0, str(data_path[0])]
+    else:
+        assert len(data_path) % 2 == 0, "weighted blend must be [w0, path0, w1, path1, ...]"
+        blend = [float(data_path[i]) if i % 2 == 0 else str(data_path[i]) for i in range(len(data_path))]
+
+    weights = [blend[i] for i in range(0, len(blend), 2)]
+    prefixes = [blend[i + 1] for i in range(0, len(blend), 2)]
+    total_w = sum(weights)
+    weights = [w / total_w for w in weights]
+    # 0.5% margin per dataset (blended_megatron_dataset_builder.py:421-426)
+    import math
+
+    sizes_per_dataset = [[int(math.ceil(s * w * 1.005)) for s in sizes] for w in weights]
+
+    split_vec = parse_and_normalize_split(split)
+    per_dataset_splits = []  # [dataset][split] -> GPTDataset | None
+    for prefix, dsizes in zip(prefixes, sizes_per_dataset):
+        indexed = MMapIndexedDataset(prefix)
+        ndocs = len(indexed)
+        bounds = get_split_indices(split_vec, ndocs)
+        per_split = []
+        for si in range(3):
+            if split_vec[si] == 0.0 or sizes[si] == 0:
+                per_split.append(None)
+            else:
+                docs = numpy.arange(bounds[si], bounds[si + 1], dtype=numpy.int32)
+                per_split.append(GPTDataset(indexed, dsizes[si], seq_length, seed=seed, documents=docs))
+        per_dataset_splits.append(per_split)
+
+    out = []
+    for si in range(3):
+        live = [ds[si] for ds in per_dataset_splits if ds[si] is not None]
+        if not live:
+            out.append(None)
+        elif len(per_dataset_splits) == 1:
+            out.append(live[0])
+        else:
+            out.append(BlendedDataset(live, weights, sizes[si]))
+    return tuple(out)
 
 
 class MegatronDataLoader:

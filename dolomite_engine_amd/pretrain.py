@@ -85,7 +85,7 @@ def evaluate(val_iter, model_wrapper, global_step: int, eval_steps: int) -> floa
 
 
 def train(args: TrainingArgs, model_wrapper, engine, lr_scheduler, train_loader, starting_step=0, metadata=None,
-          val_loader=None):
+          val_loader=None, eval_steps: int = 8):
     tp = args.training_parameters
     ga = tp.gradient_accumulation_steps
     world = get_world_size()
@@ -153,7 +153,7 @@ def train(args: TrainingArgs, model_wrapper, engine, lr_scheduler, train_loader,
             and val_loader is not None
             and global_step % tp.eval_interval == 0
         ):
-            evaluate(iter(val_loader), model_wrapper, global_step, eval_steps=8)
+            evaluate(iter(val_loader), model_wrapper, global_step, eval_steps=eval_steps)
 
         if args.save_args is not None and args.save_args.save_interval and global_step % args.save_args.save_interval == 0:
             save_checkpoint(
@@ -193,19 +193,28 @@ def main(argv=None):
     model_wrapper, engine, lr_scheduler = build_engine(args)
 
     tp = args.training_parameters
+    val_loader = None
+    eval_steps = 8
     if args.datasets and args.datasets[0].class_name == "MegatronDataset":
-        from .megatron import GPTDataset, MegatronDataLoader, MMapIndexedDataset
+        # Full reference wiring (data/megatron/__init__.py:18-110): weighted
+        # blend across data_path entries, document-level train/val/test
+        # split, eval_steps from class_args.
+        from .megatron import MegatronDataLoader, build_train_val_test_datasets, train_val_test_samples
 
         ca = args.datasets[0].class_args
-        indexed = MMapIndexedDataset(ca["data_path"])
+        eval_steps = int(ca.get("eval_steps", 8))
         world = get_world_size()
-        gpt = GPTDataset(
-            indexed,
-            num_samples=tp.num_training_steps * tp.micro_batch_size * tp.gradient_accumulation_steps * world,
-            seq_length=tp.sequence_length,
-            seed=args.random_args.seed,
+        sizes = train_val_test_samples(
+            tp.num_training_steps, tp.micro_batch_size, tp.gradient_accumulation_steps,
+            tp.eval_interval if tp.eval_during_training else None, eval_steps, world,
         )
-        train_loader = MegatronDataLoader(gpt, tp.micro_batch_size)
+        train_ds, val_ds, _test_ds = build_train_val_test_datasets(
+            ca["data_path"], ca.get("split", "969,30,1"), sizes,
+            seq_length=tp.sequence_length, seed=int(ca.get("seed", args.random_args.seed)),
+        )
+        train_loader = MegatronDataLoader(train_ds, tp.micro_batch_size)
+        if val_ds is not None and tp.eval_during_training and tp.eval_interval is not None:
+            val_loader = MegatronDataLoader(val_ds, tp.micro_batch_size)
     else:
         train_loader = SyntheticPretrainingDataLoader(
             tp.micro_batch_size, tp.sequence_length, model_wrapper.config.vocab_size, seed=args.random_args.seed
@@ -228,15 +237,15 @@ def main(argv=None):
         if not args.load_args.load_starting_iteration:
             starting_step = 0
 
-    val_loader = None
-    if args.training_parameters.eval_during_training and args.training_parameters.eval_interval is not None:
+    if val_loader is None and tp.eval_during_training and tp.eval_interval is not None:
         # held-out synthetic stream (different seed space than training)
         val_loader = SyntheticPretrainingDataLoader(
             tp.micro_batch_size, tp.sequence_length, model_wrapper.config.vocab_size,
             seed=args.random_args.seed + 777777,
         )
 
-    train(args, model_wrapper, engine, lr_scheduler, train_loader, starting_step, metadata, val_loader=val_loader)
+    train(args, model_wrapper, engine, lr_scheduler, train_loader, starting_step, metadata,
+          val_loader=val_loader, eval_steps=eval_steps)
 
 
 if __name__ == "__main__":

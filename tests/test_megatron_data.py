@@ -121,3 +121,90 @@ def test_megatron_dataloader_resume(tmp_path):
     again = [next(dl2) for _ in range(2)]
     for a, b in zip(more, again):
         torch.testing.assert_close(a["text"], b["text"], rtol=0, atol=0)
+
+
+def test_separate_final_epoch_matches_reference_semantics(tmp_path):
+    """Multi-epoch run consuming a small slice of the last epoch: the
+    document index must keep the final epoch contiguous (separately
+    shuffled) and the shuffle index must shuffle the two ranges
+    independently (reference gpt_dataset.py:300-317, 442-510)."""
+    g = numpy.random.RandomState(3)
+    docs = [g.randint(0, 100, size=20).astype(numpy.int32) for _ in range(10)]
+    ds = _write_corpus(tmp_path, docs)
+    seq_length = 16
+    tokens_per_epoch = 200
+    samples_per_epoch = (tokens_per_epoch - 1) // seq_length  # 12
+    # 3 epochs with barely any samples from the last -> separate_final_epoch
+    num_samples = 2 * samples_per_epoch + 1
+    gpt = GPTDataset(ds, num_samples=num_samples, seq_length=seq_length, seed=5)
+
+    # reference construction with the same RandomState sequence
+    rng = numpy.random.RandomState(5)
+    documents = numpy.arange(10, dtype=numpy.int32)
+    first = numpy.tile(documents, 2).astype(numpy.int32)
+    rng.shuffle(first)
+    last = documents.copy()
+    rng.shuffle(last)
+    expect_doc = numpy.concatenate([first, last])
+    numpy.testing.assert_array_equal(gpt.doc_idx, expect_doc)
+
+    total = gpt.sample_idx.shape[0] - 1
+    ns_sans_final = (2 * tokens_per_epoch - 1) // seq_length
+    f = numpy.arange(0, ns_sans_final, dtype=numpy.uint32)
+    rng.shuffle(f)
+    l = numpy.arange(ns_sans_final, total, dtype=numpy.uint32)
+    rng.shuffle(l)
+    numpy.testing.assert_array_equal(gpt.shuffle_idx, numpy.concatenate([f, l]))
+    # windows still well-formed
+    for i in (0, num_samples - 1):
+        assert len(gpt[i]["text"]) == seq_length + 1
+
+
+def test_blended_train_val_test_split(tmp_path):
+    """build_train_val_test_datasets: weighted two-corpus blend with a
+    document split; per-split datasets draw only from their document range
+    and the blend follows the greedy max-error assignment."""
+    from dolomite_engine_amd.megatron import BlendedDataset, build_train_val_test_datasets
+
+    g = numpy.random.RandomState(11)
+    for name, base in (("c0", 0), ("c1", 50000)):
+        b = MMapIndexedDatasetBuilder(tmp_path / name, dtype=numpy.int32)
+        for i in range(40):
+            b.add_document(base + g.randint(0, 1000, size=30).astype(numpy.int32))
+        b.finalize()
+
+    sizes = (64, 16, 8)
+    train, val, test = build_train_val_test_datasets(
+        [0.3, str(tmp_path / "c0"), 0.7, str(tmp_path / "c1")],
+        split="8,1,1", sizes=sizes, seq_length=16, seed=9,
+    )
+    assert isinstance(train, BlendedDataset) and isinstance(val, BlendedDataset)
+    assert len(train) == sizes[0] and len(val) == sizes[1] and len(test) == sizes[2]
+    # blend ratio of the underlying assignment approximately follows weights
+    frac1 = float(numpy.mean(train.dataset_index == 1))
+    assert 0.5 < frac1 < 0.9
+    # val documents come from the val split range only: tokens of c0 are
+    # <50000, c1 >=50000; both corpora contribute across the blend
+    seen = {int(train[i]["text"][0] >= 50000) for i in range(len(train))}
+    assert seen == {0, 1}
+    for i in range(len(val)):
+        assert len(val[i]["text"]) == 17
+
+
+def test_single_path_split_val_loader(tmp_path):
+    """Single un-weighted corpus path: split gives disjoint document
+    ranges for train/val."""
+    from dolomite_engine_amd.megatron import build_train_val_test_datasets
+
+    b = MMapIndexedDatasetBuilder(tmp_path / "c", dtype=numpy.int32)
+    for i in range(100):
+        b.add_document(numpy.full(21, i, dtype=numpy.int32))
+    b.finalize()
+    train, val, test = build_train_val_test_datasets(
+        str(tmp_path / "c"), split="90,5,5", sizes=(50, 10, 10), seq_length=20, seed=1,
+    )
+    train_docs = {int(train[i]["text"][0]) for i in range(len(train))}
+    val_docs = {int(val[i]["text"][0]) for i in range(len(val))}
+    test_docs = {int(test[i]["text"][0]) for i in range(len(test))}
+    assert train_docs and val_docs and test_docs
+    assert max(train_docs) < 90 and min(val_docs) >= 90 and val_docs.isdisjoint(test_docs)
