@@ -114,7 +114,15 @@ def load_checkpoint_for_training(
             b.master.copy_(b.flat_param[b.shard_slice].float())
 
     if load_optimizer and engine is not None:
-        engine.load_state_dict(torch.load(d / "optimizer" / f"optimizer-{rank}.pt", weights_only=False))
+        opt_dir = d / "optimizer"
+        saved = sorted(opt_dir.glob("optimizer-*.pt"), key=lambda p: int(p.stem.split("-")[1]))
+        sd0 = torch.load(saved[min(rank, len(saved) - 1)], weights_only=False)
+        if sd0["world_size"] != engine.world:
+            # world size changed: reshard from every saved rank's shard
+            all_shards = [torch.load(p, weights_only=False) for p in saved]
+            engine.load_state_dict(all_shards[0], all_shards=all_shards)
+        else:
+            engine.load_state_dict(torch.load(opt_dir / f"optimizer-{rank}.pt", weights_only=False))
     if load_lr_scheduler:
         lr_scheduler.load_state_dict(torch.load(d / "lr_scheduler.pt", weights_only=False))
     if load_rng_state:

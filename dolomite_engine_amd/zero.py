@@ -4,12 +4,15 @@ SHARD_GRAD_OP semantics) designed for RCCL over xGMI:
 
   - parameters are repointed into per-bucket flat bf16 (or fp32) buffers,
     packed in reverse registration order (≈ backward completion order);
-  - gradients accumulate into per-bucket flat fp32 buffers via
-    post-accumulate-grad hooks (fp32 accumulation ≥ the reference's
-    communication_dtype=fp32 setting);
-  - on the sync microstep, each completed bucket's fp32 grad is
-    reduce-scattered (AVG) on a side HIP stream, overlapping the rest of
-    backward (xGMI: ring RS is per-link bound, buckets sized accordingly);
+  - each param's .grad IS a view into the bucket's flat param-dtype grad
+    buffer, so autograd accumulates microstep grads in place — no per-param
+    copy kernels, exactly like the reference FSDP's FlatParameter grads
+    (accumulation in param dtype matches FSDP no_sync semantics);
+  - on the sync microstep, each completed bucket's grad is upcast to fp32
+    and reduce-scattered (AVG) on a side HIP stream, overlapping the rest
+    of backward (fp32 on the wire = the reference's
+    communication_dtype=fp32; xGMI: ring RS is per-link bound, buckets
+    sized accordingly);
   - each rank owns a contiguous shard of every bucket: fp32 master weights +
     AdamW m/v live only for the shard; the step is ONE fused HIP AdamW kernel
     per bucket (optimization/optimizer.py:74 semantics) writing updated bf16
@@ -31,15 +34,18 @@ from .utils import get_rank, get_world_size, is_initialized
 class _Bucket:
     def __init__(self, params, dtype, device, world):
         self.params = params  # list[(param, offset)]
-        numel = sum(p.numel() for p, _ in params)
-        self.numel_padded = ((numel + 4 * world - 1) // (4 * world)) * (4 * world)
+        self.numel = sum(p.numel() for p, _ in params)
+        self.numel_padded = _padded_numel(self.numel, world)
         self.shard_size = self.numel_padded // world
         self.flat_param = torch.zeros(self.numel_padded, dtype=dtype, device=device)
-        self.flat_grad = torch.zeros(self.numel_padded, dtype=torch.float32, device=device)
+        # param-dtype flat grads: every p.grad is a view, autograd
+        # accumulates in place (FSDP FlatParameter semantics)
+        self.flat_grad = torch.zeros(self.numel_padded, dtype=dtype, device=device)
         for p, off in params:
             with torch.no_grad():
                 self.flat_param[off : off + p.numel()].copy_(p.data.reshape(-1))
             p.data = self.flat_param[off : off + p.numel()].view(p.shape)
+            p.grad = self.flat_grad[off : off + p.numel()].view(p.shape)
         rank = get_rank()
         self.shard_slice = slice(rank * self.shard_size, (rank + 1) * self.shard_size)
         self.master = self.flat_param[self.shard_slice].float()
@@ -48,6 +54,10 @@ class _Bucket:
         self.shard_grad = torch.zeros_like(self.master)
         self.pending = 0
         self.comm_event = None
+
+
+def _padded_numel(numel: int, world: int) -> int:
+    return ((numel + 4 * world - 1) // (4 * world)) * (4 * world)
 
 
 class ZeRO2Engine:
@@ -76,6 +86,7 @@ class ZeRO2Engine:
         device = params[0].device
         dtype = params[0].dtype
         self.device = device
+        self._cast_scratch = None
         self.on_gpu = device.type == "cuda"
         self.overlap = overlap_comm and self.on_gpu and self.world > 1
         self.comm_stream = torch.cuda.Stream() if self.overlap else None
@@ -110,19 +121,40 @@ class ZeRO2Engine:
                 b.pending = len(b.params)
 
     def _grad_hook(self, p: torch.Tensor) -> None:
+        # autograd already accumulated into the bucket view — the hook only
+        # tracks bucket completion so the reduce can launch early. If the
+        # view was detached externally (zero_grad(set_to_none=True)),
+        # autograd created a fresh grad tensor: fold it in and repoint.
         b, off = self._param_bucket[p]
         g = p.grad
-        b.flat_grad[off : off + p.numel()].add_(g.reshape(-1).float())
-        p.grad = None
+        if g is not None:
+            view = b.flat_grad[off : off + p.numel()]
+            if g.data_ptr() != view.data_ptr():
+                view.add_(g.reshape(-1))
+                p.grad = view.view(p.shape)
         if self._sync and self.world > 1:
             b.pending -= 1
             if b.pending == 0 and self.overlap:
                 self._launch_reduce(b)
 
+    def _comm_cast(self, b: _Bucket) -> torch.Tensor:
+        # persistent fp32 staging for the wire copy — stream-ordered reuse
+        # across buckets is safe (all reduces run on the one comm stream)
+        if b.flat_grad.dtype == torch.float32:
+            return b.flat_grad
+        if self._cast_scratch is None or self._cast_scratch.numel() < b.numel_padded:
+            self._cast_scratch = torch.empty(
+                max(bb.numel_padded for bb in self.buckets), dtype=torch.float32, device=self.device
+            )
+        out = self._cast_scratch[: b.numel_padded]
+        out.copy_(b.flat_grad)
+        return out
+
     def _launch_reduce(self, b: _Bucket) -> None:
         self.comm_stream.wait_stream(torch.cuda.current_stream())
         with torch.cuda.stream(self.comm_stream):
-            dist.reduce_scatter_tensor(b.shard_grad, b.flat_grad, op=dist.ReduceOp.AVG)
+            # fp32 on the wire (reference communication_dtype=fp32)
+            dist.reduce_scatter_tensor(b.shard_grad, self._comm_cast(b), op=dist.ReduceOp.AVG)
             b.comm_event = torch.cuda.Event()
             b.comm_event.record()
 
@@ -132,10 +164,11 @@ class ZeRO2Engine:
             return
         backend = dist.get_backend()
         if backend == "nccl":
-            dist.reduce_scatter_tensor(b.shard_grad, b.flat_grad, op=dist.ReduceOp.AVG)
+            dist.reduce_scatter_tensor(b.shard_grad, self._comm_cast(b), op=dist.ReduceOp.AVG)
         else:  # gloo: no reduce_scatter — all_reduce then slice (same math)
-            dist.all_reduce(b.flat_grad, op=dist.ReduceOp.SUM)
-            b.shard_grad.copy_(b.flat_grad[b.shard_slice]).div_(self.world)
+            g32 = b.flat_grad.float()
+            dist.all_reduce(g32, op=dist.ReduceOp.SUM)
+            b.shard_grad.copy_(g32[b.shard_slice]).div_(self.world)
 
     # ---- step-side --------------------------------------------------------
 
@@ -219,8 +252,18 @@ class ZeRO2Engine:
             ],
         }
 
-    def load_state_dict(self, sd: dict) -> None:
-        assert sd["world_size"] == self.world, "optimizer resharding not implemented yet"
+    def load_state_dict(self, sd: dict, all_shards: list | None = None) -> None:
+        """Load this rank's optimizer shard. A checkpoint saved at a
+        different world size reshards (reference checkpointing.py:105-109
+        DCP-reshard equivalence): pass every saved rank's state dict in
+        `all_shards` and each bucket's full fp32 state is reassembled,
+        re-padded for this world size and re-sliced."""
+        if sd["world_size"] != self.world:
+            assert all_shards is not None and len(all_shards) == sd["world_size"], (
+                f"world size changed {sd['world_size']} -> {self.world}: resharding needs all saved shards"
+            )
+            self._load_resharded(all_shards)
+            return
         assert len(sd["buckets"]) == len(self.buckets)
         self.step_count = sd["step"]
         for b, s in zip(self.buckets, sd["buckets"]):
@@ -229,6 +272,26 @@ class ZeRO2Engine:
             b.exp_avg_sq.copy_(s["exp_avg_sq"])
             b.flat_param[b.shard_slice].copy_(b.master.to(b.flat_param.dtype))
             self._allgather_params(b)
+        self._drain_comm_events()
+
+    def _load_resharded(self, shards: list) -> None:
+        saved_world = shards[0]["world_size"]
+        assert all(s["world_size"] == saved_world for s in shards)
+        assert all(len(s["buckets"]) == len(self.buckets) for s in shards)
+        self.step_count = shards[0]["step"]
+        for i, b in enumerate(self.buckets):
+            for key, dst in (("master", b.master), ("exp_avg", b.exp_avg), ("exp_avg_sq", b.exp_avg_sq)):
+                # saved layout: bucket padded for saved_world, split into
+                # equal shards; valid data is the first b.numel elements
+                full = torch.cat([s["buckets"][i][key].to(dst.device).float() for s in shards])
+                flat = torch.zeros(b.numel_padded, dtype=torch.float32, device=dst.device)
+                flat[: b.numel].copy_(full[: b.numel])
+                dst.copy_(flat[b.shard_slice])
+            b.flat_param[b.shard_slice].copy_(b.master.to(b.flat_param.dtype))
+            self._allgather_params(b)
+        self._drain_comm_events()
+
+    def _drain_comm_events(self) -> None:
         if self.overlap:
             for b in self.buckets:
                 if b.comm_event is not None:
