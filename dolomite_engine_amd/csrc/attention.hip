@@ -201,7 +201,7 @@ __device__ __forceinline__ void xcd_remap_tile_bh(int& tile, int& b, int& h) {
 // ===========================================================================
 
 template <int DPAD>
-__global__ void __launch_bounds__(512) fa_fwd_kernel(
+__global__ void __launch_bounds__(512, 4) fa_fwd_kernel(
     const __bf16* __restrict__ q, const __bf16* __restrict__ k, const __bf16* __restrict__ v,
     __bf16* __restrict__ o, float* __restrict__ lse,
     const int32_t* __restrict__ cu, int H, int Hkv, int D, int G,
@@ -222,8 +222,11 @@ __global__ void __launch_bounds__(512) fa_fwd_kernel(
 
     int tile_id = blockIdx.x, b = blockIdx.y, h = blockIdx.z;
     xcd_remap_tile_bh(tile_id, b, h);
-    const int s0 = cu[b];
-    const int L = cu[b + 1] - s0;
+    // cu-derived values are wave-uniform: pin them to SGPRs so per-lane
+    // address arithmetic built on them does not hold VGPR pairs across the
+    // main loop (measured: the compiler otherwise spills pointer pairs)
+    const int s0 = __builtin_amdgcn_readfirstlane(cu[b]);
+    const int L = __builtin_amdgcn_readfirstlane(cu[b + 1]) - s0;
     // heaviest tiles (largest qs -> most k-tiles) dispatch FIRST: in-order
     // dispatch otherwise schedules the long-pole causal workgroups last
     // 8 waves x 16 rows = 128 q rows per workgroup: K/V staging and
@@ -238,10 +241,16 @@ __global__ void __launch_bounds__(512) fa_fwd_kernel(
     const int lr = lane & 15;   // fragment col / A-row index
     const int lg = lane >> 4;   // fragment k-group / C-row group
 
+    // NBUF=2 (DPAD<=96): double-buffered K/V images — ONE barrier per tile
+    // and the next tile's global loads issued before this tile's compute
+    // (T14 issue-early / write-late), still 2 workgroups/CU at ~76 KB.
+    // DPAD=128 double-buffered would exceed the 80 KB two-workgroup budget,
+    // so it keeps the single-buffer 2-barrier schedule.
+    constexpr int NBUF = (DPAD <= 96) ? 2 : 1;
     extern __shared__ char smem_raw[];
-    __bf16* Klds = (__bf16*)smem_raw;              // [64][SK]
-    __bf16* Vlds = Klds + 64 * SK;                 // [64 key][SK] (row-major, PI23 rows)
-    __bf16* Plds = Vlds + 64 * SK;                 // [8 waves][16][SV]
+    __bf16* Klds = (__bf16*)smem_raw;              // [NBUF][64][SK]
+    __bf16* Vlds = Klds + NBUF * 64 * SK;          // [NBUF][64 key][SK] (PI23 rows)
+    __bf16* Plds = Vlds + NBUF * 64 * SK;          // [8 waves][16][SV]
     __bf16* Pw = Plds + wave * 16 * SV;
     // V is stored row-major (coalesced staging) and consumed as the PV
     // B-fragment via ds_read_b64_tr_b16 — no transposed image, no scatter.
@@ -282,24 +291,23 @@ __global__ void __launch_bounds__(512) fa_fwd_kernel(
     // loses VALU arbitration; one setprio for it, no per-cluster flips.
     if (__builtin_amdgcn_readfirstlane(threadIdx.x) >= 256) __builtin_amdgcn_s_setprio(1);
 
-    for (int kt = 0; kt < ntiles; ++kt) {
-        const int ks = kt * 64;
-        // --- cooperative staging: K -> [key][d], V -> transposed [d][key] ---
-        {
-            const int pieces = 64 * DPAD / 8;  // 8-elem pieces
-            for (int pidx = threadIdx.x; pidx < pieces; pidx += 512) {
-                int key = pidx / (DPAD / 8);
-                int d0 = (pidx % (DPAD / 8)) * 8;
-                bool kv_valid = (ks + key) < kend;
-                const __bf16* kp = k + (int64_t)(s0 + (kv_valid ? ks + key : 0)) * k_ts + (int64_t)kvh * k_hs + d0;
-                bf16x8 kk = load_bf16x8_guard(kp, d0, D, kv_valid);
-                *(bf16x8*)&Klds[key * SK + d0] = kk;
-                const __bf16* vp = v + (int64_t)(s0 + (kv_valid ? ks + key : 0)) * v_ts + (int64_t)kvh * v_hs + d0;
-                *(bf16x8*)&Vlds[PI23(key) * SK + d0] = load_bf16x8_guard(vp, d0, D, kv_valid);
-            }
-        }
-        __syncthreads();
+    constexpr int PIECES = 64 * DPAD / 8;  // 8-elem staging pieces
+    constexpr int NP = (PIECES + 511) / 512;
 
+    // direct cooperative staging (prologue and the NBUF==1 path)
+    auto stage_direct = [&](int ks_, __bf16* Kw, __bf16* Vw) {
+        for (int pidx = threadIdx.x; pidx < PIECES; pidx += 512) {
+            int key = pidx / (DPAD / 8);
+            int d0 = (pidx % (DPAD / 8)) * 8;
+            bool kv_valid = (ks_ + key) < kend;
+            const __bf16* kp = k + (int64_t)(s0 + (kv_valid ? ks_ + key : 0)) * k_ts + (int64_t)kvh * k_hs + d0;
+            *(bf16x8*)&Kw[key * SK + d0] = load_bf16x8_guard(kp, d0, D, kv_valid);
+            const __bf16* vp = v + (int64_t)(s0 + (kv_valid ? ks_ + key : 0)) * v_ts + (int64_t)kvh * v_hs + d0;
+            *(bf16x8*)&Vw[PI23(key) * SK + d0] = load_bf16x8_guard(vp, d0, D, kv_valid);
+        }
+    };
+
+    auto compute_tile = [&](const __bf16* Kb, unsigned aV0c, unsigned aV1c, int ks) {
         // --- QK^T: 4 key-blocks of 16, accumulate over KCH chunks ---
         f32x4 sc[4];
 #pragma unroll
@@ -307,51 +315,45 @@ __global__ void __launch_bounds__(512) fa_fwd_kernel(
             sc[cb] = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
             for (int kc = 0; kc < KCH; ++kc) {
-                bf16x8 kf = *(const bf16x8*)&Klds[(cb * 16 + lr) * SK + kc * 32 + lg * 8];
+                bf16x8 kf = *(const bf16x8*)&Kb[(cb * 16 + lr) * SK + kc * 32 + lg * 8];
                 sc[cb] = MFMA16(qf[kc], kf, sc[cb]);
             }
         }
 
-        // --- mask + online softmax (fp32, per C-row) ---
-        float p_val[4][4];
-        float alpha[4];
+        // --- mask + online softmax (fp32), streamed per C-row: the row's
+        // 4 scores live in 4 registers and P goes straight to the LDS strip
+        // (keeps the per-tile live set small for the 128-VGPR cap) ---
         const float scale2 = scale * DOL_LOG2E;  // exp2-domain scores
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
             const int qpos = qs + wave * 16 + lg * 4 + r;
+            float pv[4];
             float rowmax = -INFINITY;
 #pragma unroll
             for (int cb = 0; cb < 4; ++cb) {
                 const int kpos = ks + cb * 16 + lr;
-                float s = sc[cb][r] * scale2;
-                if (kpos > qpos || kpos >= kend || qpos >= L) s = -INFINITY;
-                p_val[cb][r] = s;
-                rowmax = fmaxf(rowmax, s);
+                float sv = sc[cb][r] * scale2;
+                if (kpos > qpos || kpos >= kend || qpos >= L) sv = -INFINITY;
+                pv[cb] = sv;
+                rowmax = fmaxf(rowmax, sv);
             }
             rowmax = qwave_reduce_max(rowmax);
             float mnew = fmaxf(m_run[r], rowmax);
             if (mnew == -INFINITY) mnew = 0.f;  // fully-masked row guard
-            alpha[r] = (m_run[r] == -INFINITY) ? 0.f : exp2f(m_run[r] - mnew);
+            float alpha = (m_run[r] == -INFINITY) ? 0.f : exp2f(m_run[r] - mnew);
             float rsum = 0.f;
 #pragma unroll
             for (int cb = 0; cb < 4; ++cb) {
-                float e = (p_val[cb][r] == -INFINITY) ? 0.f : exp2f(p_val[cb][r] - mnew);
-                p_val[cb][r] = e;
+                float e = (pv[cb] == -INFINITY) ? 0.f : exp2f(pv[cb] - mnew);
+                Pw[(lg * 4 + r) * SV + cb * 16 + lr] = (__bf16)e;
                 rsum += e;
             }
             rsum = qwave_reduce_sum(rsum);
-            l_run[r] = l_run[r] * alpha[r] + rsum;
+            l_run[r] = l_run[r] * alpha + rsum;
             m_run[r] = mnew;
+#pragma unroll
+            for (int dc = 0; dc < DCH; ++dc) o_acc[dc][r] *= alpha;
         }
-        // rescale accumulator and stash P (bf16) in this wave's LDS tile
-#pragma unroll
-        for (int dc = 0; dc < DCH; ++dc)
-#pragma unroll
-            for (int r = 0; r < 4; ++r) o_acc[dc][r] *= alpha[r];
-#pragma unroll
-        for (int cb = 0; cb < 4; ++cb)
-#pragma unroll
-            for (int r = 0; r < 4; ++r) Pw[(lg * 4 + r) * SV + cb * 16 + lr] = (__bf16)p_val[cb][r];
 
         // --- PV: A = P (this wave's rows), B via the pipelined tr16 ladder
         // over the row-major PI23 V image (issue frag i+1, counted-wait
@@ -362,13 +364,13 @@ __global__ void __launch_bounds__(512) fa_fwd_kernel(
             lgkm_drain2x8(pf0, pf1);
             __builtin_amdgcn_sched_barrier(0);
             bf16x4 vlo[2], vhi[2];
-            tr16_issue<0>(aV0, aV1, vlo[0], vhi[0]);
+            tr16_issue<0>(aV0c, aV1c, vlo[0], vhi[0]);
 #define FWD_PV_STEP(i)                                                                                  \
     if constexpr ((i) < 2 * DCH) {                                                                      \
         constexpr int kc2_ = (i) / DCH, dc_ = (i) % DCH;                                                \
         if constexpr ((i) + 1 < 2 * DCH) {                                                              \
             constexpr int kn_ = ((i) + 1) / DCH, dn_ = ((i) + 1) % DCH;                                 \
-            tr16_issue<(kn_ * 32 * SK + dn_ * 16) * 2>(aV0, aV1, vlo[((i) + 1) & 1], vhi[((i) + 1) & 1]); \
+            tr16_issue<(kn_ * 32 * SK + dn_ * 16) * 2>(aV0c, aV1c, vlo[((i) + 1) & 1], vhi[((i) + 1) & 1]); \
             lgkm_wait2<2>(vlo[(i) & 1], vhi[(i) & 1]);                                                  \
         } else {                                                                                        \
             lgkm_wait2<0>(vlo[(i) & 1], vhi[(i) & 1]);                                                  \
@@ -382,7 +384,60 @@ __global__ void __launch_bounds__(512) fa_fwd_kernel(
 #undef FWD_PV_STEP
             __builtin_amdgcn_sched_barrier(0);
         }
-        __syncthreads();  // K/V LDS reused next tile
+    };
+
+    if constexpr (NBUF == 2) {
+        stage_direct(0, Klds, Vlds);
+        __syncthreads();
+        for (int kt = 0; kt < ntiles; ++kt) {
+            const int cur = kt & 1;
+            const int ks = kt * 64;
+            // T14 issue-early: next tile's K/V global loads into registers
+            bf16x8 stK[NP], stV[NP];
+            const bool have_next = kt + 1 < ntiles;
+            if (have_next) {
+                const int ksn = ks + 64;
+#pragma unroll
+                for (int i = 0; i < NP; ++i) {
+                    int pidx = (int)threadIdx.x + i * 512;
+                    bool act = pidx < PIECES;
+                    int key = act ? pidx / (DPAD / 8) : 0;
+                    int d0 = act ? (pidx % (DPAD / 8)) * 8 : 0;
+                    bool kv_valid = act && (ksn + key) < kend;
+                    const __bf16* kp = k + (int64_t)(s0 + (kv_valid ? ksn + key : 0)) * k_ts + (int64_t)kvh * k_hs + d0;
+                    stK[i] = load_bf16x8_guard(kp, d0, D, kv_valid);
+                    const __bf16* vp = v + (int64_t)(s0 + (kv_valid ? ksn + key : 0)) * v_ts + (int64_t)kvh * v_hs + d0;
+                    stV[i] = load_bf16x8_guard(vp, d0, D, kv_valid);
+                }
+            }
+            compute_tile(Klds + cur * 64 * SK,
+                         aV0 + (unsigned)(cur * 64 * SK * 2),
+                         aV1 + (unsigned)(cur * 64 * SK * 2), ks);
+            // write-late: staged registers into the other buffer; the one
+            // barrier also retires this tile's reads before the next write
+            if (have_next) {
+                __bf16* Kw = Klds + (cur ^ 1) * 64 * SK;
+                __bf16* Vw = Vlds + (cur ^ 1) * 64 * SK;
+#pragma unroll
+                for (int i = 0; i < NP; ++i) {
+                    int pidx = (int)threadIdx.x + i * 512;
+                    if (pidx < PIECES) {
+                        int key = pidx / (DPAD / 8);
+                        int d0 = (pidx % (DPAD / 8)) * 8;
+                        *(bf16x8*)&Kw[key * SK + d0] = stK[i];
+                        *(bf16x8*)&Vw[PI23(key) * SK + d0] = stV[i];
+                    }
+                }
+            }
+            __syncthreads();
+        }
+    } else {
+        for (int kt = 0; kt < ntiles; ++kt) {
+            stage_direct(kt * 64, Klds, Vlds);
+            __syncthreads();
+            compute_tile(Klds, aV0, aV1, kt * 64);
+            __syncthreads();  // K/V LDS reused next tile
+        }
     }
 
     // --- epilogue: normalize, store O and LSE ---
@@ -413,7 +468,7 @@ static int launch_fa_fwd(hipStream_t stream, const __bf16* q, const __bf16* k, c
                          int64_t q_ts, int64_t q_gs, int64_t k_ts, int64_t k_hs,
                          int64_t v_ts, int64_t v_hs, int max_tiles, float scale) {
     dim3 grid(max_tiles, batch, H), block(512);
-    size_t shmem = (64 * (DPAD + 16) * 2 + 8 * 16 * 72) * sizeof(__bf16);
+    size_t shmem = ((size_t)(DPAD <= 96 ? 2 : 1) * 64 * (DPAD + 16) * 2 + 8 * 16 * 72) * sizeof(__bf16);
     hipLaunchKernelGGL((fa_fwd_kernel<DPAD>), grid, block, shmem, stream,
                        q, k, v, o, lse, cu, H, Hkv, D, G,
                        q_ts, q_gs, k_ts, k_hs, v_ts, v_hs, (int64_t)H * D, T, scale);
@@ -524,8 +579,11 @@ __global__ void __launch_bounds__(512, DPAD <= 96 ? 4 : 2) fa_bwd_dkv_kernel(
     int tile_id = blockIdx.x, b = blockIdx.y, h = blockIdx.z;
     xcd_remap_tile_bh(tile_id, b, h);
     const int kvh = h / G;
-    const int s0 = cu[b];
-    const int L = cu[b + 1] - s0;
+    // cu-derived values are wave-uniform: pin them to SGPRs so per-lane
+    // address arithmetic built on them does not hold VGPR pairs across the
+    // main loop (measured: the compiler otherwise spills pointer pairs)
+    const int s0 = __builtin_amdgcn_readfirstlane(cu[b]);
+    const int L = __builtin_amdgcn_readfirstlane(cu[b + 1]) - s0;
     // 8 waves x 16 keys = 128-key strip per workgroup
     const int ks = tile_id * 128;
     if (ks >= L) return;
@@ -700,8 +758,11 @@ __global__ void __launch_bounds__(512, 4) fa_bwd_dq_kernel(
     int tile_id = blockIdx.x, b = blockIdx.y, h = blockIdx.z;
     xcd_remap_tile_bh(tile_id, b, h);
     const int kvh = h / G;
-    const int s0 = cu[b];
-    const int L = cu[b + 1] - s0;
+    // cu-derived values are wave-uniform: pin them to SGPRs so per-lane
+    // address arithmetic built on them does not hold VGPR pairs across the
+    // main loop (measured: the compiler otherwise spills pointer pairs)
+    const int s0 = __builtin_amdgcn_readfirstlane(cu[b]);
+    const int L = __builtin_amdgcn_readfirstlane(cu[b + 1]) - s0;
     // 8 waves x 16 rows = 128 q rows per workgroup; heaviest tiles first
     const int ntile_seq = (L + 127) / 128;
     if (tile_id >= ntile_seq) return;
