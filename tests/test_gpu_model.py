@@ -37,8 +37,8 @@ def test_padding_free_bf16_vs_reference_golden(golden_dir, case):
     out = model(input_ids=input_ids, position_ids=position_ids, cu_seqlens=cu, max_seqlen=S, labels=labels)
     B, SS = fx["input_ids"].shape
     ref_logits = fx["logits"].reshape(B * SS, -1)
-    torch.testing.assert_close(out.logits.float().cpu(), ref_logits, rtol=5e-2, atol=1e-1)
-    torch.testing.assert_close(out.loss.float().cpu(), fx["loss"], rtol=2e-2, atol=2e-2)
+    torch.testing.assert_close(out.logits.float().cpu(), ref_logits, rtol=1e-2, atol=2e-2)
+    torch.testing.assert_close(out.loss.float().cpu(), fx["loss"], rtol=1e-3, atol=1e-4)
 
 
 @pytest.mark.parametrize("case", ["mqa_rope_rmsnorm_gelu", "gqa_rope_rmsnorm_swiglu"])
@@ -129,7 +129,7 @@ def test_llama_shaped_finetune_step_8k(golden_dir):
     with torch.no_grad():
         _, oloss = omodel(ids.cpu(), pos.cpu(), cu.cpu(), S, labels=ids.cpu())
     rel = abs(float(out.loss) - float(oloss)) / abs(float(oloss))
-    assert rel < 2e-2, (float(out.loss), float(oloss))
+    assert rel < 1e-4, (float(out.loss), float(oloss))
 
 
 def test_moe_padding_free_bf16_vs_reference_golden(golden_dir):
@@ -143,8 +143,8 @@ def test_moe_padding_free_bf16_vs_reference_golden(golden_dir):
     input_ids, position_ids, cu, S, labels = _pack(fx)
     out = model(input_ids=input_ids, position_ids=position_ids, cu_seqlens=cu, max_seqlen=S, labels=labels)
     B, SS = fx["input_ids"].shape
-    torch.testing.assert_close(out.logits.float().cpu(), fx["logits"].reshape(B * SS, -1), rtol=5e-2, atol=1e-1)
-    torch.testing.assert_close(out.loss.float().cpu(), fx["loss"], rtol=2e-2, atol=2e-2)
+    torch.testing.assert_close(out.logits.float().cpu(), fx["logits"].reshape(B * SS, -1), rtol=1e-2, atol=2e-2)
+    torch.testing.assert_close(out.loss.float().cpu(), fx["loss"], rtol=1e-3, atol=1e-4)
     out.loss.backward()
     params = dict(model.named_parameters())
     # padding-free excludes the aux loss -> compare vs CLM-only golden grads
@@ -239,7 +239,7 @@ def test_gpu_training_trajectory_matches_oracle(golden_dir):
                 oracle.adamw_step_ref(p.data, p.grad * coef, m, v, step + 1, 1e-3, 0.9, 0.95, 1e-10, 0.1)
 
     for a, b_ in zip(gpu_losses, ref_losses):
-        assert abs(a - b_) / abs(b_) < 2e-2, (gpu_losses, ref_losses)
+        assert abs(a - b_) / abs(b_) < 5e-3, (gpu_losses, ref_losses)
 
 
 def test_gpu_pretrain_cli_end_to_end(tmp_path):
