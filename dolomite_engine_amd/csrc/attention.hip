@@ -241,12 +241,13 @@ __global__ void __launch_bounds__(512, 4) fa_fwd_kernel(
     const int lr = lane & 15;   // fragment col / A-row index
     const int lg = lane >> 4;   // fragment k-group / C-row group
 
-    // NBUF=2 (DPAD<=96): double-buffered K/V images — ONE barrier per tile
-    // and the next tile's global loads issued before this tile's compute
-    // (T14 issue-early / write-late), still 2 workgroups/CU at ~76 KB.
-    // DPAD=128 double-buffered would exceed the 80 KB two-workgroup budget,
-    // so it keeps the single-buffer 2-barrier schedule.
-    constexpr int NBUF = (DPAD <= 96) ? 2 : 1;
+    // Measured NEGATIVE and disabled: the NBUF=2 double-buffered K/V
+    // pipeline (issue-early loads, write-late before the single barrier)
+    // cost fa_fwd 5.44 -> 5.84 ms at B=16 — the write pass serializes
+    // behind the MFMAs exactly as guide T14's "write BEFORE the barrier"
+    // variant warns, and at 2 workgroups/CU the co-resident workgroup
+    // already hides the staging latency the pipeline was buying.
+    constexpr int NBUF = 1;
     extern __shared__ char smem_raw[];
     __bf16* Klds = (__bf16*)smem_raw;              // [NBUF][64][SK]
     __bf16* Vlds = Klds + NBUF * 64 * SK;          // [NBUF][64 key][SK] (PI23 rows)

