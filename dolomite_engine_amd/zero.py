@@ -70,6 +70,7 @@ class ZeRO2Engine:
         weight_decay: float = 0.1,
         bucket_mb: int = 128,
         overlap_comm: bool = True,
+        force_collectives: bool = False,
     ):
         self.model = model
         self.lr = lr
@@ -80,6 +81,12 @@ class ZeRO2Engine:
         self.world = get_world_size()
         self.rank = get_rank()
         self._sync = True
+        # force_collectives runs the RCCL reduce-scatter/all-gather +
+        # side-stream/event branch even at world size 1 (RS/AG are then
+        # identities): lets a single leased GPU execute the exact code the
+        # 8-GPU overlap path runs, instead of that branch being dead until
+        # a multi-GPU node exists. Covered by tests/test_gpu_model.py.
+        self.use_coll = self.world > 1 or (force_collectives and is_initialized())
 
         params = [p for p in model.parameters() if p.requires_grad]
         assert params
@@ -88,7 +95,7 @@ class ZeRO2Engine:
         self.device = device
         self._cast_scratch = None
         self.on_gpu = device.type == "cuda"
-        self.overlap = overlap_comm and self.on_gpu and self.world > 1
+        self.overlap = overlap_comm and self.on_gpu and self.use_coll
         self.comm_stream = torch.cuda.Stream() if self.overlap else None
 
         # pack in reverse registration order ≈ backward completion order
@@ -132,7 +139,7 @@ class ZeRO2Engine:
             if g.data_ptr() != view.data_ptr():
                 view.add_(g.reshape(-1))
                 p.grad = view.view(p.shape)
-        if self._sync and self.world > 1:
+        if self._sync and self.use_coll:
             b.pending -= 1
             if b.pending == 0 and self.overlap:
                 self._launch_reduce(b)
@@ -159,7 +166,7 @@ class ZeRO2Engine:
             b.comm_event.record()
 
     def _reduce_sync(self, b: _Bucket) -> None:
-        if self.world == 1:
+        if not self.use_coll:
             b.shard_grad.copy_(b.flat_grad[b.shard_slice])
             return
         backend = dist.get_backend()
@@ -219,7 +226,7 @@ class ZeRO2Engine:
         return grad_norm
 
     def _allgather_params(self, b: _Bucket) -> None:
-        if self.world == 1:
+        if not self.use_coll:
             return
         backend = dist.get_backend()
         if backend == "nccl" and self.overlap:

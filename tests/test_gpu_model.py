@@ -333,3 +333,70 @@ def test_gpu_finetune_cli_end_to_end(tmp_path):
     p.write_text(yaml.safe_dump(cfg))
     finetune.main(["--config", str(p)])
     assert (tmp_path / "ckpt" / "global_step4" / "model").exists()
+
+
+def test_zero2_rccl_collective_branch_on_one_gpu():
+    """Execute the RCCL reduce-scatter / all-gather side-stream overlap
+    branch on hardware (zero.py _launch_reduce/_allgather_params — dead
+    code until an 8-GPU node exists). A world-1 NCCL(=RCCL) process group
+    makes RS/AVG and AG identities, so a force_collectives engine must
+    reproduce the plain engine's training bit for bit while running the
+    real comm-stream launches, RCCL calls and event waits."""
+    import os
+
+    import torch.distributed as dist
+
+    from dolomite_engine_amd.hf_models import GPTDolomiteConfig, GPTDolomiteForCausalLM
+    from dolomite_engine_amd.ops import fused_cross_entropy
+    from dolomite_engine_amd.zero import ZeRO2Engine
+
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+    os.environ.setdefault("MASTER_PORT", "29571")
+    assert not dist.is_initialized()
+    dist.init_process_group("nccl", rank=0, world_size=1)
+    try:
+
+        def run(force):
+            kw = dict(
+                vocab_size=1024, n_positions=512, n_embd=256, n_layer=2, n_head=4,
+                attention_head_type="mqa", n_inner=512, activation_function="gelu_pytorch_tanh",
+                normalization_function="rmsnorm", position_embedding_type="rope",
+                resid_pdrop=0.0, embd_pdrop=0.0, attn_pdrop=0.0, tie_word_embeddings=False,
+            )
+            torch.manual_seed(33)
+            cfg = GPTDolomiteConfig(**kw)
+            cfg._attn_implementation = "flash_attention_2"
+            model = GPTDolomiteForCausalLM(cfg, use_padding_free_transformer=True).to(torch.bfloat16).cuda()
+            engine = ZeRO2Engine(
+                model, lr=1e-3, betas=(0.9, 0.95), eps=1e-10, weight_decay=0.1,
+                bucket_mb=1, overlap_comm=True, force_collectives=force,
+            )
+            if force:
+                assert engine.use_coll and engine.overlap and engine.comm_stream is not None
+            else:
+                assert not engine.use_coll
+            B, S = 2, 128
+            losses = []
+            for step in range(3):
+                engine.zero_grad()
+                engine.set_sync(True)
+                g = torch.Generator().manual_seed(900 + step)
+                tokens = torch.randint(0, 1024, (B, S + 1), generator=g).cuda()
+                ids, labels = tokens[:, :-1].reshape(-1), tokens[:, 1:].reshape(-1)
+                pos = torch.arange(S).repeat(B).cuda()
+                cu = torch.arange(0, B * S + 1, S, dtype=torch.int32).cuda()
+                out = model(input_ids=ids, position_ids=pos, cu_seqlens=cu, max_seqlen=S)
+                loss = fused_cross_entropy(out.logits, labels)
+                loss.backward()
+                engine.step(lr=1e-3, grad_clip=1.0)
+                losses.append(float(loss))
+            torch.cuda.synchronize()
+            return losses, {k: v.clone() for k, v in model.state_dict().items()}
+
+        losses_plain, state_plain = run(force=False)
+        losses_coll, state_coll = run(force=True)
+        assert losses_plain == losses_coll, (losses_plain, losses_coll)
+        for k in state_plain:
+            assert torch.equal(state_plain[k], state_coll[k]), f"{k} differs"
+    finally:
+        dist.destroy_process_group()
