@@ -389,6 +389,24 @@ def main():
                 "launches": p["count"],
             }
 
+    # secondary roofline: fa_fwd alongside the dominant kernel (the bwd
+    # region usually dominates; VERDICT r01 asked for both to be visible)
+    roofline_fwd = None
+    if profile and "fa_varlen_fwd" in profile and (roofline is None or roofline["kernel"] != "fa_varlen_fwd"):
+        p = profile["fa_varlen_fwd"]
+        achieved = fa_fwd_flops / (p["avg_ms"] / 1e3)
+        roofline_fwd = {
+            "kernel": "fa_varlen_fwd",
+            "bound": "mfma",
+            "achieved": achieved / 1e12,
+            "peak": 2500.0,
+            "unit": "TFLOP/s",
+            "frac": achieved / 1e12 / 2500.0,
+            "traffic": (traffic_cal.get("fa_varlen_fwd") or {}).get("hbm_read_bytes_per_launch"),
+            "avg_launch_ms": p["avg_ms"],
+            "launches": p["count"],
+        }
+
     cpu = None
     if not args.skip_cpu_baseline and world == 1 and args.model == "3b":
         cpu = cpu_baseline(args)
@@ -419,6 +437,7 @@ def main():
             "parallelism": f"dp{world}",
         },
         "roofline": roofline,
+        "roofline_fwd": roofline_fwd,
         "cpu_baseline": cpu,
         "kernel_profile": {k: {kk: round(vv, 3) if isinstance(vv, float) else vv for kk, vv in v.items()} for k, v in profile.items()},
     }

@@ -22,33 +22,33 @@ static int build_sample_idx_impl(const int32_t* sizes, const int32_t* doc_idx,
                                  int64_t num_samples) {
     if (seq_length <= 1 || num_epochs <= 0 || tokens_per_epoch <= 1) return 9101;
 
-    int64_t sample_index = 0;
-    int64_t doc_idx_index = 0;
-    int32_t doc_offset = 0;
-    sample_idx[2 * sample_index] = (IdxT)doc_idx_index;
-    sample_idx[2 * sample_index + 1] = (IdxT)doc_offset;
-    ++sample_index;
+    int64_t row = 0;
+    int64_t stream_pos = 0;
+    int32_t tok_off = 0;
+    sample_idx[2 * row] = (IdxT)stream_pos;
+    sample_idx[2 * row + 1] = (IdxT)tok_off;
+    ++row;
 
-    while (sample_index <= num_samples) {
-        int32_t remaining_seq_length = seq_length + 1;
-        while (remaining_seq_length != 0) {
-            int32_t doc_id = doc_idx[doc_idx_index];
-            int32_t doc_length = sizes[doc_id] - doc_offset;
-            remaining_seq_length -= doc_length;
-            if (remaining_seq_length <= 0) {
+    while (row <= num_samples) {
+        int32_t window_left = seq_length + 1;
+        while (window_left != 0) {
+            int32_t doc = doc_idx[stream_pos];
+            int32_t doc_left = sizes[doc] - tok_off;
+            window_left -= doc_left;
+            if (window_left <= 0) {
                 // the window ends inside this document; the next window
                 // starts at the last token consumed (windows overlap by one
                 // token, matching the seq_length+1 sampling)
-                doc_offset += (remaining_seq_length + doc_length - 1);
-                remaining_seq_length = 0;
+                tok_off += (window_left + doc_left - 1);
+                window_left = 0;
             } else {
-                ++doc_idx_index;
-                doc_offset = 0;
+                ++stream_pos;
+                tok_off = 0;
             }
         }
-        sample_idx[2 * sample_index] = (IdxT)doc_idx_index;
-        sample_idx[2 * sample_index + 1] = (IdxT)doc_offset;
-        ++sample_index;
+        sample_idx[2 * row] = (IdxT)stream_pos;
+        sample_idx[2 * row + 1] = (IdxT)tok_off;
+        ++row;
     }
     return 0;
 }
@@ -69,28 +69,28 @@ extern "C" int dolomite_build_sample_idx_i64(const int32_t* sizes, const int32_t
                                           sample_idx, num_samples);
 }
 
-extern "C" int dolomite_build_blending_indices(int16_t* dataset_index, int64_t* dataset_sample_index,
+extern "C" int dolomite_build_blending_indices(int16_t* dataset_index, int64_t* dataset_row,
                                                const double* weights, int32_t num_datasets,
                                                int64_t size) {
     if (num_datasets <= 0 || num_datasets > 32767) return 9102;
-    int64_t* current_samples = new int64_t[num_datasets];
-    for (int32_t i = 0; i < num_datasets; ++i) current_samples[i] = 0;
+    int64_t* drawn = new int64_t[num_datasets];
+    for (int32_t i = 0; i < num_datasets; ++i) drawn[i] = 0;
 
     for (int64_t sample_idx = 0; sample_idx < size; ++sample_idx) {
-        double sample_idx_double = sample_idx > 1 ? (double)sample_idx : 1.0;
-        int32_t max_error_index = 0;
-        double max_error = weights[0] * sample_idx_double - (double)current_samples[0];
+        double t = sample_idx > 1 ? (double)sample_idx : 1.0;
+        int32_t pick = 0;
+        double max_deficit = weights[0] * t - (double)drawn[0];
         for (int32_t d = 1; d < num_datasets; ++d) {
-            double error = weights[d] * sample_idx_double - (double)current_samples[d];
-            if (error > max_error) {
-                max_error = error;
-                max_error_index = d;
+            double deficit = weights[d] * t - (double)drawn[d];
+            if (deficit > max_deficit) {
+                max_deficit = deficit;
+                pick = d;
             }
         }
-        dataset_index[sample_idx] = (int16_t)max_error_index;
-        dataset_sample_index[sample_idx] = current_samples[max_error_index];
-        current_samples[max_error_index] += 1;
+        dataset_index[sample_idx] = (int16_t)pick;
+        dataset_row[sample_idx] = drawn[pick];
+        drawn[pick] += 1;
     }
-    delete[] current_samples;
+    delete[] drawn;
     return 0;
 }
