@@ -127,6 +127,13 @@ template <int N>
 __device__ __forceinline__ void lgkm_wait4(bf16x4& a, bf16x4& b, bf16x4& c, bf16x4& d) {
     asm volatile("s_waitcnt lgkmcnt(%c4)" : "+v"(a), "+v"(b), "+v"(c), "+v"(d) : "i"(N));
 }
+template <int N>
+__device__ __forceinline__ void lgkm_wait8(bf16x4& a, bf16x4& b, bf16x4& c, bf16x4& d,
+                                           bf16x4& e, bf16x4& f, bf16x4& g, bf16x4& h) {
+    asm volatile("s_waitcnt lgkmcnt(%c8)"
+                 : "+v"(a), "+v"(b), "+v"(c), "+v"(d), "+v"(e), "+v"(f), "+v"(g), "+v"(h)
+                 : "i"(N));
+}
 
 typedef __bf16 bf16x8v __attribute__((ext_vector_type(8)));
 __device__ __forceinline__ bf16x8 tr16_join8(bf16x4 lo, bf16x4 hi) {
@@ -218,7 +225,9 @@ __global__ void __launch_bounds__(512, 4) fa_fwd_kernel(
     //   P strips: stride 72 (b16 writes conflict-free; reads 2-way — the
     //     reverse trade costs more write cycles than it saves).
     constexpr int SK = DPAD + 16;    // K/V LDS row stride (elems)
-    constexpr int SV = 64 + 8;       // P strip row stride
+    // per-wave P strip, TRANSPOSED [64 key][16 q] PI23 rows (see dkv note):
+    // b64-packed writes, A-fragment read through the tr16 ladder
+    constexpr int STQ = 16 + 2;
 
     int tile_id = blockIdx.x, b = blockIdx.y, h = blockIdx.z;
     xcd_remap_tile_bh(tile_id, b, h);
@@ -251,8 +260,8 @@ __global__ void __launch_bounds__(512, 4) fa_fwd_kernel(
     extern __shared__ char smem_raw[];
     __bf16* Klds = (__bf16*)smem_raw;              // [NBUF][64][SK]
     __bf16* Vlds = Klds + NBUF * 64 * SK;          // [NBUF][64 key][SK] (PI23 rows)
-    __bf16* Plds = Vlds + NBUF * 64 * SK;          // [8 waves][16][SV]
-    __bf16* Pw = Plds + wave * 16 * SV;
+    __bf16* Plds = Vlds + NBUF * 64 * SK;          // [8 waves][64 key][STQ]
+    __bf16* Pw = Plds + wave * 64 * STQ;
     // V is stored row-major (coalesced staging) and consumed as the PV
     // B-fragment via ds_read_b64_tr_b16 — no transposed image, no scatter.
 
@@ -284,6 +293,8 @@ __global__ void __launch_bounds__(512, 4) fa_fwd_kernel(
     const int tr_cc = (lane & 3) * 4;
     const unsigned aV0 = (unsigned)(size_t)(Vlds + PI23(tr_g8 + tr_jj) * SK + tr_cc);
     const unsigned aV1 = (unsigned)(size_t)(Vlds + PI23(tr_g8 + 4 + tr_jj) * SK + tr_cc);
+    const unsigned aP0 = (unsigned)(size_t)(Pw + PI23(tr_g8 + tr_jj) * STQ + tr_cc);
+    const unsigned aP1 = (unsigned)(size_t)(Pw + PI23(tr_g8 + 4 + tr_jj) * STQ + tr_cc);
 
     const int kend = min(L, qs + 128);
     const int ntiles = (kend + 63) / 64;
@@ -325,6 +336,7 @@ __global__ void __launch_bounds__(512, 4) fa_fwd_kernel(
         // 4 scores live in 4 registers and P goes straight to the LDS strip
         // (keeps the per-tile live set small for the 128-VGPR cap) ---
         const float scale2 = scale * DOL_LOG2E;  // exp2-domain scores
+        bf16x4 pcb[4];
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
             const int qpos = qs + wave * 16 + lg * 4 + r;
@@ -346,7 +358,7 @@ __global__ void __launch_bounds__(512, 4) fa_fwd_kernel(
 #pragma unroll
             for (int cb = 0; cb < 4; ++cb) {
                 float e = (pv[cb] == -INFINITY) ? 0.f : exp2f(pv[cb] - mnew);
-                Pw[(lg * 4 + r) * SV + cb * 16 + lr] = (__bf16)e;
+                pcb[cb][r] = (__bf16)e;
                 rsum += e;
             }
             rsum = qwave_reduce_sum(rsum);
@@ -355,17 +367,24 @@ __global__ void __launch_bounds__(512, 4) fa_fwd_kernel(
 #pragma unroll
             for (int dc = 0; dc < DCH; ++dc) o_acc[dc][r] *= alpha;
         }
+#pragma unroll
+        for (int cb = 0; cb < 4; ++cb)
+            *(bf16x4*)&Pw[PI23(cb * 16 + lr) * STQ + lg * 4] = pcb[cb];
 
         // --- PV: A = P (this wave's rows), B via the pipelined tr16 ladder
         // over the row-major PI23 V image (issue frag i+1, counted-wait
         // frag i, MFMA — LDS latency hides under the matrix pipe) ---
         {
-            bf16x8 pf0 = *(const bf16x8*)&Pw[lr * SV + lg * 8];
-            bf16x8 pf1 = *(const bf16x8*)&Pw[lr * SV + 32 + lg * 8];
-            lgkm_drain2x8(pf0, pf1);
+            asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");  // exact counts from here
             __builtin_amdgcn_sched_barrier(0);
+            bf16x4 p_lo[2], p_hi[2];
+            tr16_issue<0>(aP0, aP1, p_lo[0], p_hi[0]);
+            tr16_issue<32 * STQ * 2>(aP0, aP1, p_lo[1], p_hi[1]);
             bf16x4 vlo[2], vhi[2];
             tr16_issue<0>(aV0c, aV1c, vlo[0], vhi[0]);
+            lgkm_wait4<2>(p_lo[0], p_hi[0], p_lo[1], p_hi[1]);
+            bf16x8 pf0 = tr16_join8(p_lo[0], p_hi[0]);
+            bf16x8 pf1 = tr16_join8(p_lo[1], p_hi[1]);
 #define FWD_PV_STEP(i)                                                                                  \
     if constexpr ((i) < 2 * DCH) {                                                                      \
         constexpr int kc2_ = (i) / DCH, dc_ = (i) % DCH;                                                \
@@ -469,7 +488,7 @@ static int launch_fa_fwd(hipStream_t stream, const __bf16* q, const __bf16* k, c
                          int64_t q_ts, int64_t q_gs, int64_t k_ts, int64_t k_hs,
                          int64_t v_ts, int64_t v_hs, int max_tiles, float scale) {
     dim3 grid(max_tiles, batch, H), block(512);
-    size_t shmem = ((size_t)(DPAD <= 96 ? 2 : 1) * 64 * (DPAD + 16) * 2 + 8 * 16 * 72) * sizeof(__bf16);
+    size_t shmem = ((size_t)(DPAD <= 96 ? 2 : 1) * 64 * (DPAD + 16) * 2 + 8 * 64 * (16 + 2)) * sizeof(__bf16);
     hipLaunchKernelGGL((fa_fwd_kernel<DPAD>), grid, block, shmem, stream,
                        q, k, v, o, lse, cu, H, Hkv, D, G,
                        q_ts, q_gs, k_ts, k_hs, v_ts, v_hs, (int64_t)H * D, T, scale);
@@ -595,11 +614,18 @@ __global__ void __launch_bounds__(512, DPAD <= 96 ? 4 : 2) fa_bwd_dkv_kernel(
     const int lg = lane >> 4;
 
     constexpr int SQ = DPAD + 16;              // PI23-rowed image stride
+    // Transposed [q][key] strips: each lane's 4 C-rows (keys) are contiguous
+    // along the key axis, so the softmax emits ONE ds_write_b64 per strip
+    // per q-block instead of 4 scalar b16 stores (32 -> 8 write
+    // instructions per wave per tile; the LDS store path was a measured
+    // co-bottleneck). Read back as MFMA A-fragments via the tr16 ladder.
+    // Stride 134: odd half-stride makes the b64 write groups conflict-free.
+    constexpr int STS = 128 + 6;
     extern __shared__ char smem_raw[];
     __bf16* Qlds = (__bf16*)smem_raw;          // [64 q][SQ] (PI23 rows)
     __bf16* dOl = Qlds + 64 * SQ;              // [64 q][SQ] (PI23 rows)
-    __bf16* dSTl = dOl + 64 * SQ;              // [128 key][ST] (dS^T, [key][q])
-    __bf16* PTl = dSTl + 128 * ST;             // [128 key][ST] (P^T, [key][q])
+    __bf16* dSTl = dOl + 64 * SQ;              // [64 q][STS] (dS^T, PI23 rows)
+    __bf16* PTl = dSTl + 64 * STS;             // [64 q][STS] (P^T, PI23 rows)
 
     const int kend = min(L, ks + 128);
 
@@ -630,6 +656,12 @@ __global__ void __launch_bounds__(512, DPAD <= 96 ? 4 : 2) fa_bwd_dkv_kernel(
     const unsigned aQ1 = (unsigned)(size_t)(Qlds + PI23(tr_g8 + 4 + tr_jj) * SQ + tr_cc);
     const unsigned aO0 = (unsigned)(size_t)(dOl + PI23(tr_g8 + tr_jj) * SQ + tr_cc);
     const unsigned aO1 = (unsigned)(size_t)(dOl + PI23(tr_g8 + 4 + tr_jj) * SQ + tr_cc);
+    // strip A-fragment bases: the wave's 16-key column block is runtime, so
+    // it lives in the base; the q-chunk (kc2*32 rows) is the immediate
+    const unsigned aPT0 = (unsigned)(size_t)(PTl + PI23(tr_g8 + tr_jj) * STS + wave * 16 + tr_cc);
+    const unsigned aPT1 = (unsigned)(size_t)(PTl + PI23(tr_g8 + 4 + tr_jj) * STS + wave * 16 + tr_cc);
+    const unsigned aDS0 = (unsigned)(size_t)(dSTl + PI23(tr_g8 + tr_jj) * STS + wave * 16 + tr_cc);
+    const unsigned aDS1 = (unsigned)(size_t)(dSTl + PI23(tr_g8 + 4 + tr_jj) * STS + wave * 16 + tr_cc);
 
     // T5 static priority (guide §5.5): the later-dispatched half of an
     // 8-wave workgroup loses VALU arbitration to the older half; one
@@ -681,29 +713,40 @@ __global__ void __launch_bounds__(512, DPAD <= 96 ? 4 : 2) fa_bwd_dkv_kernel(
             const bool qok = qpos < L;
             float lsev = qok ? lse[(int64_t)h * T_total + s0 + qpos] * DOL_LOG2E : 0.f;
             float delv = qok ? delta[(int64_t)h * T_total + s0 + qpos] : 0.f;
+            bf16x4 pvv, dsv;
 #pragma unroll
             for (int r = 0; r < 4; ++r) {
                 const int kpos = ks + wave * 16 + lg * 4 + r;
                 bool ok = full_tile || (qok && kpos < kend && kpos <= qpos);
                 float pv = ok ? exp2f(st[r] * (scale * DOL_LOG2E) - lsev) : 0.f;
                 float ds = ok ? pv * (dpt[r] - delv) * scale : 0.f;
-                PTl[(wave * 16 + lg * 4 + r) * ST + cb * 16 + lr] = (__bf16)pv;
-                dSTl[(wave * 16 + lg * 4 + r) * ST + cb * 16 + lr] = (__bf16)ds;
+                pvv[r] = (__bf16)pv;
+                dsv[r] = (__bf16)ds;
             }
+            const int srow = PI23(cb * 16 + lr) * STS + wave * 16 + lg * 4;
+            *(bf16x4*)&PTl[srow] = pvv;
+            *(bf16x4*)&dSTl[srow] = dsv;
         }
 
         // dV += P^T*dO ; dK += dS^T*Q (contraction over q) — pipelined tr16
-        // ladder: issue pair i+1 (4 reads), counted-wait pair i, 2 MFMAs.
+        // ladder. The strip A-fragments come from the transposed strips via
+        // tr16 too: issue all 8 strip reads + pair0, one counted wait.
         {
-            bf16x8 ptf0 = *(const bf16x8*)&PTl[(wave * 16 + lr) * ST + lg * 8];
-            bf16x8 dstf0 = *(const bf16x8*)&dSTl[(wave * 16 + lr) * ST + lg * 8];
-            bf16x8 ptf1 = *(const bf16x8*)&PTl[(wave * 16 + lr) * ST + 32 + lg * 8];
-            bf16x8 dstf1 = *(const bf16x8*)&dSTl[(wave * 16 + lr) * ST + 32 + lg * 8];
-            lgkm_drain4x8(ptf0, dstf0, ptf1, dstf1);
+            asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");  // exact counts from here
             __builtin_amdgcn_sched_barrier(0);
+            bf16x4 pt_lo[2], pt_hi[2], ds_lo[2], ds_hi[2];
+            tr16_issue<0>(aPT0, aPT1, pt_lo[0], pt_hi[0]);
+            tr16_issue<0>(aDS0, aDS1, ds_lo[0], ds_hi[0]);
+            tr16_issue<32 * STS * 2>(aPT0, aPT1, pt_lo[1], pt_hi[1]);
+            tr16_issue<32 * STS * 2>(aDS0, aDS1, ds_lo[1], ds_hi[1]);
             bf16x4 olo[2], ohi[2], qlo[2], qhi[2];
             tr16_issue<0>(aO0, aO1, olo[0], ohi[0]);
             tr16_issue<0>(aQ0, aQ1, qlo[0], qhi[0]);
+            lgkm_wait8<4>(pt_lo[0], pt_hi[0], ds_lo[0], ds_hi[0], pt_lo[1], pt_hi[1], ds_lo[1], ds_hi[1]);
+            bf16x8 ptf0 = tr16_join8(pt_lo[0], pt_hi[0]);
+            bf16x8 dstf0 = tr16_join8(ds_lo[0], ds_hi[0]);
+            bf16x8 ptf1 = tr16_join8(pt_lo[1], pt_hi[1]);
+            bf16x8 dstf1 = tr16_join8(ds_lo[1], ds_hi[1]);
 #define DKV_STEP(i)                                                                                     \
     if constexpr ((i) < 2 * DCH) {                                                                      \
         constexpr int kc2_ = (i) / DCH, dc_ = (i) % DCH;                                                \
@@ -778,11 +821,15 @@ __global__ void __launch_bounds__(512, 4) fa_bwd_dq_kernel(
     // come from the PI23-rowed row-major K image via the tr16 ladder, like
     // dkv. Saves DPAD*96 LDS elements and the per-piece scatter writes.
     constexpr int SQ = DPAD + 16;              // image stride
+    // per-wave dS strip, TRANSPOSED [64 key][16 q] with PI23 rows: the
+    // C-fragment's 4 q-rows pack into one ds_write_b64 per key block and
+    // the dQ A-fragment reads back through the tr16 ladder (see dkv note)
+    constexpr int STQ = 16 + 2;                // odd half-stride: b64 writes conflict-free
     extern __shared__ char smem_raw[];
     __bf16* Klds = (__bf16*)smem_raw;          // [64 key][SQ] (PI23 rows)
     __bf16* Vlds = Klds + 64 * SQ;             // [64 key][SQ] (natural rows)
-    __bf16* dSl = Vlds + 64 * SQ;              // [8 waves][16 q][ST] (dS strips)
-    __bf16* dSw = dSl + wave * 16 * ST;
+    __bf16* dSl = Vlds + 64 * SQ;              // [8 waves][64 key][STQ]
+    __bf16* dSw = dSl + wave * 64 * STQ;
 
     const int64_t q_hoff = (int64_t)(h / G) * q_gs + (int64_t)(h % G) * D;
     const int64_t do_hoff = (int64_t)h * D;
@@ -793,6 +840,8 @@ __global__ void __launch_bounds__(512, 4) fa_bwd_dq_kernel(
     const int tr_cc = (lane & 3) * 4;
     const unsigned aK0 = (unsigned)(size_t)(Klds + PI23(tr_g8 + tr_jj) * SQ + tr_cc);
     const unsigned aK1 = (unsigned)(size_t)(Klds + PI23(tr_g8 + 4 + tr_jj) * SQ + tr_cc);
+    const unsigned aS0 = (unsigned)(size_t)(dSw + PI23(tr_g8 + tr_jj) * STQ + tr_cc);
+    const unsigned aS1 = (unsigned)(size_t)(dSw + PI23(tr_g8 + 4 + tr_jj) * STQ + tr_cc);
 
     // this wave's Q and dO fragments (A-layout: i = lr -> q row)
     const int qrow = qs + wave * 16 + lr;
@@ -859,6 +908,7 @@ __global__ void __launch_bounds__(512, 4) fa_bwd_dq_kernel(
                 bf16x8 vb = *(const bf16x8*)&Vlds[(cb * 16 + lr) * SQ + d0];
                 dp = MFMA16(dfr[kc], vb, dp);
             }
+            bf16x4 dsv;
 #pragma unroll
             for (int r = 0; r < 4; ++r) {
                 const int qpos = qs + wave * 16 + lg * 4 + r;
@@ -866,19 +916,25 @@ __global__ void __launch_bounds__(512, 4) fa_bwd_dq_kernel(
                 bool ok = full_tile || (qpos < L && kpos < kend_total && kpos <= qpos);
                 float pv = ok ? exp2f(sc[r] * (scale * DOL_LOG2E) - lsev[r]) : 0.f;
                 float ds = ok ? pv * (dp[r] - delv[r]) * scale : 0.f;
-                dSw[(lg * 4 + r) * ST + cb * 16 + lr] = (__bf16)ds;
+                dsv[r] = (__bf16)ds;
             }
+            *(bf16x4*)&dSw[PI23(cb * 16 + lr) * STQ + lg * 4] = dsv;
         }
 
         // dQ += dS*K (contraction over this tile's keys) — pipelined tr16
-        // ladder over the PI23 K image
+        // ladder over the PI23 K image; dS A-fragments via tr16 from the
+        // transposed strip
         {
-            bf16x8 dsf0 = *(const bf16x8*)&dSw[lr * ST + lg * 8];
-            bf16x8 dsf1 = *(const bf16x8*)&dSw[lr * ST + 32 + lg * 8];
-            lgkm_drain2x8(dsf0, dsf1);
+            asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");  // exact counts from here
             __builtin_amdgcn_sched_barrier(0);
+            bf16x4 s_lo[2], s_hi[2];
+            tr16_issue<0>(aS0, aS1, s_lo[0], s_hi[0]);
+            tr16_issue<32 * STQ * 2>(aS0, aS1, s_lo[1], s_hi[1]);
             bf16x4 klo[2], khi[2];
             tr16_issue<0>(aK0, aK1, klo[0], khi[0]);
+            lgkm_wait4<2>(s_lo[0], s_hi[0], s_lo[1], s_hi[1]);
+            bf16x8 dsf0 = tr16_join8(s_lo[0], s_hi[0]);
+            bf16x8 dsf1 = tr16_join8(s_lo[1], s_hi[1]);
 #define DQ_STEP(i)                                                                                      \
     if constexpr ((i) < 2 * DCH) {                                                                      \
         constexpr int kc2_ = (i) / DCH, dc_ = (i) % DCH;                                                \
@@ -925,13 +981,13 @@ static int launch_fa_bwd(hipStream_t stream, const __bf16* q, const __bf16* k, c
     dim3 block(512);
     dim3 grid(max_tiles, batch, H);
     constexpr int SQ = DPAD + 16;
-    size_t shmem_dkv = (size_t)(128 * ST * 2 + 64 * SQ * 2) * sizeof(__bf16);
+    size_t shmem_dkv = (size_t)(64 * (128 + 6) * 2 + 64 * SQ * 2) * sizeof(__bf16);
     hipLaunchKernelGGL((fa_bwd_dkv_kernel<DPAD>), grid, block, shmem_dkv, stream,
                        q, k, v, dout, lse, delta, dk_acc, dv_acc, cu, H, Hkv, D, G,
                        q_ts, q_gs, k_ts, k_hs, v_ts, v_hs, do_ts, T, scale);
     int err = dol_last_error();
     if (err) return err;
-    size_t shmem_dq = (size_t)(64 * SQ * 2 + 8 * 16 * ST) * sizeof(__bf16);
+    size_t shmem_dq = (size_t)(64 * SQ * 2 + 8 * 64 * (16 + 2)) * sizeof(__bf16);
     hipLaunchKernelGGL((fa_bwd_dq_kernel<DPAD>), grid, block, shmem_dq, stream,
                        q, k, v, dout, lse, delta, dqkv_q, cu, H, Hkv, D, G,
                        q_ts, q_gs, k_ts, k_hs, v_ts, v_hs, do_ts, T, scale);

@@ -722,8 +722,12 @@ extern "C" int dolomite_ce_fwd(dolomite_stream_t stream,
 template <typename T, int V>
 __global__ void __launch_bounds__(256) ce_bwd_kernel(
     const T* __restrict__ logits, const int64_t* __restrict__ labels,
-    const float* __restrict__ lse, T* __restrict__ dlogits, float gs,
+    const float* __restrict__ lse, T* __restrict__ dlogits,
+    const float* __restrict__ gs_dev,
     int64_t V_dim, int64_t row_stride, int ignore_index) {
+    // device-side grad scale: removes the round-1 host sync that read
+    // gout/n_valid back to the CPU before every CE backward launch
+    const float gs = *gs_dev;
     int64_t t = blockIdx.x;
     const T* xr = logits + t * row_stride;
     T* dr = dlogits + t * row_stride;
@@ -765,24 +769,24 @@ __global__ void __launch_bounds__(256) ce_bwd_kernel(
 
 extern "C" int dolomite_ce_bwd(dolomite_stream_t stream,
                                const void* logits, const int64_t* labels, const float* lse,
-                               void* dlogits, float grad_scale,
+                               void* dlogits, const float* grad_scale_dev,
                                int64_t T_rows, int64_t V_dim, int64_t row_stride,
                                int ignore_index, int dtype) {
     dim3 grid((uint32_t)T_rows), block(256);
     if (dtype == DOLOMITE_BF16) {
         if (V_dim % 8 == 0)
             hipLaunchKernelGGL((ce_bwd_kernel<uint16_t, 8>), grid, block, 0, (hipStream_t)stream,
-                               (const uint16_t*)logits, labels, lse, (uint16_t*)dlogits, grad_scale, V_dim, row_stride, ignore_index);
+                               (const uint16_t*)logits, labels, lse, (uint16_t*)dlogits, grad_scale_dev, V_dim, row_stride, ignore_index);
         else
             hipLaunchKernelGGL((ce_bwd_kernel<uint16_t, 1>), grid, block, 0, (hipStream_t)stream,
-                               (const uint16_t*)logits, labels, lse, (uint16_t*)dlogits, grad_scale, V_dim, row_stride, ignore_index);
+                               (const uint16_t*)logits, labels, lse, (uint16_t*)dlogits, grad_scale_dev, V_dim, row_stride, ignore_index);
     } else {
         if (V_dim % 4 == 0)
             hipLaunchKernelGGL((ce_bwd_kernel<float, 4>), grid, block, 0, (hipStream_t)stream,
-                               (const float*)logits, labels, lse, (float*)dlogits, grad_scale, V_dim, row_stride, ignore_index);
+                               (const float*)logits, labels, lse, (float*)dlogits, grad_scale_dev, V_dim, row_stride, ignore_index);
         else
             hipLaunchKernelGGL((ce_bwd_kernel<float, 1>), grid, block, 0, (hipStream_t)stream,
-                               (const float*)logits, labels, lse, (float*)dlogits, grad_scale, V_dim, row_stride, ignore_index);
+                               (const float*)logits, labels, lse, (float*)dlogits, grad_scale_dev, V_dim, row_stride, ignore_index);
     }
     return dol_last_error();
 }

@@ -527,14 +527,15 @@ class FusedCrossEntropy(torch.autograd.Function):
     def backward(ctx, gout):
         logits, labels, lse, n_valid = ctx.saved_tensors
         T, V = logits.shape
-        gs = float(gout) / float(n_valid)
         if logits.is_cuda:
+            # device-side scale: no host readback of gout/n_valid
+            gs_dev = (gout.float() / n_valid.float()).reshape(1).contiguous()
             dlogits = torch.empty_like(logits)
             with hip.prof("ce_bwd"):
              hip.check(
                 hip.lib().dolomite_ce_bwd(
                     hip.stream(), hip.ptr(logits), hip.ptr(labels), hip.ptr(lse),
-                    hip.ptr(dlogits), gs, T, V, logits.stride(0), ctx.ignore_index, hip.dt(logits),
+                    hip.ptr(gs_dev), T, V, logits.stride(0), ctx.ignore_index, hip.dt(logits),
                 ),
                 "ce_bwd",
             )

@@ -177,9 +177,11 @@ int dolomite_ce_fwd(dolomite_stream_t stream,
                     float* row_loss, float* lse,
                     int64_t T, int64_t V, int64_t row_stride,
                     int ignore_index, int dtype);
+/* grad_scale_dev: DEVICE pointer to one fp32 (upstream grad / n_valid),
+ * so the backward launch needs no host readback of device scalars. */
 int dolomite_ce_bwd(dolomite_stream_t stream,
                     const void* logits, const int64_t* labels, const float* lse,
-                    void* dlogits, float grad_scale,
+                    void* dlogits, const float* grad_scale_dev,
                     int64_t T, int64_t V, int64_t row_stride,
                     int ignore_index, int dtype);
 
