@@ -535,11 +535,13 @@ class FusedCrossEntropy(torch.autograd.Function):
              hip.check(
                 hip.lib().dolomite_ce_bwd(
                     hip.stream(), hip.ptr(logits), hip.ptr(labels), hip.ptr(lse),
-                    hip.ptr(gs_dev), T, V, logits.stride(0), ctx.ignore_index, hip.dt(logits),
+                    hip.ptr(dlogits), hip.ptr(gs_dev), T, V, logits.stride(0),
+                    ctx.ignore_index, hip.dt(logits),
                 ),
                 "ce_bwd",
             )
         else:
+            gs = float(gout) / float(n_valid)
             p = torch.softmax(logits.float(), dim=-1)
             onehot = torch.zeros_like(p)
             onehot.scatter_(-1, labels.clamp_min(0).unsqueeze(-1), 1.0)
