@@ -226,8 +226,9 @@ __global__ void __launch_bounds__(512, 4) fa_fwd_kernel(
     //     reverse trade costs more write cycles than it saves).
     constexpr int SK = DPAD + 16;    // K/V LDS row stride (elems)
     // per-wave P strip, TRANSPOSED [64 key][16 q] PI23 rows (see dkv note):
-    // b64-packed writes, A-fragment read through the tr16 ladder
-    constexpr int STQ = 16 + 2;
+    // b64-packed writes, A-fragment read through the tr16 ladder;
+    // stride 20 keeps rows 8B-aligned
+    constexpr int STQ = 16 + 4;
 
     int tile_id = blockIdx.x, b = blockIdx.y, h = blockIdx.z;
     xcd_remap_tile_bh(tile_id, b, h);
@@ -488,7 +489,7 @@ static int launch_fa_fwd(hipStream_t stream, const __bf16* q, const __bf16* k, c
                          int64_t q_ts, int64_t q_gs, int64_t k_ts, int64_t k_hs,
                          int64_t v_ts, int64_t v_hs, int max_tiles, float scale) {
     dim3 grid(max_tiles, batch, H), block(512);
-    size_t shmem = ((size_t)(DPAD <= 96 ? 2 : 1) * 64 * (DPAD + 16) * 2 + 8 * 64 * (16 + 2)) * sizeof(__bf16);
+    size_t shmem = ((size_t)(DPAD <= 96 ? 2 : 1) * 64 * (DPAD + 16) * 2 + 8 * 64 * (16 + 4)) * sizeof(__bf16);
     hipLaunchKernelGGL((fa_fwd_kernel<DPAD>), grid, block, shmem, stream,
                        q, k, v, o, lse, cu, H, Hkv, D, G,
                        q_ts, q_gs, k_ts, k_hs, v_ts, v_hs, (int64_t)H * D, T, scale);
@@ -619,8 +620,12 @@ __global__ void __launch_bounds__(512, DPAD <= 96 ? 4 : 2) fa_bwd_dkv_kernel(
     // per q-block instead of 4 scalar b16 stores (32 -> 8 write
     // instructions per wave per tile; the LDS store path was a measured
     // co-bottleneck). Read back as MFMA A-fragments via the tr16 ladder.
-    // Stride 134: odd half-stride makes the b64 write groups conflict-free.
-    constexpr int STS = 128 + 6;
+    // Stride 132: rows stay 8-byte aligned (ds_write_b64 / tr16 reads need
+    // natural alignment — stride*2 must be a multiple of 8; the odd
+    // half-stride that would fully de-conflict the write groups is
+    // misaligned). 132/2=66 leaves a 2-way write-group conflict, which on
+    // ds_write_b64 costs ~2 LDS-array cycles per instruction.
+    constexpr int STS = 128 + 4;
     extern __shared__ char smem_raw[];
     __bf16* Qlds = (__bf16*)smem_raw;          // [64 q][SQ] (PI23 rows)
     __bf16* dOl = Qlds + 64 * SQ;              // [64 q][SQ] (PI23 rows)
@@ -824,7 +829,7 @@ __global__ void __launch_bounds__(512, 4) fa_bwd_dq_kernel(
     // per-wave dS strip, TRANSPOSED [64 key][16 q] with PI23 rows: the
     // C-fragment's 4 q-rows pack into one ds_write_b64 per key block and
     // the dQ A-fragment reads back through the tr16 ladder (see dkv note)
-    constexpr int STQ = 16 + 2;                // odd half-stride: b64 writes conflict-free
+    constexpr int STQ = 16 + 4;                // 8B-aligned rows (see dkv stride note)
     extern __shared__ char smem_raw[];
     __bf16* Klds = (__bf16*)smem_raw;          // [64 key][SQ] (PI23 rows)
     __bf16* Vlds = Klds + 64 * SQ;             // [64 key][SQ] (natural rows)
@@ -981,13 +986,13 @@ static int launch_fa_bwd(hipStream_t stream, const __bf16* q, const __bf16* k, c
     dim3 block(512);
     dim3 grid(max_tiles, batch, H);
     constexpr int SQ = DPAD + 16;
-    size_t shmem_dkv = (size_t)(64 * (128 + 6) * 2 + 64 * SQ * 2) * sizeof(__bf16);
+    size_t shmem_dkv = (size_t)(64 * (128 + 4) * 2 + 64 * SQ * 2) * sizeof(__bf16);
     hipLaunchKernelGGL((fa_bwd_dkv_kernel<DPAD>), grid, block, shmem_dkv, stream,
                        q, k, v, dout, lse, delta, dk_acc, dv_acc, cu, H, Hkv, D, G,
                        q_ts, q_gs, k_ts, k_hs, v_ts, v_hs, do_ts, T, scale);
     int err = dol_last_error();
     if (err) return err;
-    size_t shmem_dq = (size_t)(64 * SQ * 2 + 8 * 64 * (16 + 2)) * sizeof(__bf16);
+    size_t shmem_dq = (size_t)(64 * SQ * 2 + 8 * 64 * (16 + 4)) * sizeof(__bf16);
     hipLaunchKernelGGL((fa_bwd_dq_kernel<DPAD>), grid, block, shmem_dq, stream,
                        q, k, v, dout, lse, delta, dqkv_q, cu, H, Hkv, D, G,
                        q_ts, q_gs, k_ts, k_hs, v_ts, v_hs, do_ts, T, scale);
