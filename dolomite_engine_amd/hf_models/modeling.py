@@ -42,6 +42,17 @@ class ParameterizedLinear(nn.Linear):
         self.std = std
         super().__init__(in_features, out_features, bias, device, dtype)
 
+    def forward(self, input):
+        # Training on GPU with a live ZeRO engine: route through the
+        # deferred-wgrad linear so dW/db run on the engine's side stream,
+        # off the backward critical path (zero.py DeferredWgradLinear).
+        if self.training and input.is_cuda and torch.is_grad_enabled():
+            from ..zero import DeferredWgradLinear, _WgradSink
+
+            if _WgradSink.current is not None:
+                return DeferredWgradLinear.apply(input, self.weight, self.bias)
+        return super().forward(input)
+
     @torch.no_grad()
     def reset_parameters(self) -> None:
         if self.std is None:

@@ -71,6 +71,7 @@ class ZeRO2Engine:
         bucket_mb: int = 128,
         overlap_comm: bool = True,
         force_collectives: bool = False,
+        defer_wgrad: bool = True,
     ):
         self.model = model
         self.lr = lr
@@ -97,6 +98,11 @@ class ZeRO2Engine:
         self.on_gpu = device.type == "cuda"
         self.overlap = overlap_comm and self.on_gpu and self.use_coll
         self.comm_stream = torch.cuda.Stream() if self.overlap else None
+        # wgrad deferral (GPU only): dW GEMMs run on this stream, off the
+        # backward critical path; drained before reduce/clip/step
+        self.defer_wgrad = defer_wgrad and self.on_gpu
+        self.wgrad_stream = torch.cuda.Stream() if self.defer_wgrad else None
+        self._wgrad_events = []
 
         # pack in reverse registration order ≈ backward completion order
         bucket_elems = bucket_mb * 1024 * 1024 // max(dtype.itemsize, 2)
@@ -116,6 +122,15 @@ class ZeRO2Engine:
             for p, off in b.params:
                 self._param_bucket[p] = (b, off)
         self._hooks = [p.register_post_accumulate_grad_hook(self._grad_hook) for p in params]
+
+        # weight -> (bucket, offset) of its Linear's bias, for wgrad deferral
+        self._linear_bias = {}
+        for m in model.modules():
+            if isinstance(m, torch.nn.Linear) and m.bias is not None:
+                if m.weight in self._param_bucket and m.bias in self._param_bucket:
+                    self._linear_bias[m.weight] = self._param_bucket[m.bias]
+        if self.defer_wgrad:
+            _WgradSink.current = self
 
     # ---- backward-side ----------------------------------------------------
 
@@ -157,8 +172,10 @@ class ZeRO2Engine:
         out.copy_(b.flat_grad)
         return out
 
-    def _launch_reduce(self, b: _Bucket) -> None:
+    def _launch_reduce(self, b: _Bucket, extra_event=None) -> None:
         self.comm_stream.wait_stream(torch.cuda.current_stream())
+        if extra_event is not None:
+            self.comm_stream.wait_event(extra_event)
         with torch.cuda.stream(self.comm_stream):
             # fp32 on the wire (reference communication_dtype=fp32)
             dist.reduce_scatter_tensor(b.shard_grad, self._comm_cast(b), op=dist.ReduceOp.AVG)
@@ -180,6 +197,7 @@ class ZeRO2Engine:
     # ---- step-side --------------------------------------------------------
 
     def _finish_reduces(self) -> None:
+        self._drain_wgrad()
         for b in self.buckets:
             if self.overlap and b.comm_event is not None:
                 torch.cuda.current_stream().wait_event(b.comm_event)
@@ -244,6 +262,7 @@ class ZeRO2Engine:
                 b.flat_param.view(self.world, b.shard_size)[i].copy_(s)
 
     def zero_grad(self) -> None:
+        self._drain_wgrad()
         for b in self.buckets:
             b.flat_grad.zero_()
             b.shard_grad.zero_()
@@ -304,3 +323,88 @@ class ZeRO2Engine:
                 if b.comm_event is not None:
                     torch.cuda.current_stream().wait_event(b.comm_event)
                     b.comm_event = None
+
+
+# ---------------------------------------------------------------------------
+# Deferred weight gradients: the wgrad GEMMs (dW = dY^T X) are independent of
+# the backward critical path (only dX feeds the next layer's backward), so
+# on GPU they run on a dedicated side stream and accumulate STRAIGHT into the
+# flat-bucket grad views, overlapped with the rest of backward + attention
+# backward kernels (which leave the MFMA pipe ~70% idle). The engine drains
+# the stream before any reduce/clip/step touches the grads.
+# ---------------------------------------------------------------------------
+
+
+class _WgradSink:
+    current = None  # the live ZeRO2Engine with defer enabled, or None
+
+
+class DeferredWgradLinear(torch.autograd.Function):
+    """F.linear whose backward computes dX inline but hands dW (and db) to
+    the engine's wgrad stream. Falls back to inline grads when no engine
+    sink is active (CPU, eval, engine-less tests)."""
+
+    @staticmethod
+    def forward(ctx, x, weight, bias):
+        ctx.save_for_backward(x, weight)
+        ctx.has_bias = bias is not None
+        return torch.nn.functional.linear(x, weight, bias)
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, weight = ctx.saved_tensors
+        dy2 = dy.reshape(-1, dy.shape[-1])
+        x2 = x.reshape(-1, x.shape[-1])
+        dx = (dy2 @ weight).view_as(x)
+        sink = _WgradSink.current
+        if sink is not None and x.is_cuda and sink._defer_wgrad(weight, dy2, x2, ctx.has_bias):
+            return dx, None, None
+        dw = dy2.t() @ x2
+        db = dy2.sum(0) if ctx.has_bias else None
+        return dx, dw, db
+
+
+def _engine_defer_wgrad(self, weight, dy2, x2, has_bias) -> bool:
+    if not self.on_gpu or not self.defer_wgrad:
+        return False
+    ent = self._param_bucket.get(weight)
+    if ent is None:
+        return False
+    bias_ent = self._linear_bias.get(weight) if has_bias else None
+    if has_bias and bias_ent is None:
+        return False  # unknown bias pairing: let autograd do it inline
+    b, off = ent
+    s = self.wgrad_stream
+    s.wait_stream(torch.cuda.current_stream())
+    with torch.cuda.stream(s):
+        n_out, n_in = dy2.shape[1], x2.shape[1]
+        wview = b.flat_grad[off : off + n_out * n_in].view(n_out, n_in)
+        # beta=1 accumulate: C += dY^T @ X with fp32 internal accumulation
+        wview.addmm_(dy2.t(), x2)
+        if bias_ent is not None:
+            bb, boff = bias_ent
+            bb.flat_grad[boff : boff + n_out].add_(dy2.sum(0))
+        ev = torch.cuda.Event()
+        ev.record()
+    self._wgrad_events.append(ev)
+    # the allocator must not recycle these while the side stream reads them
+    dy2.record_stream(s)
+    x2.record_stream(s)
+    if self._sync and self.use_coll:
+        for bx in ({b, bias_ent[0]} if bias_ent is not None else {b}):
+            bx.pending -= 1
+            if bx.pending == 0 and self.overlap:
+                self._launch_reduce(bx, extra_event=ev)
+    return True
+
+
+def _engine_drain_wgrad(self) -> None:
+    if self._wgrad_events:
+        cur = torch.cuda.current_stream()
+        for ev in self._wgrad_events:
+            cur.wait_event(ev)
+        self._wgrad_events.clear()
+
+
+ZeRO2Engine._defer_wgrad = _engine_defer_wgrad
+ZeRO2Engine._drain_wgrad = _engine_drain_wgrad
