@@ -92,6 +92,10 @@ MODEL_MOE = dict(
     add_bias=False,
     num_experts=32,
     num_experts_per_tok=2,
+    # train the reference's actual MoE pretraining objective: CLM + the
+    # mixtral load-balancing aux loss on the packed router logits
+    output_router_logits=True,
+    router_aux_loss_coef=0.001,
     bos_token_id=0,
     eos_token_id=0,
     pad_token_id=0,

@@ -776,7 +776,12 @@ __global__ void __launch_bounds__(512, DPAD <= 96 ? 4 : 2) fa_bwd_dkv_kernel(
         }
     }
 
-    // join dK/dV (G q-head contributors per kv strip)
+    // Deterministic join: each (kv-tile, q-head) workgroup owns its keys'
+    // (t, h) slot of the per-head partial buffers exclusively, so dK/dV
+    // are plain stores (half the traffic of atomic RMW); fa_grad_finalize
+    // reduces the G q-head contributions in a FIXED order. The round-1
+    // fp32 atomicAdd join was arrival-order nondeterministic, which made
+    // bit-exact checkpoint resume a coin flip.
 #pragma unroll
     for (int dc = 0; dc < DCH; ++dc) {
 #pragma unroll
@@ -784,9 +789,9 @@ __global__ void __launch_bounds__(512, DPAD <= 96 ? 4 : 2) fa_bwd_dkv_kernel(
             const int kpos = ks + wave * 16 + lg * 4 + r;
             const int d = dc * 16 + lr;
             if (kpos < kend && d < D) {
-                int64_t idx = ((int64_t)(s0 + kpos) * Hkv + kvh) * D + d;
-                atomicAdd(&dk_acc[idx], dkr[dc][r]);
-                atomicAdd(&dv_acc[idx], dvr[dc][r]);
+                int64_t idx = ((int64_t)(s0 + kpos) * H + h) * D + d;
+                dk_acc[idx] = dkr[dc][r];
+                dv_acc[idx] = dvr[dc][r];
             }
         }
     }
@@ -1030,11 +1035,13 @@ extern "C" int dolomite_fa_varlen_bwd(dolomite_stream_t stream,
 // Backward pass 3: cast dq/dk/dv fp32 accumulators into the packed dqkv.
 // ===========================================================================
 
+// Reduce the (T, H, D) per-q-head fp32 partials over each kv head's G
+// contributors in fixed order and cast into the packed dqkv.
 template <typename T>
 __global__ void __launch_bounds__(256) fa_grad_finalize_kernel(
     const float* __restrict__ dk_acc, const float* __restrict__ dv_acc,
     T* __restrict__ dqkv, int64_t total_kv,
-    int Hkv, int D, int64_t row_ts, int64_t k_off, int64_t kv_hs, int64_t v_off) {
+    int Hkv, int D, int G, int64_t row_ts, int64_t k_off, int64_t kv_hs, int64_t v_off) {
     int64_t kidx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
     bool is_v = kidx >= total_kv;
     if (is_v) kidx -= total_kv;
@@ -1044,13 +1051,16 @@ __global__ void __launch_bounds__(256) fa_grad_finalize_kernel(
     int j = rem / D;
     int d = rem % D;
     const float* src = is_v ? dv_acc : dk_acc;
+    int64_t base = ((t * Hkv + j) * (int64_t)G) * D + d;
+    float acc = 0.f;
+    for (int g = 0; g < G; ++g) acc += src[base + (int64_t)g * D];
     int64_t off = is_v ? v_off : k_off;
-    store_from_f32(&dqkv[t * row_ts + off + (int64_t)j * kv_hs + d], src[kidx]);
+    store_from_f32(&dqkv[t * row_ts + off + (int64_t)j * kv_hs + d], acc);
 }
 
 extern "C" int dolomite_fa_grad_finalize(dolomite_stream_t stream,
                                          const float* dk_acc, const float* dv_acc,
-                                         void* dqkv, int64_t T, int Hkv, int D,
+                                         void* dqkv, int64_t T, int Hkv, int D, int G,
                                          int64_t row_tstride,
                                          int64_t k_off, int64_t kv_hstride, int64_t v_off, int dtype) {
     int64_t total_kv = T * (int64_t)Hkv * D;
@@ -1060,11 +1070,11 @@ extern "C" int dolomite_fa_grad_finalize(dolomite_stream_t stream,
     if (dtype == DOLOMITE_BF16)
         hipLaunchKernelGGL((fa_grad_finalize_kernel<uint16_t>), grid, block, 0, (hipStream_t)stream,
                            dk_acc, dv_acc, (uint16_t*)dqkv, total_kv,
-                           Hkv, D, row_tstride, k_off, kv_hstride, v_off);
+                           Hkv, D, G, row_tstride, k_off, kv_hstride, v_off);
     else
         hipLaunchKernelGGL((fa_grad_finalize_kernel<float>), grid, block, 0, (hipStream_t)stream,
                            dk_acc, dv_acc, (float*)dqkv, total_kv,
-                           Hkv, D, row_tstride, k_off, kv_hstride, v_off);
+                           Hkv, D, G, row_tstride, k_off, kv_hstride, v_off);
     return dol_last_error();
 }
 

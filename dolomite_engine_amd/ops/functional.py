@@ -420,8 +420,10 @@ class VarlenAttention(torch.autograd.Function):
                 ),
                 "fa_bwd_preprocess",
             )
-            dk_acc = torch.zeros(T, lo.Hkv, lo.D, dtype=torch.float32, device=qkv.device)
-            dv_acc = torch.zeros(T, lo.Hkv, lo.D, dtype=torch.float32, device=qkv.device)
+            # (T, H, D) per-q-head partials, written exclusively per
+            # workgroup (no zero-init needed, every valid slot is stored)
+            dk_acc = torch.empty(T, lo.H, lo.D, dtype=torch.float32, device=qkv.device)
+            dv_acc = torch.empty(T, lo.H, lo.D, dtype=torch.float32, device=qkv.device)
             dqkv = torch.empty_like(qkv)
             with hip.prof("fa_varlen_bwd"):
              hip.check(
@@ -439,7 +441,7 @@ class VarlenAttention(torch.autograd.Function):
             hip.check(
                 hip.lib().dolomite_fa_grad_finalize(
                     hip.stream(), hip.ptr(dk_acc), hip.ptr(dv_acc), hip.ptr(dqkv),
-                    T, lo.Hkv, lo.D, lo.row_len,
+                    T, lo.Hkv, lo.D, lo.G, lo.row_len,
                     lo.k_off, lo.kv_hstride, lo.v_off, hip.dt(qkv),
                 ),
                 "fa_grad_finalize",

@@ -149,9 +149,12 @@ int dolomite_fa_varlen_bwd(dolomite_stream_t stream,
                            int64_t v_tstride, int64_t v_hstride,
                            int64_t do_tstride,
                            float scale, int dtype);
+/* dk_acc/dv_acc are (T, H, D) fp32 PER-Q-HEAD partials written exclusively
+ * (no atomics) by dolomite_fa_varlen_bwd; the finalize reduces each kv
+ * head's G contributors in fixed order — deterministic gradients. */
 int dolomite_fa_grad_finalize(dolomite_stream_t stream,
                               const float* dk_acc, const float* dv_acc,
-                              void* dqkv, int64_t T, int Hkv, int D,
+                              void* dqkv, int64_t T, int Hkv, int D, int G,
                               int64_t row_tstride,
                               int64_t k_off, int64_t kv_hstride, int64_t v_off, int dtype);
 
