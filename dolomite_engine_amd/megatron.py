@@ -174,14 +174,80 @@ def _build_shuffle_index(num_samples: int, total_size: int, rng):
     return numpy.concatenate((first, last))
 
 
+
+
+# FIM (fill-in-the-middle) augmentation, gpt_dataset.py:16-20, 513-600.
+FIM_PREFIX = "<fim_prefix>"
+FIM_MIDDLE = "<fim_middle>"
+FIM_SUFFIX = "<fim_suffix>"
+FIM_PAD = "<fim_pad>"
+
+
+def fim_permute(sample, np_rng, fim_rate, fim_spm_rate, tokenizer,
+                suffix_tok_id, prefix_tok_id, middle_tok_id, pad_tok_id):
+    """One document segment -> PSM/SPM FIM rearrangement with probability
+    fim_rate (gpt_dataset.py:513-600, truncate_or_pad=False call form)."""
+    if not np_rng.binomial(1, fim_rate):
+        return sample
+    contents = tokenizer.detokenize(sample)
+    boundaries = list(np_rng.randint(low=0, high=len(contents) + 1, size=2))
+    boundaries.sort()
+    prefix = numpy.array(tokenizer.tokenize(contents[: boundaries[0]]), dtype=numpy.int64)
+    middle = numpy.array(tokenizer.tokenize(contents[boundaries[0] : boundaries[1]]), dtype=numpy.int64)
+    suffix = numpy.array(tokenizer.tokenize(contents[boundaries[1] :]), dtype=numpy.int64)
+    if np_rng.binomial(1, fim_spm_rate):  # SPM (FIM paper variant 2)
+        return numpy.concatenate([[prefix_tok_id, suffix_tok_id], suffix, [middle_tok_id], prefix, middle])
+    return numpy.concatenate([[prefix_tok_id], prefix, [suffix_tok_id], suffix, [middle_tok_id], middle])
+
+
+def fim_transform(sample, np_rng, fim_rate, fim_spm_rate, tokenizer,
+                  suffix_tok_id, prefix_tok_id, middle_tok_id, pad_tok_id):
+    """Apply FIM per document segment (split on eod), then truncate/pad back
+    to the original window length (gpt_dataset.py:170-239)."""
+    sample_len = sample.shape[0]
+    eod = tokenizer.eod
+    breaks = numpy.argwhere(sample == eod)
+    kw = dict(fim_rate=fim_rate, fim_spm_rate=fim_spm_rate, tokenizer=tokenizer,
+              suffix_tok_id=suffix_tok_id, prefix_tok_id=prefix_tok_id,
+              middle_tok_id=middle_tok_id, pad_tok_id=pad_tok_id)
+    if breaks.shape != (0, 1):
+        cur = 0
+        parts = []
+        for loc in numpy.nditer(breaks):
+            if loc - cur > 0:
+                parts += [fim_permute(sample[cur:loc], np_rng, **kw), numpy.array([eod], dtype=numpy.int64)]
+            cur = int(loc) + 1
+        parts.append(fim_permute(sample[cur:], np_rng, **kw))
+        sample = numpy.concatenate(parts)
+    else:
+        sample = fim_permute(sample, np_rng, **kw)
+    diff = sample.shape[0] - sample_len
+    if diff > 0:
+        sample = sample[:sample_len]
+    elif diff < 0:
+        sample = numpy.concatenate([sample, numpy.full(-diff, pad_tok_id, dtype=numpy.int64)])
+    assert sample.shape[0] == sample_len
+    return sample
+
+
 class GPTDataset:
     """Doc/sample/shuffle-index dataset over an MMapIndexedDataset
     (gpt_dataset.py:241-401; indices built in memory, not cached to disk)."""
 
     def __init__(self, indexed_dataset: MMapIndexedDataset, num_samples: int, seq_length: int, seed: int = 1234,
-                 documents: numpy.ndarray | None = None):
+                 documents: numpy.ndarray | None = None,
+                 fim_rate: float = 0.0, fim_spm_rate: float = 0.5, tokenizer=None):
         self.indexed = indexed_dataset
         self.seq_length = seq_length
+        self.fim_rate = fim_rate
+        self.fim_spm_rate = fim_spm_rate
+        self.tokenizer = tokenizer
+        if fim_rate != 0:
+            assert 0 <= fim_rate <= 1 and tokenizer is not None, "FIM needs a tokenizer and 0<=rate<=1"
+            self.np_rng = numpy.random.RandomState(seed=seed)
+            self.suffix_tok_id, self.prefix_tok_id, self.middle_tok_id, self.pad_tok_id = (
+                tokenizer.convert_tokens_to_ids(t) for t in (FIM_SUFFIX, FIM_PREFIX, FIM_MIDDLE, FIM_PAD)
+            )
         if documents is None:
             documents = numpy.arange(len(indexed_dataset), dtype=numpy.int32)
         self.documents = documents
@@ -243,7 +309,13 @@ class GPTDataset:
             parts.append(self.indexed.get(int(self.doc_idx[doc_index_end]), length=int(doc_index_end_offset) + 1))
             text = numpy.concatenate(parts)
         assert len(text) == self.seq_length + 1, (len(text), self.seq_length + 1)
-        return {"text": torch.from_numpy(text.astype(numpy.int64))}
+        text = text.astype(numpy.int64)
+        if self.fim_rate != 0:
+            text = fim_transform(
+                text, self.np_rng, self.fim_rate, self.fim_spm_rate, self.tokenizer,
+                self.suffix_tok_id, self.prefix_tok_id, self.middle_tok_id, self.pad_tok_id,
+            )
+        return {"text": torch.from_numpy(text)}
 
 
 def parse_and_normalize_split(split: str) -> list:
@@ -301,7 +373,8 @@ def train_val_test_samples(num_training_steps: int, micro_batch_size: int, gradi
     return train, val, test
 
 
-def build_train_val_test_datasets(data_path, split: str, sizes: tuple, seq_length: int, seed: int):
+def build_train_val_test_datasets(data_path, split: str, sizes: tuple, seq_length: int, seed: int,
+                                  fim_rate: float = 0.0, fim_spm_rate: float = 0.5, tokenizer=None):
     """Build the train/val/test GPTDatasets, blended across weighted paths
     (data/megatron/__init__.py:18-110 options 1 and 2).
 
@@ -338,7 +411,8 @@ def build_train_val_test_datasets(data_path, split: str, sizes: tuple, seq_lengt
                 per_split.append(None)
             else:
                 docs = numpy.arange(bounds[si], bounds[si + 1], dtype=numpy.int32)
-                per_split.append(GPTDataset(indexed, dsizes[si], seq_length, seed=seed, documents=docs))
+                per_split.append(GPTDataset(indexed, dsizes[si], seq_length, seed=seed, documents=docs,
+                                            fim_rate=fim_rate, fim_spm_rate=fim_spm_rate, tokenizer=tokenizer))
         per_dataset_splits.append(per_split)
 
     out = []

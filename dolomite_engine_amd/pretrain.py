@@ -208,9 +208,20 @@ def main(argv=None):
             tp.num_training_steps, tp.micro_batch_size, tp.gradient_accumulation_steps,
             tp.eval_interval if tp.eval_during_training else None, eval_steps, world,
         )
+        fim_rate = float(ca.get("fim_rate", 0))
+        tok = None
+        if fim_rate:
+            from transformers import AutoTokenizer
+
+            tok = AutoTokenizer.from_pretrained(args.tokenizer_args.tokenizer_name)
+            # megatron-tokenizer façade the FIM transform expects
+            tok.eod = tok.eos_token_id
+            tok.detokenize = tok.decode
+            tok.tokenize = lambda s_, _t=tok: _t.encode(s_, add_special_tokens=False)
         train_ds, val_ds, _test_ds = build_train_val_test_datasets(
             ca["data_path"], ca.get("split", "969,30,1"), sizes,
             seq_length=tp.sequence_length, seed=int(ca.get("seed", args.random_args.seed)),
+            fim_rate=fim_rate, fim_spm_rate=float(ca.get("fim_spm_rate", 0.5)), tokenizer=tok,
         )
         train_loader = MegatronDataLoader(train_ds, tp.micro_batch_size)
         if val_ds is not None and tp.eval_during_training and tp.eval_interval is not None:

@@ -54,6 +54,7 @@ class _Bucket:
         self.shard_grad = torch.zeros_like(self.master)
         self.pending = 0
         self.comm_event = None
+        self.wgrad_event = None  # last deferred-wgrad event touching this bucket
 
 
 def _padded_numel(numel: int, world: int) -> int:
@@ -71,7 +72,10 @@ class ZeRO2Engine:
         bucket_mb: int = 128,
         overlap_comm: bool = True,
         force_collectives: bool = False,
-        defer_wgrad: bool = True,
+        # Measured perf-NEUTRAL on 1 MI355X (attention kernels at 2 WGs/CU
+        # leave no LDS for co-scheduled GEMM workgroups; the streams
+        # time-slice): off by default, available for multi-GPU experiments.
+        defer_wgrad: bool = False,
     ):
         self.model = model
         self.lr = lr
@@ -141,6 +145,7 @@ class ZeRO2Engine:
         if sync:
             for b in self.buckets:
                 b.pending = len(b.params)
+                b.wgrad_event = None
 
     def _grad_hook(self, p: torch.Tensor) -> None:
         # autograd already accumulated into the bucket view — the hook only
@@ -172,10 +177,13 @@ class ZeRO2Engine:
         out.copy_(b.flat_grad)
         return out
 
-    def _launch_reduce(self, b: _Bucket, extra_event=None) -> None:
+    def _launch_reduce(self, b: _Bucket) -> None:
         self.comm_stream.wait_stream(torch.cuda.current_stream())
-        if extra_event is not None:
-            self.comm_stream.wait_event(extra_event)
+        if b.wgrad_event is not None:
+            # a deferred wgrad may still be writing this bucket's views on
+            # the wgrad stream — the reduce must order after it regardless
+            # of which param's completion triggered the launch
+            self.comm_stream.wait_event(b.wgrad_event)
         with torch.cuda.stream(self.comm_stream):
             # fp32 on the wire (reference communication_dtype=fp32)
             dist.reduce_scatter_tensor(b.shard_grad, self._comm_cast(b), op=dist.ReduceOp.AVG)
@@ -390,11 +398,13 @@ def _engine_defer_wgrad(self, weight, dy2, x2, has_bias) -> bool:
     # the allocator must not recycle these while the side stream reads them
     dy2.record_stream(s)
     x2.record_stream(s)
+    for bx in ({b, bias_ent[0]} if bias_ent is not None else {b}):
+        bx.wgrad_event = ev
     if self._sync and self.use_coll:
         for bx in ({b, bias_ent[0]} if bias_ent is not None else {b}):
             bx.pending -= 1
             if bx.pending == 0 and self.overlap:
-                self._launch_reduce(bx, extra_event=ev)
+                self._launch_reduce(bx)
     return True
 
 

@@ -208,3 +208,59 @@ def test_single_path_split_val_loader(tmp_path):
     test_docs = {int(test[i]["text"][0]) for i in range(len(test))}
     assert train_docs and val_docs and test_docs
     assert max(train_docs) < 90 and min(val_docs) >= 90 and val_docs.isdisjoint(test_docs)
+
+
+class _FakeFimTokenizer:
+    """Identity 'tokenizer' over comma-separated token strings, with the
+    FIM sentinel vocabulary (tests the transform, not BPE)."""
+
+    eod = 0
+    _special = {"<fim_suffix>": 9001, "<fim_prefix>": 9002, "<fim_middle>": 9003, "<fim_pad>": 9004}
+
+    def detokenize(self, arr):
+        return ",".join(str(int(x)) for x in arr)
+
+    def tokenize(self, s):
+        return [int(x) for x in s.split(",") if x != ""]
+
+    def convert_tokens_to_ids(self, t):
+        return self._special[t]
+
+
+def test_fim_transform_semantics():
+    """gpt_dataset.py:513-600: PSM/SPM rearrangement per eod-split segment,
+    window length preserved, sentinels present, rate=0 is identity."""
+    from dolomite_engine_amd.megatron import fim_transform
+
+    tok = _FakeFimTokenizer()
+    rng = numpy.random.RandomState(0)
+    sample = numpy.array([5, 6, 7, 0, 8, 9, 10, 11], dtype=numpy.int64)
+
+    out0 = fim_transform(sample.copy(), rng, 0.0, 0.5, tok, 9001, 9002, 9003, 9004)
+    numpy.testing.assert_array_equal(out0, sample)
+
+    rng = numpy.random.RandomState(1)
+    out = fim_transform(sample.copy(), rng, 1.0, 0.5, tok, 9001, 9002, 9003, 9004)
+    assert out.shape == sample.shape
+    assert (out >= 9001).any(), out  # sentinels inserted
+    # eod-separated structure: eod still present (first segment + separator)
+    assert (out == 0).sum() >= 1
+    # the multiset of original tokens in the output is a subset of the input
+    kept = [t for t in out if t < 9000 and t != 0]
+    assert set(kept) <= set(sample.tolist())
+
+
+def test_gpt_dataset_fim_rate(tmp_path):
+    from dolomite_engine_amd.megatron import GPTDataset
+
+    docs = [numpy.arange(1, 30, dtype=numpy.int32) for _ in range(6)]
+    ds = _write_corpus(tmp_path, docs)
+    tok = _FakeFimTokenizer()
+    gpt = GPTDataset(ds, num_samples=8, seq_length=16, seed=3, fim_rate=1.0, tokenizer=tok)
+    for i in range(4):
+        t = gpt[i]["text"]
+        assert t.shape == (17,)
+    # rate 0 path unchanged vs plain dataset
+    a = GPTDataset(ds, num_samples=8, seq_length=16, seed=3)
+    b = GPTDataset(ds, num_samples=8, seq_length=16, seed=3, fim_rate=0.0, tokenizer=tok)
+    torch.testing.assert_close(a[0]["text"], b[0]["text"])
