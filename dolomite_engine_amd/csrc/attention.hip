@@ -225,10 +225,7 @@ __global__ void __launch_bounds__(512, 4) fa_fwd_kernel(
     //   P strips: stride 72 (b16 writes conflict-free; reads 2-way — the
     //     reverse trade costs more write cycles than it saves).
     constexpr int SK = DPAD + 16;    // K/V LDS row stride (elems)
-    // per-wave P strip, TRANSPOSED [64 key][16 q] PI23 rows (see dkv note):
-    // b64-packed writes, A-fragment read through the tr16 ladder;
-    // stride 20 keeps rows 8B-aligned
-    constexpr int STQ = 16 + 4;
+    constexpr int SV = 64 + 8;       // P strip row stride
 
     int tile_id = blockIdx.x, b = blockIdx.y, h = blockIdx.z;
     xcd_remap_tile_bh(tile_id, b, h);
@@ -261,8 +258,8 @@ __global__ void __launch_bounds__(512, 4) fa_fwd_kernel(
     extern __shared__ char smem_raw[];
     __bf16* Klds = (__bf16*)smem_raw;              // [NBUF][64][SK]
     __bf16* Vlds = Klds + NBUF * 64 * SK;          // [NBUF][64 key][SK] (PI23 rows)
-    __bf16* Plds = Vlds + NBUF * 64 * SK;          // [8 waves][64 key][STQ]
-    __bf16* Pw = Plds + wave * 64 * STQ;
+    __bf16* Plds = Vlds + NBUF * 64 * SK;          // [8 waves][16][SV]
+    __bf16* Pw = Plds + wave * 16 * SV;
     // V is stored row-major (coalesced staging) and consumed as the PV
     // B-fragment via ds_read_b64_tr_b16 — no transposed image, no scatter.
 
@@ -294,8 +291,6 @@ __global__ void __launch_bounds__(512, 4) fa_fwd_kernel(
     const int tr_cc = (lane & 3) * 4;
     const unsigned aV0 = (unsigned)(size_t)(Vlds + PI23(tr_g8 + tr_jj) * SK + tr_cc);
     const unsigned aV1 = (unsigned)(size_t)(Vlds + PI23(tr_g8 + 4 + tr_jj) * SK + tr_cc);
-    const unsigned aP0 = (unsigned)(size_t)(Pw + PI23(tr_g8 + tr_jj) * STQ + tr_cc);
-    const unsigned aP1 = (unsigned)(size_t)(Pw + PI23(tr_g8 + 4 + tr_jj) * STQ + tr_cc);
 
     const int kend = min(L, qs + 128);
     const int ntiles = (kend + 63) / 64;
@@ -337,7 +332,6 @@ __global__ void __launch_bounds__(512, 4) fa_fwd_kernel(
         // 4 scores live in 4 registers and P goes straight to the LDS strip
         // (keeps the per-tile live set small for the 128-VGPR cap) ---
         const float scale2 = scale * DOL_LOG2E;  // exp2-domain scores
-        bf16x4 pcb[4];
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
             const int qpos = qs + wave * 16 + lg * 4 + r;
@@ -359,7 +353,7 @@ __global__ void __launch_bounds__(512, 4) fa_fwd_kernel(
 #pragma unroll
             for (int cb = 0; cb < 4; ++cb) {
                 float e = (pv[cb] == -INFINITY) ? 0.f : exp2f(pv[cb] - mnew);
-                pcb[cb][r] = (__bf16)e;
+                Pw[(lg * 4 + r) * SV + cb * 16 + lr] = (__bf16)e;
                 rsum += e;
             }
             rsum = qwave_reduce_sum(rsum);
@@ -368,24 +362,17 @@ __global__ void __launch_bounds__(512, 4) fa_fwd_kernel(
 #pragma unroll
             for (int dc = 0; dc < DCH; ++dc) o_acc[dc][r] *= alpha;
         }
-#pragma unroll
-        for (int cb = 0; cb < 4; ++cb)
-            *(bf16x4*)&Pw[PI23(cb * 16 + lr) * STQ + lg * 4] = pcb[cb];
 
         // --- PV: A = P (this wave's rows), B via the pipelined tr16 ladder
         // over the row-major PI23 V image (issue frag i+1, counted-wait
         // frag i, MFMA — LDS latency hides under the matrix pipe) ---
         {
-            asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");  // exact counts from here
+            bf16x8 pf0 = *(const bf16x8*)&Pw[lr * SV + lg * 8];
+            bf16x8 pf1 = *(const bf16x8*)&Pw[lr * SV + 32 + lg * 8];
+            lgkm_drain2x8(pf0, pf1);
             __builtin_amdgcn_sched_barrier(0);
-            bf16x4 p_lo[2], p_hi[2];
-            tr16_issue<0>(aP0, aP1, p_lo[0], p_hi[0]);
-            tr16_issue<32 * STQ * 2>(aP0, aP1, p_lo[1], p_hi[1]);
             bf16x4 vlo[2], vhi[2];
             tr16_issue<0>(aV0c, aV1c, vlo[0], vhi[0]);
-            lgkm_wait4<2>(p_lo[0], p_hi[0], p_lo[1], p_hi[1]);
-            bf16x8 pf0 = tr16_join8(p_lo[0], p_hi[0]);
-            bf16x8 pf1 = tr16_join8(p_lo[1], p_hi[1]);
 #define FWD_PV_STEP(i)                                                                                  \
     if constexpr ((i) < 2 * DCH) {                                                                      \
         constexpr int kc2_ = (i) / DCH, dc_ = (i) % DCH;                                                \
@@ -489,7 +476,7 @@ static int launch_fa_fwd(hipStream_t stream, const __bf16* q, const __bf16* k, c
                          int64_t q_ts, int64_t q_gs, int64_t k_ts, int64_t k_hs,
                          int64_t v_ts, int64_t v_hs, int max_tiles, float scale) {
     dim3 grid(max_tiles, batch, H), block(512);
-    size_t shmem = ((size_t)(DPAD <= 96 ? 2 : 1) * 64 * (DPAD + 16) * 2 + 8 * 64 * (16 + 4)) * sizeof(__bf16);
+    size_t shmem = ((size_t)(DPAD <= 96 ? 2 : 1) * 64 * (DPAD + 16) * 2 + 8 * 16 * 72) * sizeof(__bf16);
     hipLaunchKernelGGL((fa_fwd_kernel<DPAD>), grid, block, shmem, stream,
                        q, k, v, o, lse, cu, H, Hkv, D, G,
                        q_ts, q_gs, k_ts, k_hs, v_ts, v_hs, (int64_t)H * D, T, scale);
@@ -831,15 +818,11 @@ __global__ void __launch_bounds__(512, 4) fa_bwd_dq_kernel(
     // come from the PI23-rowed row-major K image via the tr16 ladder, like
     // dkv. Saves DPAD*96 LDS elements and the per-piece scatter writes.
     constexpr int SQ = DPAD + 16;              // image stride
-    // per-wave dS strip, TRANSPOSED [64 key][16 q] with PI23 rows: the
-    // C-fragment's 4 q-rows pack into one ds_write_b64 per key block and
-    // the dQ A-fragment reads back through the tr16 ladder (see dkv note)
-    constexpr int STQ = 16 + 4;                // 8B-aligned rows (see dkv stride note)
     extern __shared__ char smem_raw[];
     __bf16* Klds = (__bf16*)smem_raw;          // [64 key][SQ] (PI23 rows)
     __bf16* Vlds = Klds + 64 * SQ;             // [64 key][SQ] (natural rows)
-    __bf16* dSl = Vlds + 64 * SQ;              // [8 waves][64 key][STQ]
-    __bf16* dSw = dSl + wave * 64 * STQ;
+    __bf16* dSl = Vlds + 64 * SQ;              // [8 waves][16 q][ST] (dS strips)
+    __bf16* dSw = dSl + wave * 16 * ST;
 
     const int64_t q_hoff = (int64_t)(h / G) * q_gs + (int64_t)(h % G) * D;
     const int64_t do_hoff = (int64_t)h * D;
@@ -850,8 +833,6 @@ __global__ void __launch_bounds__(512, 4) fa_bwd_dq_kernel(
     const int tr_cc = (lane & 3) * 4;
     const unsigned aK0 = (unsigned)(size_t)(Klds + PI23(tr_g8 + tr_jj) * SQ + tr_cc);
     const unsigned aK1 = (unsigned)(size_t)(Klds + PI23(tr_g8 + 4 + tr_jj) * SQ + tr_cc);
-    const unsigned aS0 = (unsigned)(size_t)(dSw + PI23(tr_g8 + tr_jj) * STQ + tr_cc);
-    const unsigned aS1 = (unsigned)(size_t)(dSw + PI23(tr_g8 + 4 + tr_jj) * STQ + tr_cc);
 
     // this wave's Q and dO fragments (A-layout: i = lr -> q row)
     const int qrow = qs + wave * 16 + lr;
@@ -918,7 +899,6 @@ __global__ void __launch_bounds__(512, 4) fa_bwd_dq_kernel(
                 bf16x8 vb = *(const bf16x8*)&Vlds[(cb * 16 + lr) * SQ + d0];
                 dp = MFMA16(dfr[kc], vb, dp);
             }
-            bf16x4 dsv;
 #pragma unroll
             for (int r = 0; r < 4; ++r) {
                 const int qpos = qs + wave * 16 + lg * 4 + r;
@@ -926,25 +906,19 @@ __global__ void __launch_bounds__(512, 4) fa_bwd_dq_kernel(
                 bool ok = full_tile || (qpos < L && kpos < kend_total && kpos <= qpos);
                 float pv = ok ? exp2f(sc[r] * (scale * DOL_LOG2E) - lsev[r]) : 0.f;
                 float ds = ok ? pv * (dp[r] - delv[r]) * scale : 0.f;
-                dsv[r] = (__bf16)ds;
+                dSw[(lg * 4 + r) * ST + cb * 16 + lr] = (__bf16)ds;
             }
-            *(bf16x4*)&dSw[PI23(cb * 16 + lr) * STQ + lg * 4] = dsv;
         }
 
         // dQ += dS*K (contraction over this tile's keys) — pipelined tr16
-        // ladder over the PI23 K image; dS A-fragments via tr16 from the
-        // transposed strip
+        // ladder over the PI23 K image
         {
-            asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");  // exact counts from here
+            bf16x8 dsf0 = *(const bf16x8*)&dSw[lr * ST + lg * 8];
+            bf16x8 dsf1 = *(const bf16x8*)&dSw[lr * ST + 32 + lg * 8];
+            lgkm_drain2x8(dsf0, dsf1);
             __builtin_amdgcn_sched_barrier(0);
-            bf16x4 s_lo[2], s_hi[2];
-            tr16_issue<0>(aS0, aS1, s_lo[0], s_hi[0]);
-            tr16_issue<32 * STQ * 2>(aS0, aS1, s_lo[1], s_hi[1]);
             bf16x4 klo[2], khi[2];
             tr16_issue<0>(aK0, aK1, klo[0], khi[0]);
-            lgkm_wait4<2>(s_lo[0], s_hi[0], s_lo[1], s_hi[1]);
-            bf16x8 dsf0 = tr16_join8(s_lo[0], s_hi[0]);
-            bf16x8 dsf1 = tr16_join8(s_lo[1], s_hi[1]);
 #define DQ_STEP(i)                                                                                      \
     if constexpr ((i) < 2 * DCH) {                                                                      \
         constexpr int kc2_ = (i) / DCH, dc_ = (i) % DCH;                                                \
@@ -997,7 +971,7 @@ static int launch_fa_bwd(hipStream_t stream, const __bf16* q, const __bf16* k, c
                        q_ts, q_gs, k_ts, k_hs, v_ts, v_hs, do_ts, T, scale);
     int err = dol_last_error();
     if (err) return err;
-    size_t shmem_dq = (size_t)(64 * SQ * 2 + 8 * 64 * (16 + 4)) * sizeof(__bf16);
+    size_t shmem_dq = (size_t)(64 * SQ * 2 + 8 * 16 * ST) * sizeof(__bf16);
     hipLaunchKernelGGL((fa_bwd_dq_kernel<DPAD>), grid, block, shmem_dq, stream,
                        q, k, v, dout, lse, delta, dqkv_q, cu, H, Hkv, D, G,
                        q_ts, q_gs, k_ts, k_hs, v_ts, v_hs, do_ts, T, scale);

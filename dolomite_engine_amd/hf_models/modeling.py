@@ -119,8 +119,14 @@ class RoPE(nn.Module):
         super().__init__()
         self.head_dim = head_dim
         self.base = float(base)
+        self.mscale = 1.0
         self.max_seq_len_cached = 0
         self.reset_parameters(max_position_embeddings)
+
+    def _get_inv_freq(self, device) -> torch.Tensor:
+        return 1.0 / (
+            self.base ** (torch.arange(0, self.head_dim, 2, dtype=torch.float32, device=device) / self.head_dim)
+        )
 
     def reset_parameters(self, seq_len: int | None = None) -> None:
         self._set_cache(seq_len or self.max_seq_len_cached, torch.device("cpu"))
@@ -135,14 +141,12 @@ class RoPE(nn.Module):
         self.max_seq_len_cached = seq_len
         if device is not None and torch.device(device).type == "meta":
             device = "cpu"
-        inv_freq = 1.0 / (
-            self.base ** (torch.arange(0, self.head_dim, 2, dtype=torch.float32, device=device) / self.head_dim)
-        )
+        inv_freq = self._get_inv_freq(device)
         t = torch.arange(seq_len, dtype=torch.float32, device=device)
         freqs = torch.outer(t, inv_freq)
         emb = torch.cat((freqs, freqs), dim=-1)
-        self.cos_cached = emb.cos()
-        self.sin_cached = emb.sin()
+        self.cos_cached = emb.cos() * self.mscale
+        self.sin_cached = emb.sin() * self.mscale
 
     def forward(self, seq_len: int, device) -> tuple[torch.Tensor, torch.Tensor]:
         if seq_len > self.max_seq_len_cached or self.cos_cached.device != torch.device(device):
