@@ -154,6 +154,48 @@ class RoPE(nn.Module):
         return self.cos_cached[:seq_len], self.sin_cached[:seq_len]
 
 
+class YaRNScaledRoPE(RoPE):
+    """NTK-by-parts YaRN scaling (reference position_embedding/rope.py:59-142):
+    per-dim blend of interpolated and extrapolated inverse frequencies with
+    a linear ramp between the beta_fast/beta_slow correction dims, and the
+    attention-magnitude mscale folded into the tables."""
+
+    def __init__(self, head_dim: int, max_position_embeddings: int, base: float,
+                 scale: float = 1.0, original_max_position_embeddings: int = 2048,
+                 extrapolation_factor: float = 1.0, attn_factor: float = 1.0,
+                 beta_fast: int = 32, beta_slow: int = 1):
+        self.scale = float(scale)
+        self.original_max_position_embeddings = original_max_position_embeddings
+        self.extrapolation_factor = extrapolation_factor
+        self.attn_factor = attn_factor
+        self.beta_fast = beta_fast
+        self.beta_slow = beta_slow
+        super().__init__(head_dim, max_position_embeddings, base)
+        self.mscale = (0.1 * math.log(self.scale) + 1.0 if self.scale > 1 else 1.0) * attn_factor
+        # rebuild the cached tables with the final mscale (the base __init__
+        # built them before mscale was known)
+        self.reset_parameters(max_position_embeddings)
+
+    def _get_inv_freq(self, device) -> torch.Tensor:
+        dim = self.head_dim
+        pos_freqs = self.base ** (torch.arange(0, dim, 2, dtype=torch.float32, device=device) / dim)
+        inv_extrapolation = 1.0 / pos_freqs
+        inv_interpolation = 1.0 / (self.scale * pos_freqs)
+
+        def correction_dim(num_rotations):
+            return (dim * math.log(self.original_max_position_embeddings / (num_rotations * 2 * math.pi))) / (
+                2 * math.log(self.base)
+            )
+
+        low = max(math.floor(correction_dim(self.beta_fast)), 0)
+        high = min(math.ceil(correction_dim(self.beta_slow)), dim - 1)
+        if low == high:
+            high += 0.001  # prevent singularity
+        ramp = ((torch.arange(dim // 2, dtype=torch.float32, device=device) - low) / (high - low)).clamp(0, 1)
+        inv_freq_mask = (1 - ramp) * self.extrapolation_factor
+        return inv_interpolation * (1 - inv_freq_mask) + inv_extrapolation * inv_freq_mask
+
+
 def apply_rotary_dense(x, cos, sin):
     """x: (B, H, S, D); cos/sin: (B, 1, S, D) in x dtype (rope.py:104-121)."""
     x1, x2 = torch.chunk(x, 2, dim=-1)
@@ -474,12 +516,17 @@ class GPTDolomiteModel(GPTDolomitePreTrainedModel):
         if self.position_embedding_type == "learned_absolute":
             self.wpe = ParameterizedEmbedding(config.n_positions, self.embed_dim, std=config.initializer_range)
         elif self.position_embedding_type == "rope":
-            self.rope = RoPE(self.head_dim, config.n_positions, config.rope_theta)
             # transformers-5 normalizes rope_scaling=None into a default dict
             rs = config.rope_scaling
-            assert rs is None or (isinstance(rs, dict) and rs.get("rope_type", "default") == "default"), (
-                "rope_scaling (YaRN etc.) is not implemented yet"
-            )
+            if rs is None or (isinstance(rs, dict) and rs.get("rope_type", rs.get("type", "default")) == "default"):
+                self.rope = RoPE(self.head_dim, config.n_positions, config.rope_theta)
+            else:
+                # reference base.py:534-547: YaRN-scaled tables
+                self.rope = YaRNScaledRoPE(
+                    self.head_dim, config.n_positions, config.rope_theta,
+                    scale=rs["factor"],
+                    original_max_position_embeddings=rs["original_max_position_embeddings"],
+                )
         else:
             raise NotImplementedError(f"position_embedding_type {self.position_embedding_type}")
 

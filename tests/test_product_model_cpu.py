@@ -167,3 +167,58 @@ def test_gradient_checkpointing_matches_plain(golden_dir):
     assert grads_a.keys() == grads_b.keys()
     for k in grads_a:
         torch.testing.assert_close(grads_a[k], grads_b[k], rtol=1e-6, atol=1e-7, msg=lambda m: f"{k}: {m}")
+
+
+import os as _os
+
+
+@pytest.mark.skipif(not _os.path.isdir("/root/reference"), reason="reference checkout not present")
+def test_yarn_rope_tables_match_reference():
+    """YaRNScaledRoPE cos/sin tables bit-match the reference implementation
+    (position_embedding/rope.py:59-142) over scales and head dims."""
+    import importlib.util
+    import sys
+
+    sys.path.insert(0, "/root/reference")
+    try:
+        spec = importlib.util.spec_from_file_location(
+            "_ref_rope", "/root/reference/dolomite_engine/hf_models/modeling_utils/position_embedding/rope.py"
+        )
+        ref_rope = importlib.util.module_from_spec(spec)
+        spec.loader.exec_module(ref_rope)
+    finally:
+        sys.path.remove("/root/reference")
+
+    from dolomite_engine_amd.hf_models.modeling import YaRNScaledRoPE
+
+    for head_dim, scale, orig in [(64, 4.0, 2048), (128, 8.0, 4096), (80, 2.0, 1024), (64, 1.0, 2048)]:
+        ours = YaRNScaledRoPE(head_dim, 512, 10000.0, scale=scale, original_max_position_embeddings=orig)
+        ref = ref_rope.YaRNScaledRoPE(
+            head_dim, max_position_embeddings=512, base=10000, scale=scale,
+            original_max_position_embeddings=orig,
+        )
+        torch.testing.assert_close(ours.cos_cached, ref.cos_cached[:512], rtol=0, atol=0)
+        torch.testing.assert_close(ours.sin_cached, ref.sin_cached[:512], rtol=0, atol=0)
+
+
+def test_yarn_model_forward_runs():
+    """A rope_scaling config builds YaRN tables and the padding-free CPU
+    path runs end to end (previously raised NotImplementedError)."""
+    from dolomite_engine_amd.hf_models import GPTDolomiteConfig, GPTDolomiteForCausalLM
+    from dolomite_engine_amd.hf_models.modeling import YaRNScaledRoPE
+
+    torch.manual_seed(0)
+    cfg = GPTDolomiteConfig(
+        vocab_size=128, n_positions=64, n_embd=64, n_layer=1, n_head=4,
+        attention_head_type="mqa", n_inner=128, activation_function="gelu_pytorch_tanh",
+        normalization_function="rmsnorm", position_embedding_type="rope",
+        rope_scaling={"rope_type": "yarn", "factor": 4.0, "original_max_position_embeddings": 16},
+        resid_pdrop=0.0, embd_pdrop=0.0, attn_pdrop=0.0, tie_word_embeddings=False,
+    )
+    cfg._attn_implementation = "flash_attention_2"
+    model = GPTDolomiteForCausalLM(cfg, use_padding_free_transformer=True)
+    assert isinstance(model.transformer.rope, YaRNScaledRoPE)
+    ids = torch.randint(0, 128, (32,))
+    out = model(input_ids=ids, position_ids=torch.arange(32),
+                cu_seqlens=torch.tensor([0, 32], dtype=torch.int32), max_seqlen=32, labels=ids)
+    assert torch.isfinite(out.loss)
