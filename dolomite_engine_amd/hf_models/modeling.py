@@ -196,6 +196,37 @@ class YaRNScaledRoPE(RoPE):
         return inv_interpolation * (1 - inv_freq_mask) + inv_extrapolation * inv_freq_mask
 
 
+class Alibi(nn.Module):
+    """Bloom-style alibi slopes and bias (reference
+    position_embedding/alibi.py:8-45): bias[b, h, k] = slope_h * position_k
+    (row-constant-shifted equivalent of slope * (k - q) under softmax)."""
+
+    def __init__(self, num_heads: int):
+        super().__init__()
+        self.num_heads = num_heads
+        self.reset_parameters()
+
+    def reset_parameters(self) -> None:
+        n = self.num_heads
+        closest = 2 ** math.floor(math.log2(n))
+        base = torch.tensor(2 ** (-(2 ** -(math.log2(closest) - 3))), dtype=torch.float32)
+        slopes = torch.pow(base, torch.arange(1, 1 + closest, dtype=torch.int32))
+        if closest != n:
+            extra_base = torch.tensor(2 ** (-(2 ** -(math.log2(2 * closest) - 3))), dtype=torch.float32)
+            extra = torch.pow(extra_base, torch.arange(1, 1 + 2 * min(closest, n - closest), 2, dtype=torch.int32))
+            slopes = torch.cat([slopes, extra], dim=0)
+        self.slopes = slopes  # plain attribute (see RoPE cache note)
+
+    def forward(self, attention_mask, batch_size: int, key_length: int, device, dtype) -> torch.Tensor:
+        slopes = self.slopes.to(device)
+        if attention_mask is None:
+            pos = torch.arange(key_length, device=device, dtype=torch.float32).unsqueeze(0).unsqueeze(0)
+            pos = pos.expand(batch_size, -1, -1)
+        else:
+            pos = (attention_mask.cumsum(dim=-1) - 1).masked_fill_(attention_mask == 0, 0).unsqueeze(1)
+        return (slopes.unsqueeze(0).unsqueeze(-1) * pos).to(dtype)  # (B, H, K)
+
+
 def apply_rotary_dense(x, cos, sin):
     """x: (B, H, S, D); cos/sin: (B, 1, S, D) in x dtype (rope.py:104-121)."""
     x1, x2 = torch.chunk(x, 2, dim=-1)
@@ -527,6 +558,13 @@ class GPTDolomiteModel(GPTDolomitePreTrainedModel):
                     scale=rs["factor"],
                     original_max_position_embeddings=rs["original_max_position_embeddings"],
                 )
+        elif self.position_embedding_type == "alibi":
+            # reference base.py:261-287 + alibi.py; supported on the dense
+            # eager/sdpa paths (the reference's flash path drops it too)
+            assert not self._use_padding_free_transformer, (
+                "alibi is not supported on the padding-free/flash path (matches the reference)"
+            )
+            self.alibi = Alibi(config.n_head)
         else:
             raise NotImplementedError(f"position_embedding_type {self.position_embedding_type}")
 
@@ -627,6 +665,22 @@ class GPTDolomiteModel(GPTDolomitePreTrainedModel):
             bias = torch.zeros(B, S, S, dtype=hs.dtype, device=device)
             bias = bias.masked_fill(~allowed, torch.finfo(hs.dtype).min)
             attention_bias = bias.unsqueeze(1)
+        if self.position_embedding_type == "alibi":
+            # (B, H, 1, K) additive slopes*position bias on top of the mask
+            # (reference base.py:484-503, _get_maybe_causal_mask:569-598).
+            # Faithful quirk: the reference's SDPA branch folds alibi into
+            # the mask ONLY when a padding mask exists — sdpa + alibi with
+            # attention_mask=None silently runs without alibi. We mirror
+            # that for drop-in parity; the eager branch always applies it.
+            if self.attention_implementation == "sdpa" and attention_mask is None:
+                pass
+            else:
+                ab = self.alibi(attention_mask, B, S, device, hs.dtype).unsqueeze(2)
+                if attention_bias is None:
+                    causal = torch.ones(S, S, dtype=torch.bool, device=device).tril()
+                    cb = torch.zeros(S, S, dtype=hs.dtype, device=device)
+                    attention_bias = cb.masked_fill(~causal, torch.finfo(hs.dtype).min).unsqueeze(0).unsqueeze(0)
+                attention_bias = attention_bias + ab
 
         for block in self.h:
             if getattr(block, "_gradient_checkpointing", False) and self.training:

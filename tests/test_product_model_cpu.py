@@ -222,3 +222,57 @@ def test_yarn_model_forward_runs():
     out = model(input_ids=ids, position_ids=torch.arange(32),
                 cu_seqlens=torch.tensor([0, 32], dtype=torch.int32), max_seqlen=32, labels=ids)
     assert torch.isfinite(out.loss)
+
+
+@pytest.mark.skipif(not _os.path.isdir("/root/reference"), reason="reference checkout not present")
+@pytest.mark.parametrize("impl", ["eager", "sdpa"])
+def test_alibi_dense_matches_reference(impl):
+    """position_embedding_type=alibi on the dense paths vs the reference
+    model run directly (base.py:261-287, alibi.py slopes)."""
+    from oracle.ref_shim import make_reference_config, make_reference_model
+
+    kw = dict(
+        vocab_size=160, n_positions=64, n_embd=48, n_layer=2, n_head=6,
+        attention_head_type="mqa", n_inner=96, activation_function="gelu_pytorch_tanh",
+        normalization_function="rmsnorm", position_embedding_type="alibi",
+        resid_pdrop=0.0, embd_pdrop=0.0, attn_pdrop=0.0, tie_word_embeddings=False,
+    )
+    torch.manual_seed(11)
+    rcfg = make_reference_config(**kw)
+    rmodel = make_reference_model(rcfg, attn_implementation=impl)
+    rmodel.eval()
+
+    from dolomite_engine_amd.hf_models import GPTDolomiteConfig, GPTDolomiteForCausalLM
+
+    cfg = GPTDolomiteConfig(**kw)
+    cfg._attn_implementation = impl
+    model = GPTDolomiteForCausalLM(cfg)
+    model.load_state_dict(rmodel.state_dict())
+    model.eval()
+
+    g = torch.Generator().manual_seed(5)
+    ids = torch.randint(0, 160, (2, 33), generator=g)
+    mask = torch.ones(2, 33, dtype=torch.long)
+    mask[1, :7] = 0  # left padding on row 1
+    with torch.no_grad():
+        ref_full = rmodel(input_ids=ids).logits
+        our_full = model(input_ids=ids).logits
+        ref_mask = rmodel(input_ids=ids, attention_mask=mask).logits
+        our_mask = model(input_ids=ids, attention_mask=mask).logits
+    torch.testing.assert_close(our_full, ref_full, rtol=1e-4, atol=1e-5)
+    valid = mask.bool()
+    torch.testing.assert_close(our_mask[valid], ref_mask[valid], rtol=1e-4, atol=1e-5)
+
+
+def test_alibi_padding_free_raises():
+    from dolomite_engine_amd.hf_models import GPTDolomiteConfig, GPTDolomiteForCausalLM
+
+    cfg = GPTDolomiteConfig(
+        vocab_size=64, n_positions=32, n_embd=32, n_layer=1, n_head=4,
+        attention_head_type="mqa", n_inner=64, position_embedding_type="alibi",
+        normalization_function="rmsnorm", activation_function="gelu_pytorch_tanh",
+        tie_word_embeddings=False,
+    )
+    cfg._attn_implementation = "flash_attention_2"
+    with pytest.raises(AssertionError):
+        GPTDolomiteForCausalLM(cfg, use_padding_free_transformer=True)
