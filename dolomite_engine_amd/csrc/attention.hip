@@ -248,13 +248,16 @@ __global__ void __launch_bounds__(512, 4) fa_fwd_kernel(
     const int lr = lane & 15;   // fragment col / A-row index
     const int lg = lane >> 4;   // fragment k-group / C-row group
 
-    // Measured NEGATIVE and disabled: the NBUF=2 double-buffered K/V
-    // pipeline (issue-early loads, write-late before the single barrier)
-    // cost fa_fwd 5.44 -> 5.84 ms at B=16 — the write pass serializes
-    // behind the MFMAs exactly as guide T14's "write BEFORE the barrier"
-    // variant warns, and at 2 workgroups/CU the co-resident workgroup
-    // already hides the staging latency the pipeline was buying.
-    constexpr int NBUF = 1;
+    // NBUF=2 (DPAD<=96): double-buffered K/V staged by LDS-DMA
+    // (buffer_load ... lds): the next tile's loads are ISSUED at the top of
+    // this tile's compute and land directly in the inactive buffer — no
+    // destination registers, no write-late pass (the register-staged
+    // variant measured -7% because that pass serialized behind the MFMAs),
+    // one barrier per tile with a vmcnt drain. OOB buffer reads return 0,
+    // which implements both the tail-key guard and the MFMA pad-column
+    // zero-fill for free. DPAD=128 keeps the single-buffer schedule
+    // (double-buffering would exceed the 80 KB 2-workgroup LDS budget).
+    constexpr int NBUF = (DPAD <= 96) ? 2 : 1;
     extern __shared__ char smem_raw[];
     __bf16* Klds = (__bf16*)smem_raw;              // [NBUF][64][SK]
     __bf16* Vlds = Klds + NBUF * 64 * SK;          // [NBUF][64 key][SK] (PI23 rows)
@@ -395,48 +398,72 @@ __global__ void __launch_bounds__(512, 4) fa_fwd_kernel(
     };
 
     if constexpr (NBUF == 2) {
-        stage_direct(0, Klds, Vlds);
+        // ---- LDS-DMA staging setup ----
+        // Chunk = one buffer_load_dwordx4..lds instruction: 64 lanes x 16 B
+        // of contiguous LDS. Per tensor image: 64*SK*2/1024 chunks; this
+        // wave owns chunks c = wave, wave+8, ... across K then V.
+        constexpr int CH = (64 * SK * 2) / 1024;   // chunks per image (SK%8==0)
+        constexpr int NC = (2 * CH + 7) / 8;       // chunks this wave may own
+        // Per-sequence, per-kv-head buffer descriptors: OOB voffsets
+        // (key >= L, or MFMA pad columns sent out of range) read as zero.
+        auto rsrcK = __builtin_amdgcn_make_buffer_rsrc(
+            (void*)(k + (int64_t)s0 * k_ts + (int64_t)kvh * k_hs), 0,
+            (unsigned)(((int64_t)(L - 1) * k_ts + D) * 2), 0);
+        auto rsrcV = __builtin_amdgcn_make_buffer_rsrc(
+            (void*)(v + (int64_t)s0 * v_ts + (int64_t)kvh * v_hs), 0,
+            (unsigned)(((int64_t)(L - 1) * v_ts + D) * 2), 0);
+        // Loop-invariant per-lane voffsets (bytes) for key tile 0; per tile
+        // add ks*stride*2. Pad columns get 0x40000000 (stays OOB after the
+        // per-tile add — tensors on this path are < 1 GiB).
+        unsigned choff[NC];
+        bool ch_is_v[NC];
+#pragma unroll
+        for (int i = 0; i < NC; ++i) {
+            int c = wave + i * 8;
+            bool isv = c >= CH;
+            int cc = isv ? c - CH : c;
+            int e0 = cc * 512 + lane * 8;
+            int row = e0 / SK;
+            int col = e0 % SK;
+            int key = isv ? PI23(row) : row;   // V image rows are PI23-placed
+            int64_t ts = isv ? v_ts : k_ts;
+            unsigned off = (unsigned)((key * ts + col) * 2);
+            if (col >= D || c >= 2 * CH) off = 0x40000000u;
+            choff[i] = off;
+            ch_is_v[i] = isv;
+        }
+        const unsigned kstep2 = (unsigned)(k_ts * 2) * 64u;  // voffset delta per 64-key tile
+        const unsigned vstep2 = (unsigned)(v_ts * 2) * 64u;
+
+        auto stage_dma = [&](int kt_, int buf_) {
+            __bf16* Kb = Klds + buf_ * 64 * SK;
+            __bf16* Vb = Vlds + buf_ * 64 * SK;
+#pragma unroll
+            for (int i = 0; i < NC; ++i) {
+                int c = wave + i * 8;
+                if (c >= 2 * CH) break;
+                bool isv = ch_is_v[i];
+                int cc = isv ? c - CH : c;
+                char* ldsp = (char*)(isv ? Vb : Kb) + cc * 1024;
+                unsigned vo = choff[i] + (unsigned)kt_ * (isv ? vstep2 : kstep2);
+                __builtin_amdgcn_raw_ptr_buffer_load_lds(
+                    isv ? rsrcV : rsrcK,
+                    (__attribute__((address_space(3))) void*)ldsp, 16, vo, 0, 0, 0);
+            }
+        };
+
+        stage_dma(0, 0);
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
         __syncthreads();
         for (int kt = 0; kt < ntiles; ++kt) {
             const int cur = kt & 1;
-            const int ks = kt * 64;
-            // T14 issue-early: next tile's K/V global loads into registers
-            bf16x8 stK[NP], stV[NP];
-            const bool have_next = kt + 1 < ntiles;
-            if (have_next) {
-                const int ksn = ks + 64;
-#pragma unroll
-                for (int i = 0; i < NP; ++i) {
-                    int pidx = (int)threadIdx.x + i * 512;
-                    bool act = pidx < PIECES;
-                    int key = act ? pidx / (DPAD / 8) : 0;
-                    int d0 = act ? (pidx % (DPAD / 8)) * 8 : 0;
-                    bool kv_valid = act && (ksn + key) < kend;
-                    const __bf16* kp = k + (int64_t)(s0 + (kv_valid ? ksn + key : 0)) * k_ts + (int64_t)kvh * k_hs + d0;
-                    stK[i] = load_bf16x8_guard(kp, d0, D, kv_valid);
-                    const __bf16* vp = v + (int64_t)(s0 + (kv_valid ? ksn + key : 0)) * v_ts + (int64_t)kvh * v_hs + d0;
-                    stV[i] = load_bf16x8_guard(vp, d0, D, kv_valid);
-                }
-            }
+            if (kt + 1 < ntiles) stage_dma(kt + 1, cur ^ 1);  // in flight under compute
             compute_tile(Klds + cur * 64 * SK,
                          aV0 + (unsigned)(cur * 64 * SK * 2),
-                         aV1 + (unsigned)(cur * 64 * SK * 2), ks);
-            // write-late: staged registers into the other buffer; the one
-            // barrier also retires this tile's reads before the next write
-            if (have_next) {
-                __bf16* Kw = Klds + (cur ^ 1) * 64 * SK;
-                __bf16* Vw = Vlds + (cur ^ 1) * 64 * SK;
-#pragma unroll
-                for (int i = 0; i < NP; ++i) {
-                    int pidx = (int)threadIdx.x + i * 512;
-                    if (pidx < PIECES) {
-                        int key = pidx / (DPAD / 8);
-                        int d0 = (pidx % (DPAD / 8)) * 8;
-                        *(bf16x8*)&Kw[key * SK + d0] = stK[i];
-                        *(bf16x8*)&Vw[PI23(key) * SK + d0] = stV[i];
-                    }
-                }
-            }
+                         aV1 + (unsigned)(cur * 64 * SK * 2), kt * 64);
+            // the DMA into buf cur^1 must land before the next iteration
+            // reads it; the barrier then publishes it workgroup-wide
+            asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
             __syncthreads();
         }
     } else {
