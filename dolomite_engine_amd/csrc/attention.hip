@@ -135,7 +135,6 @@ __device__ __forceinline__ void lgkm_wait8(bf16x4& a, bf16x4& b, bf16x4& c, bf16
                  : "i"(N));
 }
 
-typedef __bf16 bf16x8v __attribute__((ext_vector_type(8)));
 __device__ __forceinline__ bf16x8 tr16_join8(bf16x4 lo, bf16x4 hi) {
     bf16x8 out;
 #pragma unroll
@@ -303,7 +302,6 @@ __global__ void __launch_bounds__(512, 4) fa_fwd_kernel(
     if (__builtin_amdgcn_readfirstlane(threadIdx.x) >= 256) __builtin_amdgcn_s_setprio(1);
 
     constexpr int PIECES = 64 * DPAD / 8;  // 8-elem staging pieces
-    constexpr int NP = (PIECES + 511) / 512;
 
     // direct cooperative staging (prologue and the NBUF==1 path)
     auto stage_direct = [&](int ks_, __bf16* Kw, __bf16* Vw) {
@@ -609,7 +607,6 @@ __global__ void __launch_bounds__(512, DPAD <= 96 ? 4 : 2) fa_bwd_dkv_kernel(
     int64_t do_ts, int64_t T_total, float scale) {
     constexpr int KCH = DPAD / 32;
     constexpr int DCH = DPAD / 16;
-    constexpr int ST = 64 + 8;                 // strip stride (see fa_fwd note)
 
     int tile_id = blockIdx.x, b = blockIdx.y, h = blockIdx.z;
     xcd_remap_tile_bh(tile_id, b, h);
