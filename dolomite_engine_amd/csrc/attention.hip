@@ -247,16 +247,18 @@ __global__ void __launch_bounds__(512, 4) fa_fwd_kernel(
     const int lr = lane & 15;   // fragment col / A-row index
     const int lg = lane >> 4;   // fragment k-group / C-row group
 
-    // NBUF=2 (DPAD<=96): double-buffered K/V staged by LDS-DMA
-    // (buffer_load ... lds): the next tile's loads are ISSUED at the top of
-    // this tile's compute and land directly in the inactive buffer — no
-    // destination registers, no write-late pass (the register-staged
-    // variant measured -7% because that pass serialized behind the MFMAs),
-    // one barrier per tile with a vmcnt drain. OOB buffer reads return 0,
-    // which implements both the tail-key guard and the MFMA pad-column
-    // zero-fill for free. DPAD=128 keeps the single-buffer schedule
-    // (double-buffering would exceed the 80 KB 2-workgroup LDS budget).
-    constexpr int NBUF = (DPAD <= 96) ? 2 : 1;
+    // Two double-buffered staging pipelines were built and MEASURED OUT:
+    //   (a) register-staged issue-early/write-late: 5.44 -> 5.84 ms (the
+    //       write pass serializes behind the MFMAs, guide T14's warning);
+    //   (b) LDS-DMA (buffer_load..lds) into the inactive buffer: wall-time
+    //       NEUTRAL (5.65 ms) and NUMERICALLY UNSAFE as compiled — hipcc
+    //       waterfalls the buffer descriptors it cannot prove wave-uniform
+    //       (T20) and issues some chunk DMAs under partial exec masks, so
+    //       inactive lanes leave stale LDS in the image (flaky parity
+    //       failures across runs; see git history for the full variant).
+    // At 2 workgroups/CU the co-resident workgroup already hides staging
+    // latency, so the single-buffer two-barrier schedule stays.
+    constexpr int NBUF = 1;
     extern __shared__ char smem_raw[];
     __bf16* Klds = (__bf16*)smem_raw;              // [NBUF][64][SK]
     __bf16* Vlds = Klds + NBUF * 64 * SK;          // [NBUF][64 key][SK] (PI23 rows)
