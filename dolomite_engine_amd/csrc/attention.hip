@@ -803,8 +803,11 @@ __global__ void __launch_bounds__(512, DPAD <= 96 ? 4 : 2) fa_bwd_dkv_kernel(
             const int d = dc * 16 + lr;
             if (kpos < kend && d < D) {
                 int64_t idx = ((int64_t)(s0 + kpos) * H + h) * D + d;
-                dk_acc[idx] = dkr[dc][r];
-                dv_acc[idx] = dvr[dc][r];
+                // non-temporal: the 1.3 GB of fp32 partials per launch must
+                // not evict the XCD-resident Q/dO slices (PMC: dkv read
+                // traffic 3.6 GB/launch vs ~1.4 GB ideal with plain stores)
+                __builtin_nontemporal_store(dkr[dc][r], &dk_acc[idx]);
+                __builtin_nontemporal_store(dvr[dc][r], &dv_acc[idx]);
             }
         }
     }
@@ -1053,7 +1056,7 @@ __global__ void __launch_bounds__(256) fa_grad_finalize_kernel(
     const float* src = is_v ? dv_acc : dk_acc;
     int64_t base = ((t * Hkv + j) * (int64_t)G) * D + d;
     float acc = 0.f;
-    for (int g = 0; g < G; ++g) acc += src[base + (int64_t)g * D];
+    for (int g = 0; g < G; ++g) acc += __builtin_nontemporal_load(&src[base + (int64_t)g * D]);
     int64_t off = is_v ? v_off : k_off;
     store_from_f32(&dqkv[t * row_ts + off + (int64_t)j * kv_hs + d], acc);
 }
