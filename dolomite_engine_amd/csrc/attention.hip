@@ -333,8 +333,14 @@ __global__ void __launch_bounds__(512, 4) fa_fwd_kernel(
 
         // --- mask + online softmax (fp32), streamed per C-row: the row's
         // 4 scores live in 4 registers and P goes straight to the LDS strip
-        // (keeps the per-tile live set small for the 128-VGPR cap) ---
+        // (keeps the per-tile live set small for the 128-VGPR cap).
+        // Wave-uniform mask split: full tiles skip the per-element compare
+        // chain entirely (the kernel is issue-bound); the -INF guards for
+        // the FIRST tile's running max stay in both branches. ---
         const float scale2 = scale * DOL_LOG2E;  // exp2-domain scores
+        // (a wave-uniform full-tile/masked split was tried here like the
+        // backward kernels' — it pushed fwd from 96 to 128 VGPRs with loop
+        // spills, so fwd keeps the single masked body)
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
             const int qpos = qs + wave * 16 + lg * 4 + r;
@@ -714,6 +720,9 @@ __global__ void __launch_bounds__(512, DPAD <= 96 ? 4 : 2) fa_bwd_dkv_kernel(
         // block's softmax straight into the LDS strips. Merging the two
         // loops keeps only ONE C-fragment pair (8 VGPRs) live instead of
         // four (32) — the register headroom the (512,4) cap needs.
+        // The kernels are ISSUE-bound (SQ counters): the causal-mask
+        // compare/select chain runs only on diagonal/edge tiles via a
+        // wave-uniform branch; ~94% of tiles take the maskless body.
         const bool full_tile = (ks + wave * 16 + 16 <= qs + 1) && (kend == ks + 64) && (qs + 64 <= L);
 #pragma unroll
         for (int cb = 0; cb < 4; ++cb) {
@@ -732,14 +741,23 @@ __global__ void __launch_bounds__(512, DPAD <= 96 ? 4 : 2) fa_bwd_dkv_kernel(
             float lsev = qok ? lse[(int64_t)h * T_total + s0 + qpos] * DOL_LOG2E : 0.f;
             float delv = qok ? delta[(int64_t)h * T_total + s0 + qpos] : 0.f;
             bf16x4 pvv, dsv;
+            if (__builtin_amdgcn_readfirstlane(full_tile ? 1 : 0)) {
 #pragma unroll
-            for (int r = 0; r < 4; ++r) {
-                const int kpos = ks + wave * 16 + lg * 4 + r;
-                bool ok = full_tile || (qok && kpos < kend && kpos <= qpos);
-                float pv = ok ? exp2f(st[r] * (scale * DOL_LOG2E) - lsev) : 0.f;
-                float ds = ok ? pv * (dpt[r] - delv) * scale : 0.f;
-                pvv[r] = (__bf16)pv;
-                dsv[r] = (__bf16)ds;
+                for (int r = 0; r < 4; ++r) {
+                    float pv = exp2f(st[r] * (scale * DOL_LOG2E) - lsev);
+                    pvv[r] = (__bf16)pv;
+                    dsv[r] = (__bf16)(pv * (dpt[r] - delv) * scale);
+                }
+            } else {
+#pragma unroll
+                for (int r = 0; r < 4; ++r) {
+                    const int kpos = ks + wave * 16 + lg * 4 + r;
+                    bool ok = qok && kpos < kend && kpos <= qpos;
+                    float pv = ok ? exp2f(st[r] * (scale * DOL_LOG2E) - lsev) : 0.f;
+                    float ds = ok ? pv * (dpt[r] - delv) * scale : 0.f;
+                    pvv[r] = (__bf16)pv;
+                    dsv[r] = (__bf16)ds;
+                }
             }
             const int srow = PI23(cb * 16 + lr) * STS + wave * 16 + lg * 4;
             *(bf16x4*)&PTl[srow] = pvv;
@@ -915,6 +933,7 @@ __global__ void __launch_bounds__(512, 4) fa_bwd_dq_kernel(
 
         // Per key-column block: S = Q*K^T and dP = dO*V^T MFMAs, then that
         // block's dS straight into the wave's strip (one C-pair live).
+        // Wave-uniform mask split (see dkv note).
         const bool full_tile = (ks + 64 <= qs + wave * 16 + 1) && (ks + 64 <= kend_total) && (qs + wave * 16 + 16 <= L);
 #pragma unroll
         for (int cb = 0; cb < 4; ++cb) {
@@ -928,14 +947,22 @@ __global__ void __launch_bounds__(512, 4) fa_bwd_dq_kernel(
                 bf16x8 vb = *(const bf16x8*)&Vlds[(cb * 16 + lr) * SQ + d0];
                 dp = MFMA16(dfr[kc], vb, dp);
             }
+            if (__builtin_amdgcn_readfirstlane(full_tile ? 1 : 0)) {
 #pragma unroll
-            for (int r = 0; r < 4; ++r) {
-                const int qpos = qs + wave * 16 + lg * 4 + r;
-                const int kpos = ks + cb * 16 + lr;
-                bool ok = full_tile || (qpos < L && kpos < kend_total && kpos <= qpos);
-                float pv = ok ? exp2f(sc[r] * (scale * DOL_LOG2E) - lsev[r]) : 0.f;
-                float ds = ok ? pv * (dp[r] - delv[r]) * scale : 0.f;
-                dSw[(lg * 4 + r) * ST + cb * 16 + lr] = (__bf16)ds;
+                for (int r = 0; r < 4; ++r) {
+                    float pv = exp2f(sc[r] * (scale * DOL_LOG2E) - lsev[r]);
+                    dSw[(lg * 4 + r) * ST + cb * 16 + lr] = (__bf16)(pv * (dp[r] - delv[r]) * scale);
+                }
+            } else {
+#pragma unroll
+                for (int r = 0; r < 4; ++r) {
+                    const int qpos = qs + wave * 16 + lg * 4 + r;
+                    const int kpos = ks + cb * 16 + lr;
+                    bool ok = qpos < L && kpos < kend_total && kpos <= qpos;
+                    float pv = ok ? exp2f(sc[r] * (scale * DOL_LOG2E) - lsev[r]) : 0.f;
+                    float ds = ok ? pv * (dp[r] - delv[r]) * scale : 0.f;
+                    dSw[(lg * 4 + r) * ST + cb * 16 + lr] = (__bf16)ds;
+                }
             }
         }
 

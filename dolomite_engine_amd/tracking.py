@@ -26,23 +26,56 @@ class RunningMean:
 
 
 class ExperimentsTracker:
-    def __init__(self, tracker_name: str | None = None, run_dir: str | None = None):
+    """reference utils/experiments_tracker.py surface: tracker_name selects
+    aim / wandb (used when the library is importable, as the reference
+    does) with the offline 'jsonl' sink always available."""
+
+    def __init__(self, tracker_name: str | None = None, run_dir: str | None = None, **init_kwargs):
         self.enabled = get_rank() == 0
         self._fh = None
-        if self.enabled and tracker_name == "jsonl" and run_dir is not None:
+        self._aim_run = None
+        self._wandb = None
+        if not self.enabled:
+            return
+        if tracker_name == "jsonl" and run_dir is not None:
             Path(run_dir).mkdir(parents=True, exist_ok=True)
             self._fh = open(Path(run_dir) / "metrics.jsonl", "a")
-        elif tracker_name not in (None, "jsonl"):
-            log_rank_0(f"tracker '{tracker_name}' unavailable offline; falling back to logs", logging.WARNING)
+        elif tracker_name == "aim":
+            try:
+                from aim import Run  # noqa: PLC0415
+
+                self._aim_run = Run(repo=init_kwargs.get("repo", run_dir), experiment=init_kwargs.get("experiment"))
+            except ImportError:
+                log_rank_0("tracker 'aim' requested but aim is not installed; falling back to logs", logging.WARNING)
+        elif tracker_name == "wandb":
+            try:
+                import wandb  # noqa: PLC0415
+
+                self._wandb = wandb
+                wandb.init(**init_kwargs)
+            except ImportError:
+                log_rank_0("tracker 'wandb' requested but wandb is not installed; falling back to logs", logging.WARNING)
+        elif tracker_name is not None:
+            log_rank_0(f"unknown tracker '{tracker_name}'; falling back to logs", logging.WARNING)
 
     def track(self, values: dict, step: int | None = None, context: str | None = None) -> None:
         if self._fh is not None:
             self._fh.write(json.dumps({"step": step, "context": context, **values}) + "\n")
             self._fh.flush()
+        if self._aim_run is not None:
+            for k, v in values.items():
+                self._aim_run.track(v, name=k, step=step, context={"subset": context} if context else None)
+        if self._wandb is not None:
+            prefix = f"{context}/" if context else ""
+            self._wandb.log({f"{prefix}{k}": v for k, v in values.items()}, step=step)
 
     def finish(self) -> None:
         if self._fh is not None:
             self._fh.close()
+        if self._aim_run is not None:
+            self._aim_run.close()
+        if self._wandb is not None:
+            self._wandb.finish()
 
 
 def track_train_metrics(
