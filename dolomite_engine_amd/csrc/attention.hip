@@ -299,6 +299,16 @@ __global__ void __launch_bounds__(512, 4) fa_fwd_kernel(
     const int kend = min(L, qs + 128);
     const int ntiles = (kend + 63) / 64;
 
+    if (D < DPAD) {  // pad columns zeroed once (see dkv note)
+        for (int pidx = threadIdx.x; pidx < 64 * (DPAD - D) / 8; pidx += 512) {
+            int key = pidx / ((DPAD - D) / 8);
+            int d0 = D + (pidx % ((DPAD - D) / 8)) * 8;
+            bf16x8 z = {};
+            *(bf16x8*)&Klds[key * SK + d0] = z;
+            *(bf16x8*)&Vlds[PI23(key) * SK + d0] = z;
+        }
+    }
+
     // T5 static priority: the later-dispatched half of an 8-wave workgroup
     // loses VALU arbitration; one setprio for it, no per-cluster flips.
     if (__builtin_amdgcn_readfirstlane(threadIdx.x) >= 256) __builtin_amdgcn_s_setprio(1);
@@ -307,14 +317,25 @@ __global__ void __launch_bounds__(512, 4) fa_fwd_kernel(
 
     // direct cooperative staging (prologue and the NBUF==1 path)
     auto stage_direct = [&](int ks_, __bf16* Kw, __bf16* Vw) {
-        for (int pidx = threadIdx.x; pidx < PIECES; pidx += 512) {
-            int key = pidx / (DPAD / 8);
-            int d0 = (pidx % (DPAD / 8)) * 8;
-            bool kv_valid = (ks_ + key) < kend;
-            const __bf16* kp = k + (int64_t)(s0 + (kv_valid ? ks_ + key : 0)) * k_ts + (int64_t)kvh * k_hs + d0;
-            *(bf16x8*)&Kw[key * SK + d0] = load_bf16x8_guard(kp, d0, D, kv_valid);
-            const __bf16* vp = v + (int64_t)(s0 + (kv_valid ? ks_ + key : 0)) * v_ts + (int64_t)kvh * v_hs + d0;
-            *(bf16x8*)&Vw[PI23(key) * SK + d0] = load_bf16x8_guard(vp, d0, D, kv_valid);
+        if (ks_ + 64 <= kend) {
+            for (int pidx = threadIdx.x; pidx < 64 * D / 8; pidx += 512) {
+                int key = pidx / (D / 8);
+                int d0 = (pidx % (D / 8)) * 8;
+                *(bf16x8*)&Kw[key * SK + d0] =
+                    *(const bf16x8*)(k + (int64_t)(s0 + ks_ + key) * k_ts + (int64_t)kvh * k_hs + d0);
+                *(bf16x8*)&Vw[PI23(key) * SK + d0] =
+                    *(const bf16x8*)(v + (int64_t)(s0 + ks_ + key) * v_ts + (int64_t)kvh * v_hs + d0);
+            }
+        } else {
+            for (int pidx = threadIdx.x; pidx < PIECES; pidx += 512) {
+                int key = pidx / (DPAD / 8);
+                int d0 = (pidx % (DPAD / 8)) * 8;
+                bool kv_valid = (ks_ + key) < kend;
+                const __bf16* kp = k + (int64_t)(s0 + (kv_valid ? ks_ + key : 0)) * k_ts + (int64_t)kvh * k_hs + d0;
+                *(bf16x8*)&Kw[key * SK + d0] = load_bf16x8_guard(kp, d0, D, kv_valid);
+                const __bf16* vp = v + (int64_t)(s0 + (kv_valid ? ks_ + key : 0)) * v_ts + (int64_t)kvh * v_hs + d0;
+                *(bf16x8*)&Vw[PI23(key) * SK + d0] = load_bf16x8_guard(vp, d0, D, kv_valid);
+            }
         }
     };
 
@@ -699,19 +720,47 @@ __global__ void __launch_bounds__(512, DPAD <= 96 ? 4 : 2) fa_bwd_dkv_kernel(
     const int64_t q_hoff = (int64_t)(h / G) * q_gs + (int64_t)(h % G) * D;
     const int64_t do_hoff = (int64_t)h * D;
 
+    // MFMA pad columns [D, DPAD) zeroed ONCE: the fast staging path below
+    // then only writes the real D columns per tile (guards and pad-fill
+    // gone from the per-tile loop; the guarded edge path rewrites pads
+    // with zeros, so the invariant holds under any tile interleave)
+    if (D < DPAD) {
+        for (int pidx = threadIdx.x; pidx < 64 * (DPAD - D) / 8; pidx += 512) {
+            int qq = pidx / ((DPAD - D) / 8);
+            int d0 = D + (pidx % ((DPAD - D) / 8)) * 8;
+            bf16x8 z = {};
+            *(bf16x8*)&Qlds[PI23(qq) * SQ + d0] = z;
+            *(bf16x8*)&dOl[PI23(qq) * SQ + d0] = z;
+        }
+    }
+
     for (int qt = qt0; qt < nqt; ++qt) {
         const int qs = qt * 64;
         __syncthreads();  // previous iteration's image reads done
         {
             const int pieces = 64 * DPAD / 8;
-            for (int pidx = threadIdx.x; pidx < pieces; pidx += 512) {
-                int qq = pidx / (DPAD / 8);
-                int d0 = (pidx % (DPAD / 8)) * 8;
-                bool valid = (qs + qq) < L;
-                const __bf16* qp = q + (int64_t)(s0 + (valid ? qs + qq : 0)) * q_ts + q_hoff + d0;
-                *(bf16x8*)&Qlds[PI23(qq) * SQ + d0] = load_bf16x8_guard(qp, d0, D, valid);
-                const __bf16* dp = dout + (int64_t)(s0 + (valid ? qs + qq : 0)) * do_ts + do_hoff + d0;
-                *(bf16x8*)&dOl[PI23(qq) * SQ + d0] = load_bf16x8_guard(dp, d0, D, valid);
+            if (qs + 64 <= L) {
+                // full interior tile (every tile when seqlens divide 64):
+                // guard-free staging over the real D columns only (pads
+                // pre-zeroed above; D % 8 == 0 on this path)
+                for (int pidx = threadIdx.x; pidx < 64 * D / 8; pidx += 512) {
+                    int qq = pidx / (D / 8);
+                    int d0 = (pidx % (D / 8)) * 8;
+                    *(bf16x8*)&Qlds[PI23(qq) * SQ + d0] =
+                        *(const bf16x8*)(q + (int64_t)(s0 + qs + qq) * q_ts + q_hoff + d0);
+                    *(bf16x8*)&dOl[PI23(qq) * SQ + d0] =
+                        *(const bf16x8*)(dout + (int64_t)(s0 + qs + qq) * do_ts + do_hoff + d0);
+                }
+            } else {
+                for (int pidx = threadIdx.x; pidx < pieces; pidx += 512) {
+                    int qq = pidx / (DPAD / 8);
+                    int d0 = (pidx % (DPAD / 8)) * 8;
+                    bool valid = (qs + qq) < L;
+                    const __bf16* qp = q + (int64_t)(s0 + (valid ? qs + qq : 0)) * q_ts + q_hoff + d0;
+                    *(bf16x8*)&Qlds[PI23(qq) * SQ + d0] = load_bf16x8_guard(qp, d0, D, valid);
+                    const __bf16* dp = dout + (int64_t)(s0 + (valid ? qs + qq : 0)) * do_ts + do_hoff + d0;
+                    *(bf16x8*)&dOl[PI23(qq) * SQ + d0] = load_bf16x8_guard(dp, d0, D, valid);
+                }
             }
         }
         __syncthreads();
@@ -907,6 +956,16 @@ __global__ void __launch_bounds__(512, 4) fa_bwd_dq_kernel(
 #pragma unroll
     for (int dc = 0; dc < DCH; ++dc) dq[dc] = {0.f, 0.f, 0.f, 0.f};
 
+    if (D < DPAD) {  // pad columns zeroed once (see dkv note)
+        for (int pidx = threadIdx.x; pidx < 64 * (DPAD - D) / 8; pidx += 512) {
+            int key = pidx / ((DPAD - D) / 8);
+            int d0 = D + (pidx % ((DPAD - D) / 8)) * 8;
+            bf16x8 z = {};
+            *(bf16x8*)&Klds[PI23(key) * SQ + d0] = z;
+            *(bf16x8*)&Vlds[key * SQ + d0] = z;
+        }
+    }
+
     const int kend_total = min(L, qs + 128);
     const int nkt = (kend_total + 63) / 64;
 
@@ -919,14 +978,25 @@ __global__ void __launch_bounds__(512, 4) fa_bwd_dq_kernel(
         // stage row-major K (PI23 rows) + V images, cooperative
         {
             const int pieces = 64 * DPAD / 8;
-            for (int pidx = threadIdx.x; pidx < pieces; pidx += 512) {
-                int key = pidx / (DPAD / 8);
-                int d0 = (pidx % (DPAD / 8)) * 8;
-                bool valid = (ks + key) < kend_total;
-                const __bf16* kp = k + (int64_t)(s0 + (valid ? ks + key : 0)) * k_ts + (int64_t)kvh * k_hs + d0;
-                *(bf16x8*)&Klds[PI23(key) * SQ + d0] = load_bf16x8_guard(kp, d0, D, valid);
-                const __bf16* vp = v + (int64_t)(s0 + (valid ? ks + key : 0)) * v_ts + (int64_t)kvh * v_hs + d0;
-                *(bf16x8*)&Vlds[key * SQ + d0] = load_bf16x8_guard(vp, d0, D, valid);
+            if (ks + 64 <= kend_total) {
+                for (int pidx = threadIdx.x; pidx < 64 * D / 8; pidx += 512) {
+                    int key = pidx / (D / 8);
+                    int d0 = (pidx % (D / 8)) * 8;
+                    *(bf16x8*)&Klds[PI23(key) * SQ + d0] =
+                        *(const bf16x8*)(k + (int64_t)(s0 + ks + key) * k_ts + (int64_t)kvh * k_hs + d0);
+                    *(bf16x8*)&Vlds[key * SQ + d0] =
+                        *(const bf16x8*)(v + (int64_t)(s0 + ks + key) * v_ts + (int64_t)kvh * v_hs + d0);
+                }
+            } else {
+                for (int pidx = threadIdx.x; pidx < pieces; pidx += 512) {
+                    int key = pidx / (DPAD / 8);
+                    int d0 = (pidx % (DPAD / 8)) * 8;
+                    bool valid = (ks + key) < kend_total;
+                    const __bf16* kp = k + (int64_t)(s0 + (valid ? ks + key : 0)) * k_ts + (int64_t)kvh * k_hs + d0;
+                    *(bf16x8*)&Klds[PI23(key) * SQ + d0] = load_bf16x8_guard(kp, d0, D, valid);
+                    const __bf16* vp = v + (int64_t)(s0 + (valid ? ks + key : 0)) * v_ts + (int64_t)kvh * v_hs + d0;
+                    *(bf16x8*)&Vlds[key * SQ + d0] = load_bf16x8_guard(vp, d0, D, valid);
+                }
             }
         }
         __syncthreads();
