@@ -168,15 +168,29 @@ __device__ __forceinline__ bf16x8 load_bf16x8_guard(const __bf16* p, int d0, int
 }
 
 // 16-lane row reduction (the 16 columns of one C-fragment row live in the
-// 16 lanes of one quarter-wave; xor over bits 0..3 stays inside it).
+// 16 lanes of one quarter-wave). DPP row_ror rotations instead of
+// __shfl_xor: hipcc lowers the xor shuffles to ds_bpermute — 32 LDS
+// instructions (plus their lgkm waits) per fwd loop iteration for an
+// issue-bound kernel. row_ror by 8/4/2/1 inside the 16-lane DPP row is
+// pure VALU and reduces the same 16-lane set.
+template <int N>
+__device__ __forceinline__ float dpp_row_ror(float v) {
+    return __int_as_float(__builtin_amdgcn_update_dpp(
+        0, __float_as_int(v), 0x120 + N, 0xF, 0xF, false));
+}
+
 __device__ __forceinline__ float qwave_reduce_max(float v) {
-#pragma unroll
-    for (int off = 8; off >= 1; off >>= 1) v = fmaxf(v, __shfl_xor(v, off, 64));
+    v = fmaxf(v, dpp_row_ror<8>(v));
+    v = fmaxf(v, dpp_row_ror<4>(v));
+    v = fmaxf(v, dpp_row_ror<2>(v));
+    v = fmaxf(v, dpp_row_ror<1>(v));
     return v;
 }
 __device__ __forceinline__ float qwave_reduce_sum(float v) {
-#pragma unroll
-    for (int off = 8; off >= 1; off >>= 1) v += __shfl_xor(v, off, 64);
+    v += dpp_row_ror<8>(v);
+    v += dpp_row_ror<4>(v);
+    v += dpp_row_ror<2>(v);
+    v += dpp_row_ror<1>(v);
     return v;
 }
 
