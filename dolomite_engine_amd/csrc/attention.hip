@@ -373,30 +373,36 @@ __global__ void __launch_bounds__(512, 4) fa_fwd_kernel(
         // chain entirely (the kernel is issue-bound); the -INF guards for
         // the FIRST tile's running max stay in both branches. ---
         const float scale2 = scale * DOL_LOG2E;  // exp2-domain scores
-        // (a wave-uniform full-tile/masked split was tried here like the
-        // backward kernels' — it pushed fwd from 96 to 128 VGPRs with loop
-        // spills, so fwd keeps the single masked body)
+        // One body, two cheap reductions of per-element work:
+        //  - the mask-compare chain runs only on edge tiles (wave-uniform
+        //    branch; a full dual-body split cost fwd 32 VGPRs + spills);
+        //  - the -INFINITY select guards are gone: v_exp_f32(-inf) is
+        //    exactly +0, and mnew is never -inf after the fully-masked-row
+        //    force, so exp2f does the zeroing for free.
+        const bool edge_tile =
+            !((ks + 63 <= qs + wave * 16) && (ks + 64 <= kend) && (qs + wave * 16 + 16 <= L));
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
-            const int qpos = qs + wave * 16 + lg * 4 + r;
             float pv[4];
-            float rowmax = -INFINITY;
 #pragma unroll
-            for (int cb = 0; cb < 4; ++cb) {
-                const int kpos = ks + cb * 16 + lr;
-                float sv = sc[cb][r] * scale2;
-                if (kpos > qpos || kpos >= kend || qpos >= L) sv = -INFINITY;
-                pv[cb] = sv;
-                rowmax = fmaxf(rowmax, sv);
+            for (int cb = 0; cb < 4; ++cb) pv[cb] = sc[cb][r] * scale2;
+            if (__builtin_amdgcn_readfirstlane(edge_tile ? 1 : 0)) {
+                const int qpos = qs + wave * 16 + lg * 4 + r;
+#pragma unroll
+                for (int cb = 0; cb < 4; ++cb) {
+                    const int kpos = ks + cb * 16 + lr;
+                    if (kpos > qpos || kpos >= kend || qpos >= L) pv[cb] = -INFINITY;
+                }
             }
+            float rowmax = fmaxf(fmaxf(pv[0], pv[1]), fmaxf(pv[2], pv[3]));
             rowmax = qwave_reduce_max(rowmax);
             float mnew = fmaxf(m_run[r], rowmax);
             if (mnew == -INFINITY) mnew = 0.f;  // fully-masked row guard
-            float alpha = (m_run[r] == -INFINITY) ? 0.f : exp2f(m_run[r] - mnew);
+            float alpha = exp2f(m_run[r] - mnew);
             float rsum = 0.f;
 #pragma unroll
             for (int cb = 0; cb < 4; ++cb) {
-                float e = (pv[cb] == -INFINITY) ? 0.f : exp2f(pv[cb] - mnew);
+                float e = exp2f(pv[cb] - mnew);
                 Pw[(lg * 4 + r) * SV + cb * 16 + lr] = (__bf16)e;
                 rsum += e;
             }
