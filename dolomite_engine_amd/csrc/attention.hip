@@ -323,6 +323,21 @@ __global__ void __launch_bounds__(512, 4) fa_fwd_kernel(
         }
     }
 
+    // tile-invariant staging piece coordinates (see dkv note)
+    int st_koff[2], st_voff[2];
+    int st_klds[2], st_vlds[2];
+#pragma unroll
+    for (int j = 0; j < 2; ++j) {
+        int pidx = (int)threadIdx.x + j * 512;
+        int pp = pidx < 64 * D / 8 ? pidx : 0;
+        int key = pp / (D / 8);
+        int d0 = (pp % (D / 8)) * 8;
+        st_koff[j] = (int)(key * k_ts + kvh * k_hs) + d0;   // fits 32 bits
+        st_voff[j] = (int)(key * v_ts + kvh * v_hs) + d0;
+        st_klds[j] = key * SK + d0;
+        st_vlds[j] = PI23(key) * SK + d0;
+    }
+
     // T5 static priority: the later-dispatched half of an 8-wave workgroup
     // loses VALU arbitration; one setprio for it, no per-cluster flips.
     if (__builtin_amdgcn_readfirstlane(threadIdx.x) >= 256) __builtin_amdgcn_s_setprio(1);
@@ -332,13 +347,13 @@ __global__ void __launch_bounds__(512, 4) fa_fwd_kernel(
     // direct cooperative staging (prologue and the NBUF==1 path)
     auto stage_direct = [&](int ks_, __bf16* Kw, __bf16* Vw) {
         if (ks_ + 64 <= kend) {
-            for (int pidx = threadIdx.x; pidx < 64 * D / 8; pidx += 512) {
-                int key = pidx / (D / 8);
-                int d0 = (pidx % (D / 8)) * 8;
-                *(bf16x8*)&Kw[key * SK + d0] =
-                    *(const bf16x8*)(k + (int64_t)(s0 + ks_ + key) * k_ts + (int64_t)kvh * k_hs + d0);
-                *(bf16x8*)&Vw[PI23(key) * SK + d0] =
-                    *(const bf16x8*)(v + (int64_t)(s0 + ks_ + key) * v_ts + (int64_t)kvh * v_hs + d0);
+            const int64_t disp = (int64_t)(s0 + ks_);
+#pragma unroll
+            for (int j = 0; j < 2; ++j) {
+                if ((int)threadIdx.x + j * 512 < 64 * D / 8) {
+                    *(bf16x8*)&Kw[st_klds[j]] = *(const bf16x8*)(k + disp * k_ts + st_koff[j]);
+                    *(bf16x8*)&Vw[st_vlds[j]] = *(const bf16x8*)(v + disp * v_ts + st_voff[j]);
+                }
             }
         } else {
             for (int pidx = threadIdx.x; pidx < PIECES; pidx += 512) {
@@ -754,15 +769,15 @@ __global__ void __launch_bounds__(512, DPAD <= 96 ? 4 : 2) fa_bwd_dkv_kernel(
         }
     }
 
+    // (hoisting these piece coordinates like dq/fwd spills at dkv's
+    // 128-VGPR cap — dkv keeps the in-loop computation)
     for (int qt = qt0; qt < nqt; ++qt) {
         const int qs = qt * 64;
         __syncthreads();  // previous iteration's image reads done
         {
             const int pieces = 64 * DPAD / 8;
             if (qs + 64 <= L) {
-                // full interior tile (every tile when seqlens divide 64):
-                // guard-free staging over the real D columns only (pads
-                // pre-zeroed above; D % 8 == 0 on this path)
+                // full interior tile: guard-free staging of the real D cols
                 for (int pidx = threadIdx.x; pidx < 64 * D / 8; pidx += 512) {
                     int qq = pidx / (D / 8);
                     int d0 = (pidx % (D / 8)) * 8;
@@ -986,6 +1001,21 @@ __global__ void __launch_bounds__(512, 4) fa_bwd_dq_kernel(
         }
     }
 
+    // tile-invariant staging piece coordinates (see dkv note)
+    int st_koff[2], st_voff[2];
+    int st_klds[2], st_vlds[2];
+#pragma unroll
+    for (int j = 0; j < 2; ++j) {
+        int pidx = (int)threadIdx.x + j * 512;
+        int pp = pidx < 64 * D / 8 ? pidx : 0;
+        int key = pp / (D / 8);
+        int d0 = (pp % (D / 8)) * 8;
+        st_koff[j] = (int)(key * k_ts + kvh * k_hs) + d0;   // fits 32 bits
+        st_voff[j] = (int)(key * v_ts + kvh * v_hs) + d0;
+        st_klds[j] = PI23(key) * SQ + d0;
+        st_vlds[j] = key * SQ + d0;
+    }
+
     const int kend_total = min(L, qs + 128);
     const int nkt = (kend_total + 63) / 64;
 
@@ -999,13 +1029,13 @@ __global__ void __launch_bounds__(512, 4) fa_bwd_dq_kernel(
         {
             const int pieces = 64 * DPAD / 8;
             if (ks + 64 <= kend_total) {
-                for (int pidx = threadIdx.x; pidx < 64 * D / 8; pidx += 512) {
-                    int key = pidx / (D / 8);
-                    int d0 = (pidx % (D / 8)) * 8;
-                    *(bf16x8*)&Klds[PI23(key) * SQ + d0] =
-                        *(const bf16x8*)(k + (int64_t)(s0 + ks + key) * k_ts + (int64_t)kvh * k_hs + d0);
-                    *(bf16x8*)&Vlds[key * SQ + d0] =
-                        *(const bf16x8*)(v + (int64_t)(s0 + ks + key) * v_ts + (int64_t)kvh * v_hs + d0);
+                const int64_t disp = (int64_t)(s0 + ks);
+#pragma unroll
+                for (int j = 0; j < 2; ++j) {
+                    if ((int)threadIdx.x + j * 512 < 64 * D / 8) {
+                        *(bf16x8*)&Klds[st_klds[j]] = *(const bf16x8*)(k + disp * k_ts + st_koff[j]);
+                        *(bf16x8*)&Vlds[st_vlds[j]] = *(const bf16x8*)(v + disp * v_ts + st_voff[j]);
+                    }
                 }
             } else {
                 for (int pidx = threadIdx.x; pidx < pieces; pidx += 512) {
