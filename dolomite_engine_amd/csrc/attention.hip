@@ -732,16 +732,18 @@ __global__ void __launch_bounds__(512, DPAD <= 96 ? 4 : 2) fa_bwd_dkv_kernel(
     const int tr_jj = (lane >> 2) & 3;
     const int tr_g8 = (lane >> 4) * 8;
     const int tr_cc = (lane & 3) * 4;
+    // Two lane bases per image PAIR: dOl sits 64*SQ elements above Qlds and
+    // PTl 64*STS above dSTl — both displacements fit the 16-bit ds_read
+    // immediate, so the second image of each pair rides the offset field
+    // instead of holding two more address registers (dkv is register-bound).
     const unsigned aQ0 = (unsigned)(size_t)(Qlds + PI23(tr_g8 + tr_jj) * SQ + tr_cc);
     const unsigned aQ1 = (unsigned)(size_t)(Qlds + PI23(tr_g8 + 4 + tr_jj) * SQ + tr_cc);
-    const unsigned aO0 = (unsigned)(size_t)(dOl + PI23(tr_g8 + tr_jj) * SQ + tr_cc);
-    const unsigned aO1 = (unsigned)(size_t)(dOl + PI23(tr_g8 + 4 + tr_jj) * SQ + tr_cc);
-    // strip A-fragment bases: the wave's 16-key column block is runtime, so
-    // it lives in the base; the q-chunk (kc2*32 rows) is the immediate
-    const unsigned aPT0 = (unsigned)(size_t)(PTl + PI23(tr_g8 + tr_jj) * STS + wave * 16 + tr_cc);
-    const unsigned aPT1 = (unsigned)(size_t)(PTl + PI23(tr_g8 + 4 + tr_jj) * STS + wave * 16 + tr_cc);
     const unsigned aDS0 = (unsigned)(size_t)(dSTl + PI23(tr_g8 + tr_jj) * STS + wave * 16 + tr_cc);
     const unsigned aDS1 = (unsigned)(size_t)(dSTl + PI23(tr_g8 + 4 + tr_jj) * STS + wave * 16 + tr_cc);
+    constexpr int DO_IMM = 64 * SQ * 2;        // dOl = Qlds + this (bytes)
+    constexpr int PT_IMM = 64 * STS * 2;       // PTl = dSTl + this (bytes)
+    static_assert(DO_IMM + 32 * SQ * 2 + 16 * 2 < 65536, "ds offset range");
+    static_assert(PT_IMM + 32 * STS * 2 + 16 * 2 < 65536, "ds offset range");
 
     // T5 static priority (guide §5.5): the later-dispatched half of an
     // 8-wave workgroup loses VALU arbitration to the older half; one
@@ -769,22 +771,35 @@ __global__ void __launch_bounds__(512, DPAD <= 96 ? 4 : 2) fa_bwd_dkv_kernel(
         }
     }
 
-    // (hoisting these piece coordinates like dq/fwd spills at dkv's
-    // 128-VGPR cap — dkv keeps the in-loop computation)
+    // staging piece coordinates hoisted like dq/fwd (the freed ladder
+    // address registers above pay for them at the 128-VGPR cap)
+    int st_qoff[2], st_dooff[2], st_lds[2];
+#pragma unroll
+    for (int j = 0; j < 2; ++j) {
+        int pidx = (int)threadIdx.x + j * 512;
+        int pp = pidx < 64 * D / 8 ? pidx : 0;
+        int qq = pp / (D / 8);
+        int d0 = (pp % (D / 8)) * 8;
+        st_qoff[j] = (int)(qq * q_ts + q_hoff) + d0;   // fits 32 bits
+        st_dooff[j] = (int)(qq * do_ts + do_hoff) + d0;
+        st_lds[j] = PI23(qq) * SQ + d0;
+    }
+
     for (int qt = qt0; qt < nqt; ++qt) {
         const int qs = qt * 64;
         __syncthreads();  // previous iteration's image reads done
         {
             const int pieces = 64 * DPAD / 8;
             if (qs + 64 <= L) {
-                // full interior tile: guard-free staging of the real D cols
-                for (int pidx = threadIdx.x; pidx < 64 * D / 8; pidx += 512) {
-                    int qq = pidx / (D / 8);
-                    int d0 = (pidx % (D / 8)) * 8;
-                    *(bf16x8*)&Qlds[PI23(qq) * SQ + d0] =
-                        *(const bf16x8*)(q + (int64_t)(s0 + qs + qq) * q_ts + q_hoff + d0);
-                    *(bf16x8*)&dOl[PI23(qq) * SQ + d0] =
-                        *(const bf16x8*)(dout + (int64_t)(s0 + qs + qq) * do_ts + do_hoff + d0);
+                // full interior tile: precomputed piece offsets + scalar
+                // displacement (see prologue note)
+                const int64_t disp = (int64_t)(s0 + qs);
+#pragma unroll
+                for (int j = 0; j < 2; ++j) {
+                    if ((int)threadIdx.x + j * 512 < 64 * D / 8) {
+                        *(bf16x8*)&Qlds[st_lds[j]] = *(const bf16x8*)(q + disp * q_ts + st_qoff[j]);
+                        *(bf16x8*)&dOl[st_lds[j]] = *(const bf16x8*)(dout + disp * do_ts + st_dooff[j]);
+                    }
                 }
             } else {
                 for (int pidx = threadIdx.x; pidx < pieces; pidx += 512) {
@@ -855,12 +870,12 @@ __global__ void __launch_bounds__(512, DPAD <= 96 ? 4 : 2) fa_bwd_dkv_kernel(
             asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");  // exact counts from here
             __builtin_amdgcn_sched_barrier(0);
             bf16x4 pt_lo[2], pt_hi[2], ds_lo[2], ds_hi[2];
-            tr16_issue<0>(aPT0, aPT1, pt_lo[0], pt_hi[0]);
+            tr16_issue<PT_IMM>(aDS0, aDS1, pt_lo[0], pt_hi[0]);
             tr16_issue<0>(aDS0, aDS1, ds_lo[0], ds_hi[0]);
-            tr16_issue<32 * STS * 2>(aPT0, aPT1, pt_lo[1], pt_hi[1]);
+            tr16_issue<PT_IMM + 32 * STS * 2>(aDS0, aDS1, pt_lo[1], pt_hi[1]);
             tr16_issue<32 * STS * 2>(aDS0, aDS1, ds_lo[1], ds_hi[1]);
             bf16x4 olo[2], ohi[2], qlo[2], qhi[2];
-            tr16_issue<0>(aO0, aO1, olo[0], ohi[0]);
+            tr16_issue<DO_IMM>(aQ0, aQ1, olo[0], ohi[0]);
             tr16_issue<0>(aQ0, aQ1, qlo[0], qhi[0]);
             lgkm_wait8<4>(pt_lo[0], pt_hi[0], ds_lo[0], ds_hi[0], pt_lo[1], pt_hi[1], ds_lo[1], ds_hi[1]);
             bf16x8 ptf0 = tr16_join8(pt_lo[0], pt_hi[0]);
@@ -873,7 +888,7 @@ __global__ void __launch_bounds__(512, DPAD <= 96 ? 4 : 2) fa_bwd_dkv_kernel(
         if constexpr ((i) + 1 < 2 * DCH) {                                                              \
             constexpr int kn_ = ((i) + 1) / DCH, dn_ = ((i) + 1) % DCH;                                 \
             constexpr int off_ = (kn_ * 32 * SQ + dn_ * 16) * 2;                                        \
-            tr16_issue<off_>(aO0, aO1, olo[((i) + 1) & 1], ohi[((i) + 1) & 1]);                         \
+            tr16_issue<off_ + DO_IMM>(aQ0, aQ1, olo[((i) + 1) & 1], ohi[((i) + 1) & 1]);                \
             tr16_issue<off_>(aQ0, aQ1, qlo[((i) + 1) & 1], qhi[((i) + 1) & 1]);                         \
             lgkm_wait4<4>(olo[(i) & 1], ohi[(i) & 1], qlo[(i) & 1], qhi[(i) & 1]);                      \
         } else {                                                                                        \
