@@ -771,35 +771,23 @@ __global__ void __launch_bounds__(512, DPAD <= 96 ? 4 : 2) fa_bwd_dkv_kernel(
         }
     }
 
-    // staging piece coordinates hoisted like dq/fwd (the freed ladder
-    // address registers above pay for them at the 128-VGPR cap)
-    int st_qoff[2], st_dooff[2], st_lds[2];
-#pragma unroll
-    for (int j = 0; j < 2; ++j) {
-        int pidx = (int)threadIdx.x + j * 512;
-        int pp = pidx < 64 * D / 8 ? pidx : 0;
-        int qq = pp / (D / 8);
-        int d0 = (pp % (D / 8)) * 8;
-        st_qoff[j] = (int)(qq * q_ts + q_hoff) + d0;   // fits 32 bits
-        st_dooff[j] = (int)(qq * do_ts + do_hoff) + d0;
-        st_lds[j] = PI23(qq) * SQ + d0;
-    }
-
+    // (hoisting the staging coordinates like dq/fwd costs 3 scratch
+    // pointer reloads per iteration at dkv's 128-VGPR cap and measured
+    // slower than recomputing the addresses — dkv keeps in-loop compute)
     for (int qt = qt0; qt < nqt; ++qt) {
         const int qs = qt * 64;
         __syncthreads();  // previous iteration's image reads done
         {
             const int pieces = 64 * DPAD / 8;
             if (qs + 64 <= L) {
-                // full interior tile: precomputed piece offsets + scalar
-                // displacement (see prologue note)
-                const int64_t disp = (int64_t)(s0 + qs);
-#pragma unroll
-                for (int j = 0; j < 2; ++j) {
-                    if ((int)threadIdx.x + j * 512 < 64 * D / 8) {
-                        *(bf16x8*)&Qlds[st_lds[j]] = *(const bf16x8*)(q + disp * q_ts + st_qoff[j]);
-                        *(bf16x8*)&dOl[st_lds[j]] = *(const bf16x8*)(dout + disp * do_ts + st_dooff[j]);
-                    }
+                // full interior tile: guard-free staging of the real D cols
+                for (int pidx = threadIdx.x; pidx < 64 * D / 8; pidx += 512) {
+                    int qq = pidx / (D / 8);
+                    int d0 = (pidx % (D / 8)) * 8;
+                    *(bf16x8*)&Qlds[PI23(qq) * SQ + d0] =
+                        *(const bf16x8*)(q + (int64_t)(s0 + qs + qq) * q_ts + q_hoff + d0);
+                    *(bf16x8*)&dOl[PI23(qq) * SQ + d0] =
+                        *(const bf16x8*)(dout + (int64_t)(s0 + qs + qq) * do_ts + do_hoff + d0);
                 }
             } else {
                 for (int pidx = threadIdx.x; pidx < pieces; pidx += 512) {
