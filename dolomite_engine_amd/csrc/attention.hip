@@ -436,19 +436,25 @@ __global__ void __launch_bounds__(512, 4) fa_fwd_kernel(
             bf16x8 pf1 = *(const bf16x8*)&Pw[lr * SV + 32 + lg * 8];
             lgkm_drain2x8(pf0, pf1);
             __builtin_amdgcn_sched_barrier(0);
-            bf16x4 vlo[2], vhi[2];
+            // depth-3 pipeline: two fragments stay in flight — ONE MFMA
+            // (~17 cyc) does not cover the ~50-cycle LDS read latency that
+            // depth-2 exposed at every counted wait
+            bf16x4 vlo[3], vhi[3];
             tr16_issue<0>(aV0c, aV1c, vlo[0], vhi[0]);
+            tr16_issue<((1 / DCH) * 32 * SK + (1 % DCH) * 16) * 2>(aV0c, aV1c, vlo[1], vhi[1]);
 #define FWD_PV_STEP(i)                                                                                  \
     if constexpr ((i) < 2 * DCH) {                                                                      \
         constexpr int kc2_ = (i) / DCH, dc_ = (i) % DCH;                                                \
-        if constexpr ((i) + 1 < 2 * DCH) {                                                              \
-            constexpr int kn_ = ((i) + 1) / DCH, dn_ = ((i) + 1) % DCH;                                 \
-            tr16_issue<(kn_ * 32 * SK + dn_ * 16) * 2>(aV0c, aV1c, vlo[((i) + 1) & 1], vhi[((i) + 1) & 1]); \
-            lgkm_wait2<2>(vlo[(i) & 1], vhi[(i) & 1]);                                                  \
+        if constexpr ((i) + 2 < 2 * DCH) {                                                              \
+            constexpr int kn_ = ((i) + 2) / DCH, dn_ = ((i) + 2) % DCH;                                 \
+            tr16_issue<(kn_ * 32 * SK + dn_ * 16) * 2>(aV0c, aV1c, vlo[((i) + 2) % 3], vhi[((i) + 2) % 3]); \
+            lgkm_wait2<4>(vlo[(i) % 3], vhi[(i) % 3]);                                                  \
+        } else if constexpr ((i) + 1 < 2 * DCH) {                                                       \
+            lgkm_wait2<2>(vlo[(i) % 3], vhi[(i) % 3]);                                                  \
         } else {                                                                                        \
-            lgkm_wait2<0>(vlo[(i) & 1], vhi[(i) & 1]);                                                  \
+            lgkm_wait2<0>(vlo[(i) % 3], vhi[(i) % 3]);                                                  \
         }                                                                                               \
-        o_acc[dc_] = MFMA16(kc2_ ? pf1 : pf0, tr16_join8(vlo[(i) & 1], vhi[(i) & 1]), o_acc[dc_]);      \
+        o_acc[dc_] = MFMA16(kc2_ ? pf1 : pf0, tr16_join8(vlo[(i) % 3], vhi[(i) % 3]), o_acc[dc_]);      \
     }
             FWD_PV_STEP(0) FWD_PV_STEP(1) FWD_PV_STEP(2) FWD_PV_STEP(3)
             FWD_PV_STEP(4) FWD_PV_STEP(5) FWD_PV_STEP(6) FWD_PV_STEP(7)
@@ -614,7 +620,17 @@ __global__ void __launch_bounds__(256) fa_bwd_preprocess_kernel(
     const T* op = o + t * o_ts + (int64_t)h * D;
     const T* dp = dout + t * do_ts + (int64_t)h * D;
     float acc = 0.f;
-    for (int d = lane; d < D; d += 64) acc += load_as_f32(op + d) * load_as_f32(dp + d);
+    // element pairs per lane: coalesced 4-byte loads instead of scalar b16
+    for (int d = lane * 2; d + 1 < D; d += 128) {
+        float o0[2], d0[2];
+        VecIO<T, 2>::load(op + d, o0);
+        VecIO<T, 2>::load(dp + d, d0);
+        acc += o0[0] * d0[0] + o0[1] * d0[1];
+    }
+    if (D & 1) {  // odd head dim tail (not on any named config)
+        int d = D - 1;
+        if (lane == 0) acc += load_as_f32(op + d) * load_as_f32(dp + d);
+    }
     acc = wave_reduce_sum(acc);
     if (lane == 0) delta[(int64_t)h * T_total + t] = acc;
 }
@@ -773,7 +789,10 @@ __global__ void __launch_bounds__(512, DPAD <= 96 ? 4 : 2) fa_bwd_dkv_kernel(
 
     // (hoisting the staging coordinates like dq/fwd costs 3 scratch
     // pointer reloads per iteration at dkv's 128-VGPR cap and measured
-    // slower than recomputing the addresses — dkv keeps in-loop compute)
+    // slower than recomputing addresses — dkv recomputes, with the
+    // runtime division replaced by an exact magic multiply: pidx < 2048
+    // and D/8 <= 16, so (pidx * (2^22/(D/8) + 1)) >> 22 is exact)
+    const unsigned st_magic = (1u << 22) / (unsigned)(D / 8) + 1;
     for (int qt = qt0; qt < nqt; ++qt) {
         const int qs = qt * 64;
         __syncthreads();  // previous iteration's image reads done
@@ -782,8 +801,8 @@ __global__ void __launch_bounds__(512, DPAD <= 96 ? 4 : 2) fa_bwd_dkv_kernel(
             if (qs + 64 <= L) {
                 // full interior tile: guard-free staging of the real D cols
                 for (int pidx = threadIdx.x; pidx < 64 * D / 8; pidx += 512) {
-                    int qq = pidx / (D / 8);
-                    int d0 = (pidx % (D / 8)) * 8;
+                    int qq = (int)(((unsigned)pidx * st_magic) >> 22);
+                    int d0 = (pidx - qq * (D / 8)) * 8;
                     *(bf16x8*)&Qlds[PI23(qq) * SQ + d0] =
                         *(const bf16x8*)(q + (int64_t)(s0 + qs + qq) * q_ts + q_hoff + d0);
                     *(bf16x8*)&dOl[PI23(qq) * SQ + d0] =
@@ -1096,19 +1115,22 @@ __global__ void __launch_bounds__(512, 4) fa_bwd_dq_kernel(
             bf16x8 dsf1 = *(const bf16x8*)&dSw[lr * ST + 32 + lg * 8];
             lgkm_drain2x8(dsf0, dsf1);
             __builtin_amdgcn_sched_barrier(0);
-            bf16x4 klo[2], khi[2];
+            bf16x4 klo[3], khi[3];
             tr16_issue<0>(aK0, aK1, klo[0], khi[0]);
+            tr16_issue<((1 / DCH) * 32 * SQ + (1 % DCH) * 16) * 2>(aK0, aK1, klo[1], khi[1]);
 #define DQ_STEP(i)                                                                                      \
     if constexpr ((i) < 2 * DCH) {                                                                      \
         constexpr int kc2_ = (i) / DCH, dc_ = (i) % DCH;                                                \
-        if constexpr ((i) + 1 < 2 * DCH) {                                                              \
-            constexpr int kn_ = ((i) + 1) / DCH, dn_ = ((i) + 1) % DCH;                                 \
-            tr16_issue<(kn_ * 32 * SQ + dn_ * 16) * 2>(aK0, aK1, klo[((i) + 1) & 1], khi[((i) + 1) & 1]); \
-            lgkm_wait2<2>(klo[(i) & 1], khi[(i) & 1]);                                                  \
+        if constexpr ((i) + 2 < 2 * DCH) {                                                              \
+            constexpr int kn_ = ((i) + 2) / DCH, dn_ = ((i) + 2) % DCH;                                 \
+            tr16_issue<(kn_ * 32 * SQ + dn_ * 16) * 2>(aK0, aK1, klo[((i) + 2) % 3], khi[((i) + 2) % 3]); \
+            lgkm_wait2<4>(klo[(i) % 3], khi[(i) % 3]);                                                  \
+        } else if constexpr ((i) + 1 < 2 * DCH) {                                                       \
+            lgkm_wait2<2>(klo[(i) % 3], khi[(i) % 3]);                                                  \
         } else {                                                                                        \
-            lgkm_wait2<0>(klo[(i) & 1], khi[(i) & 1]);                                                  \
+            lgkm_wait2<0>(klo[(i) % 3], khi[(i) % 3]);                                                  \
         }                                                                                               \
-        dq[dc_] = MFMA16(kc2_ ? dsf1 : dsf0, tr16_join8(klo[(i) & 1], khi[(i) & 1]), dq[dc_]);          \
+        dq[dc_] = MFMA16(kc2_ ? dsf1 : dsf0, tr16_join8(klo[(i) % 3], khi[(i) % 3]), dq[dc_]);          \
     }
             DQ_STEP(0) DQ_STEP(1) DQ_STEP(2) DQ_STEP(3)
             DQ_STEP(4) DQ_STEP(5) DQ_STEP(6) DQ_STEP(7)

@@ -54,9 +54,21 @@ __device__ __forceinline__ void store_from_f32<uint16_t>(uint16_t* p, float v) {
 
 // ---- wave / block reductions ---------------------------------------------
 
+// 16-lane-internal steps as DPP row_ror (pure VALU; __shfl_xor lowers to
+// ds_bpermute LDS instructions), cross-row steps as shuffles.
+template <int N>
+__device__ __forceinline__ float dpp_ror_f32(float v) {
+    return __int_as_float(__builtin_amdgcn_update_dpp(
+        0, __float_as_int(v), 0x120 + N, 0xF, 0xF, false));
+}
+
 __device__ __forceinline__ float wave_reduce_sum(float v) {
-#pragma unroll
-    for (int off = 32; off >= 1; off >>= 1) v += __shfl_xor(v, off, 64);
+    v += dpp_ror_f32<8>(v);
+    v += dpp_ror_f32<4>(v);
+    v += dpp_ror_f32<2>(v);
+    v += dpp_ror_f32<1>(v);
+    v += __shfl_xor(v, 16, 64);
+    v += __shfl_xor(v, 32, 64);
     return v;
 }
 
@@ -119,6 +131,31 @@ struct VecIO<float, 4> {
 #pragma unroll
         for (int e = 0; e < 4; ++e) v[e] = in[e];
         *(f4_t*)p = v;
+    }
+};
+
+template <>
+struct VecIO<uint16_t, 2> {
+    static __device__ __forceinline__ void load(const uint16_t* p, float* out) {
+        uint32_t v = *(const uint32_t*)p;
+        out[0] = bf16_to_f32((uint16_t)(v & 0xffff));
+        out[1] = bf16_to_f32((uint16_t)(v >> 16));
+    }
+    static __device__ __forceinline__ void store(uint16_t* p, const float* in) {
+        uint32_t v = (uint32_t)f32_to_bf16(in[0]) | ((uint32_t)f32_to_bf16(in[1]) << 16);
+        *(uint32_t*)p = v;
+    }
+};
+
+template <>
+struct VecIO<float, 2> {
+    static __device__ __forceinline__ void load(const float* p, float* out) {
+        out[0] = p[0];
+        out[1] = p[1];
+    }
+    static __device__ __forceinline__ void store(float* p, const float* in) {
+        p[0] = in[0];
+        p[1] = in[1];
     }
 };
 

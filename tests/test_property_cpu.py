@@ -66,7 +66,10 @@ def test_build_sample_idx_property(doc_lens, seq_length, num_epochs):
     """Native C builder vs the pure-python helpers.cpp restatement for
     arbitrary document length mixes."""
     from dolomite_engine_amd.megatron import build_sample_idx
-    from tests.test_megatron_data import _sample_idx_python
+    try:
+        from tests.test_megatron_data import _sample_idx_python
+    except ModuleNotFoundError:  # subset invocations without the package path
+        from test_megatron_data import _sample_idx_python
 
     sizes = numpy.array(doc_lens, dtype=numpy.int32)
     doc_idx = numpy.tile(numpy.arange(len(sizes), dtype=numpy.int32), num_epochs)
