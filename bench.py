@@ -237,8 +237,9 @@ def cpu_baseline(args):
 
     L = cfg.n_layer
     t_attn1(256)  # warm up the autograd/GEMM paths before timing
-    ta_512 = min(t_attn1(512), t_attn1(512))
     ta_S = t_attn1(args.seq_len)
+    # measure the 512 leg fully warm (thread-pool spin-up was inflating it)
+    ta_512 = min(t_attn1(512), t_attn1(512))
 
     # AdamW on a 1/8 sample of the params, scaled up
     n_sample = 0

@@ -173,24 +173,18 @@ __device__ __forceinline__ bf16x8 load_bf16x8_guard(const __bf16* p, int d0, int
 // instructions (plus their lgkm waits) per fwd loop iteration for an
 // issue-bound kernel. row_ror by 8/4/2/1 inside the 16-lane DPP row is
 // pure VALU and reduces the same 16-lane set.
-template <int N>
-__device__ __forceinline__ float dpp_row_ror(float v) {
-    return __int_as_float(__builtin_amdgcn_update_dpp(
-        0, __float_as_int(v), 0x120 + N, 0xF, 0xF, false));
-}
-
 __device__ __forceinline__ float qwave_reduce_max(float v) {
-    v = fmaxf(v, dpp_row_ror<8>(v));
-    v = fmaxf(v, dpp_row_ror<4>(v));
-    v = fmaxf(v, dpp_row_ror<2>(v));
-    v = fmaxf(v, dpp_row_ror<1>(v));
+    v = fmaxf(v, dpp_ror_f32<8>(v));
+    v = fmaxf(v, dpp_ror_f32<4>(v));
+    v = fmaxf(v, dpp_ror_f32<2>(v));
+    v = fmaxf(v, dpp_ror_f32<1>(v));
     return v;
 }
 __device__ __forceinline__ float qwave_reduce_sum(float v) {
-    v += dpp_row_ror<8>(v);
-    v += dpp_row_ror<4>(v);
-    v += dpp_row_ror<2>(v);
-    v += dpp_row_ror<1>(v);
+    v += dpp_ror_f32<8>(v);
+    v += dpp_ror_f32<4>(v);
+    v += dpp_ror_f32<2>(v);
+    v += dpp_ror_f32<1>(v);
     return v;
 }
 
