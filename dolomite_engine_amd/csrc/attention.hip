@@ -864,45 +864,59 @@ __global__ void __launch_bounds__(512, DPAD <= 96 ? 4 : 2) fa_bwd_dkv_kernel(
             *(bf16x4*)&dSTl[srow] = dsv;
         }
 
-        // dV += P^T*dO ; dK += dS^T*Q (contraction over q) — pipelined tr16
-        // ladder. The strip A-fragments come from the transposed strips via
-        // tr16 too: issue all 8 strip reads + pair0, one counted wait.
+        // dV += P^T*dO then dK += dS^T*Q (contraction over q) — TWO
+        // independent depth-3 tr16 ladders: dO feeds only dV and Q only dK,
+        // so splitting halves the live A-fragment registers (one strip pair
+        // instead of two) and the freed budget funds a third in-flight
+        // fragment — one MFMA per wait no longer exposes LDS latency.
         {
             asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");  // exact counts from here
             __builtin_amdgcn_sched_barrier(0);
-            bf16x4 pt_lo[2], pt_hi[2], ds_lo[2], ds_hi[2];
-            tr16_issue<PT_IMM>(aDS0, aDS1, pt_lo[0], pt_hi[0]);
-            tr16_issue<0>(aDS0, aDS1, ds_lo[0], ds_hi[0]);
-            tr16_issue<PT_IMM + 32 * STS * 2>(aDS0, aDS1, pt_lo[1], pt_hi[1]);
-            tr16_issue<32 * STS * 2>(aDS0, aDS1, ds_lo[1], ds_hi[1]);
-            bf16x4 olo[2], ohi[2], qlo[2], qhi[2];
-            tr16_issue<DO_IMM>(aQ0, aQ1, olo[0], ohi[0]);
-            tr16_issue<0>(aQ0, aQ1, qlo[0], qhi[0]);
-            lgkm_wait8<4>(pt_lo[0], pt_hi[0], ds_lo[0], ds_hi[0], pt_lo[1], pt_hi[1], ds_lo[1], ds_hi[1]);
-            bf16x8 ptf0 = tr16_join8(pt_lo[0], pt_hi[0]);
-            bf16x8 dstf0 = tr16_join8(ds_lo[0], ds_hi[0]);
-            bf16x8 ptf1 = tr16_join8(pt_lo[1], pt_hi[1]);
-            bf16x8 dstf1 = tr16_join8(ds_lo[1], ds_hi[1]);
-#define DKV_STEP(i)                                                                                     \
+            bf16x4 a_lo[2], a_hi[2];
+            tr16_issue<PT_IMM>(aDS0, aDS1, a_lo[0], a_hi[0]);
+            tr16_issue<PT_IMM + 32 * STS * 2>(aDS0, aDS1, a_lo[1], a_hi[1]);
+            bf16x4 blo[3], bhi[3];
+            tr16_issue<DO_IMM>(aQ0, aQ1, blo[0], bhi[0]);
+            tr16_issue<DO_IMM + ((1 / DCH) * 32 * SQ + (1 % DCH) * 16) * 2>(aQ0, aQ1, blo[1], bhi[1]);
+            lgkm_wait4<4>(a_lo[0], a_hi[0], a_lo[1], a_hi[1]);
+            bf16x8 af0 = tr16_join8(a_lo[0], a_hi[0]);
+            bf16x8 af1 = tr16_join8(a_lo[1], a_hi[1]);
+#define DKV_HALF(i, IMM, ACC, AF0, AF1)                                                                 \
     if constexpr ((i) < 2 * DCH) {                                                                      \
         constexpr int kc2_ = (i) / DCH, dc_ = (i) % DCH;                                                \
-        if constexpr ((i) + 1 < 2 * DCH) {                                                              \
-            constexpr int kn_ = ((i) + 1) / DCH, dn_ = ((i) + 1) % DCH;                                 \
-            constexpr int off_ = (kn_ * 32 * SQ + dn_ * 16) * 2;                                        \
-            tr16_issue<off_ + DO_IMM>(aQ0, aQ1, olo[((i) + 1) & 1], ohi[((i) + 1) & 1]);                \
-            tr16_issue<off_>(aQ0, aQ1, qlo[((i) + 1) & 1], qhi[((i) + 1) & 1]);                         \
-            lgkm_wait4<4>(olo[(i) & 1], ohi[(i) & 1], qlo[(i) & 1], qhi[(i) & 1]);                      \
+        if constexpr ((i) + 2 < 2 * DCH) {                                                              \
+            constexpr int kn_ = ((i) + 2) / DCH, dn_ = ((i) + 2) % DCH;                                 \
+            tr16_issue<(kn_ * 32 * SQ + dn_ * 16) * 2 + (IMM)>(aQ0, aQ1, blo[((i) + 2) % 3], bhi[((i) + 2) % 3]); \
+            lgkm_wait2<4>(blo[(i) % 3], bhi[(i) % 3]);                                                  \
+        } else if constexpr ((i) + 1 < 2 * DCH) {                                                       \
+            lgkm_wait2<2>(blo[(i) % 3], bhi[(i) % 3]);                                                  \
         } else {                                                                                        \
-            lgkm_wait4<0>(olo[(i) & 1], ohi[(i) & 1], qlo[(i) & 1], qhi[(i) & 1]);                      \
+            lgkm_wait2<0>(blo[(i) % 3], bhi[(i) % 3]);                                                  \
         }                                                                                               \
-        dvr[dc_] = MFMA16(kc2_ ? ptf1 : ptf0, tr16_join8(olo[(i) & 1], ohi[(i) & 1]), dvr[dc_]);        \
-        dkr[dc_] = MFMA16(kc2_ ? dstf1 : dstf0, tr16_join8(qlo[(i) & 1], qhi[(i) & 1]), dkr[dc_]);      \
+        ACC[dc_] = MFMA16(kc2_ ? (AF1) : (AF0), tr16_join8(blo[(i) % 3], bhi[(i) % 3]), ACC[dc_]);      \
     }
-            DKV_STEP(0) DKV_STEP(1) DKV_STEP(2) DKV_STEP(3)
-            DKV_STEP(4) DKV_STEP(5) DKV_STEP(6) DKV_STEP(7)
-            DKV_STEP(8) DKV_STEP(9) DKV_STEP(10) DKV_STEP(11)
-            DKV_STEP(12) DKV_STEP(13) DKV_STEP(14) DKV_STEP(15)
-#undef DKV_STEP
+#define DKV_A(i) DKV_HALF(i, DO_IMM, dvr, af0, af1)
+            DKV_A(0) DKV_A(1) DKV_A(2) DKV_A(3)
+            DKV_A(4) DKV_A(5) DKV_A(6) DKV_A(7)
+            DKV_A(8) DKV_A(9) DKV_A(10) DKV_A(11)
+            DKV_A(12) DKV_A(13) DKV_A(14) DKV_A(15)
+#undef DKV_A
+            // half B: dK += dS^T * Q. The strip A-fragments and the first
+            // two Q B-fragments are issued, then one counted wait.
+            tr16_issue<0>(aDS0, aDS1, a_lo[0], a_hi[0]);
+            tr16_issue<32 * STS * 2>(aDS0, aDS1, a_lo[1], a_hi[1]);
+            tr16_issue<0>(aQ0, aQ1, blo[0], bhi[0]);
+            tr16_issue<((1 / DCH) * 32 * SQ + (1 % DCH) * 16) * 2>(aQ0, aQ1, blo[1], bhi[1]);
+            lgkm_wait4<4>(a_lo[0], a_hi[0], a_lo[1], a_hi[1]);
+            af0 = tr16_join8(a_lo[0], a_hi[0]);
+            af1 = tr16_join8(a_lo[1], a_hi[1]);
+#define DKV_B(i) DKV_HALF(i, 0, dkr, af0, af1)
+            DKV_B(0) DKV_B(1) DKV_B(2) DKV_B(3)
+            DKV_B(4) DKV_B(5) DKV_B(6) DKV_B(7)
+            DKV_B(8) DKV_B(9) DKV_B(10) DKV_B(11)
+            DKV_B(12) DKV_B(13) DKV_B(14) DKV_B(15)
+#undef DKV_B
+#undef DKV_HALF
             __builtin_amdgcn_sched_barrier(0);
         }
     }
