@@ -12,12 +12,16 @@
 //     wave), k-tile 64 keys streamed through LDS as row-major [key][d]
 //     images (stride DPAD+16, conflict-free b128 B-frag reads); the PV
 //     B-fragment comes from the PI23-rowed V image via ds_read_b64_tr_b16.
-//   - online softmax in fp32 VGPRs, m/l carried per row, row reductions via
-//     16-lane __shfl_xor (the C-fragment's 16 columns of one row).
-//   - P routed through a per-wave LDS tile to re-shape C-layout -> A-layout.
-//   - LSE (H,T) fp32 written for backward; backward recomputes P, computes
-//     dk/dv exclusively per (kv-tile, kv-head) workgroup looping q-tiles and
-//     grouped q-heads, dq via fp32 device atomics then a finalize cast.
+//   - online softmax in fp32 VGPRs, m/l carried per row, row reductions as
+//     DPP row_ror rotations (the C-fragment's 16 columns of one row).
+//   - P routed through a per-wave LDS tile to re-shape C-layout -> A-layout;
+//     the PV B-fragment comes from the PI23 V image via depth-3 counted
+//     tr16 ladders.
+//   - LSE (H,T) fp32 written for backward; backward recomputes P. dkv
+//     writes per-q-head (T, H, D) fp32 partials (exclusive stores, no
+//     atomics) reduced in fixed order by the finalize kernel —
+//     deterministic gradients; dq accumulates dQ in registers and stores
+//     bf16 straight into the packed dqkv.
 //   - dropout unsupported (hot-path configs run attn_pdrop = 0).
 //
 // Fragment layout assumption for mfma_f32_16x16x32_bf16 (verified on
