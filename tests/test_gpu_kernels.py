@@ -169,6 +169,8 @@ def test_rope_packed_roundtrip_and_oracle(head_type, H, Hkv, D):
         ("mha", 4, 4, 64, [64, 64]),
         ("mqa", 4, 1, 16, [10, 5]),       # tiny-config shape (D=16 -> DPAD 32)
         ("mqa", 8, 1, 80, [4096]),        # one full-length 3B-shaped sequence
+        ("mqa", 8, 1, 80, [7, 0, 9]),     # empty sequence in the packed batch
+        ("gqa", 8, 2, 128, [65, 63]),     # edge tiles on both sides of 64
     ],
 )
 def test_varlen_attention_fwd_bwd_vs_oracle(head_type, H, Hkv, D, lens):
