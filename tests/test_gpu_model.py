@@ -451,3 +451,42 @@ def test_deferred_wgrad_matches_inline():
     assert abs(l_inline[1] - l_defer[1]) / abs(l_inline[1]) < 2e-3, (l_inline, l_defer)
     for k in s_inline:
         torch.testing.assert_close(s_defer[k], s_inline[k], rtol=2e-2, atol=2e-3, msg=lambda m: f"{k}: {m}")
+
+
+def test_dense_sdpa_vs_packed_on_hardware():
+    """Dense bf16 sdpa forward vs the packed padding-free HIP path on the
+    same weights/inputs, on hardware — the reference's own cross-impl
+    consistency check (gpt_dolomite_test.py) at its bf16 tolerance."""
+    from dolomite_engine_amd.hf_models import GPTDolomiteConfig, GPTDolomiteForCausalLM
+
+    kw = dict(
+        vocab_size=512, n_positions=256, n_embd=256, n_layer=2, n_head=4,
+        attention_head_type="gqa", num_key_value_heads=2, n_inner=512,
+        activation_function="swiglu", normalization_function="rmsnorm",
+        position_embedding_type="rope", resid_pdrop=0.0, embd_pdrop=0.0,
+        attn_pdrop=0.0, tie_word_embeddings=False,
+    )
+    torch.manual_seed(3)
+    cfg = GPTDolomiteConfig(**kw)
+    cfg._attn_implementation = "sdpa"
+    dense = GPTDolomiteForCausalLM(cfg).to(torch.bfloat16).cuda().eval()
+
+    cfg2 = GPTDolomiteConfig(**kw)
+    cfg2._attn_implementation = "flash_attention_2"
+    packed = GPTDolomiteForCausalLM(cfg2, use_padding_free_transformer=True).to(torch.bfloat16).cuda().eval()
+    packed.load_state_dict(dense.state_dict())
+
+    B, S = 2, 96
+    g = torch.Generator().manual_seed(77)
+    ids = torch.randint(0, 512, (B, S), generator=g).cuda()
+    with torch.no_grad():
+        out_d = dense(input_ids=ids).logits
+        out_p = packed(
+            input_ids=ids.reshape(-1),
+            position_ids=torch.arange(S).repeat(B).cuda(),
+            cu_seqlens=torch.arange(0, B * S + 1, S, dtype=torch.int32).cuda(),
+            max_seqlen=S,
+        ).logits
+    torch.testing.assert_close(
+        out_p.float().reshape(B, S, -1), out_d.float(), rtol=5e-3, atol=5e-3
+    )
