@@ -1,9 +1,10 @@
-"""Data pipeline (thin, round 1): synthetic Megatron-shaped pretraining
-batches ({"text": (micro_batch_size, sequence_length+1) int64}, reference
+"""Data pipeline: synthetic Megatron-shaped pretraining batches
+({"text": (micro_batch_size, sequence_length+1) int64}, reference
 data/megatron returns sequence_length+1 windows), and a list-packing
 finetune collator (data/utils.py:8-93, padding-free branch).
 
-The on-disk Megatron mmap .bin/.idx pipeline is a "next" row.
+The on-disk Megatron mmap .bin/.idx pipeline (blending, splits, FIM,
+native index builders) lives in megatron.py.
 """
 
 import torch
