@@ -50,6 +50,11 @@ class ModelWrapper(nn.Module):
         if model_name is None:
             cfg_kwargs = dict(pretrained_config)
             model_type = cfg_kwargs.pop("model_type", "gpt_dolomite")
+            if model_type not in _MODEL_CLASSES:
+                raise NotImplementedError(
+                    f"model family '{model_type}' is out of scope (DESIGN.md §6: "
+                    f"rnn/crosslayer/dense-moe families); built: {sorted(_MODEL_CLASSES)}"
+                )
             config_class, model_class = _MODEL_CLASSES[model_type]
             config = config_class(**cfg_kwargs)
             config._attn_implementation = attention_implementation
@@ -63,6 +68,11 @@ class ModelWrapper(nn.Module):
             from transformers import AutoConfig
 
             model_type = AutoConfig.from_pretrained(model_name).model_type
+            if model_type not in _MODEL_CLASSES:
+                raise NotImplementedError(
+                    f"model family '{model_type}' is out of scope (DESIGN.md §6); "
+                    f"built: {sorted(_MODEL_CLASSES)}"
+                )
             config_class, model_class = _MODEL_CLASSES[model_type]
             config = config_class.from_pretrained(model_name)
             config._attn_implementation = attention_implementation
