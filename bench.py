@@ -134,7 +134,6 @@ MICRO_BATCH = 16  # tokens per rank per step = 16 * 4096 = 65536 (fills HBM bett
 
 
 def build(args, device):
-    from dolomite_engine_amd.hf_models import GPTDolomiteConfig, GPTDolomiteForCausalLM
     from dolomite_engine_amd.model_wrapper import ModelWrapperForPretraining
     from dolomite_engine_amd.zero import ZeRO2Engine
 

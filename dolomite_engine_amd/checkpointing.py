@@ -12,7 +12,6 @@
 """
 
 import json
-import os
 import random
 from pathlib import Path
 

@@ -13,7 +13,7 @@ from .data import get_finetuning_dataloader
 from .model_wrapper import ModelWrapperForFinetuning
 from .optimization import get_scheduler
 from .train_utils import train_step
-from .utils import get_rank, get_world_size, init_distributed, log_rank_0
+from .utils import get_rank, init_distributed, log_rank_0
 from .zero import ZeRO2Engine
 
 

@@ -6,13 +6,13 @@ import time
 
 import torch
 
-from .arguments import Mode, TrainingArgs, parse_args
+from .arguments import TrainingArgs, parse_args
 from .checkpointing import load_checkpoint_for_training, save_checkpoint
 from .data import SyntheticPretrainingDataLoader
 from .model_wrapper import ModelWrapperForPretraining
 from .optimization import get_scheduler
 from .train_utils import get_model_tflops, train_step
-from .utils import get_rank, get_world_size, init_distributed, log_rank_0
+from .utils import get_rank, get_world_size, init_distributed
 from .zero import ZeRO2Engine
 
 
