@@ -34,3 +34,30 @@ for row in cur.execute(sel):
     agg[key][1] += en - st
 for (name, grid), (cnt, tot) in sorted(agg.items(), key=lambda kv: -kv[1][1]):
     print(f"{name[:58]:60s} grid={str(grid):24s} calls={cnt:4d} total={tot/1e6:9.1f}ms avg={tot/1e6/cnt:7.3f}ms")
+
+
+def modal_split(db_cur, u, ksym, name_filter, mod, labels):
+    """Bucket a kernel's dispatches by (order index % mod) — the per-layer
+    dispatch cycle is deterministic, so this splits shared Tensile kernels
+    by GEMM shape."""
+    rows = []
+    for kid, st, en in db_cur.execute(
+        f"SELECT kernel_id, {start_c}, {end_c} FROM rocpd_kernel_dispatch_{u} ORDER BY {start_c}"
+    ):
+        if name_filter in ksym.get(kid, ""):
+            rows.append(en - st)
+    # drop the trailing odd calls (lm_head has its own grid bucket anyway)
+    n = len(rows) - (len(rows) % mod)
+    agg = defaultdict(lambda: [0, 0])
+    for i, d in enumerate(rows[:n]):
+        agg[i % mod][0] += 1
+        agg[i % mod][1] += d
+    for m in range(mod):
+        c, t = agg[m]
+        lab = labels[m] if m < len(labels) else str(m)
+        print(f"  pos {m} ({lab:8s}): calls={c:4d} avg={t/1e6/max(c,1):7.3f}ms")
+
+
+if len(sys.argv) > 3 and sys.argv[3] == "mod4":
+    print("MT256x256x32 fwd calls by position in the layer cycle:")
+    modal_split(cur, u, ksym, "MT256x256x32", 4, ["c_attn", "c_proj", "c_fc", "mlp_proj"])
