@@ -4,6 +4,10 @@ run ~400 TF in-step vs >1 PF isolated. Variants:
            fresh output allocations per call (exactly the model's fwd shape cycle)
   cycle  — the 4 GEMMs back-to-back, no elementwise kernels in between
   prealloc — cycle but with out= preallocated buffers (no allocator traffic)
+  autograd — chain under autograd with the graph kept alive (real activation
+             footprint, ~150 GB retained, autograd dispatch overhead)
+  ballast  — cycle with ~120 GB of dead retained tensors (memory footprint
+             alone, no autograd)
 Pure analysis tool.
 """
 import time
@@ -40,11 +44,18 @@ def run(variant: str, passes: int = 3):
     # preallocated outputs for the no-alloc variant
     o = [torch.empty(M, QKV, dtype=dt, device=dev), torch.empty(M, H, dtype=dt, device=dev),
          torch.empty(M, FF, dtype=dt, device=dev), torch.empty(M, H, dtype=dt, device=dev)]
+    ballast = []
+    if variant == "ballast":
+        for _ in range(60):
+            ballast.append(torch.empty(1 << 30, dtype=dt, device=dev))  # 60 x 2 GiB
     for p in range(passes + 1):  # pass 0 = warmup, untimed
-        h = x
+        h = x.requires_grad_(False)
+        if variant == "autograd":
+            h = x.clone().requires_grad_(True)
+        keep = []
         for li in range(L):
             evs = pairs[li]
-            if variant == "chain":
+            if variant in ("chain", "autograd"):
                 h = h * torch.rsqrt(h.float().pow(2).mean(-1, keepdim=True) + 1e-5).to(dt)  # norm mimic
                 evs[0][0].record(); qkv = torch.nn.functional.linear(h, w_qkv, b_qkv); evs[0][1].record()
                 a = qkv[:, :H].contiguous()  # attn-output mimic (copy kernel)
@@ -55,7 +66,9 @@ def run(variant: str, passes: int = 3):
                 u = torch.nn.functional.gelu(u)
                 evs[3][0].record(); h2 = torch.nn.functional.linear(u, w_mproj, b_mproj); evs[3][1].record()
                 h = h + h2
-            elif variant == "cycle":
+                if variant == "autograd":
+                    keep.append((qkv, u, h2))  # hold the graph + activations
+            elif variant in ("cycle", "ballast"):
                 evs[0][0].record(); qkv = torch.nn.functional.linear(h, w_qkv, b_qkv); evs[0][1].record()
                 a = qkv[:, :H].contiguous()
                 evs[1][0].record(); a = torch.nn.functional.linear(a, w_proj, b_proj); evs[1][1].record()
@@ -77,5 +90,9 @@ def run(variant: str, passes: int = 3):
     print(f"{variant:9s} " + "  ".join(f"{names[i]}={tfs[i]*n/acc[i]:5.0f}TF({acc[i]/n*1e3:5.2f}ms)" for i in range(4)))
 
 
-for v in ["chain", "cycle", "prealloc"]:
+import sys
+
+variants = sys.argv[1:] or ["chain", "cycle", "prealloc", "autograd", "ballast"]
+for v in variants:
     run(v)
+    torch.cuda.empty_cache()
