@@ -1,0 +1,79 @@
+"""Event-timed fwd GEMMs inside the REAL bench step (flagship 3B, B=16),
+independent of rocprof attribution: CUDA-event pairs around every Linear
+module's forward, synced at step end. Pure analysis tool."""
+import argparse
+from collections import defaultdict
+
+import torch
+
+import bench as B
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--model", default="3b")
+    ap.add_argument("--micro-batch", type=int, default=B.MICRO_BATCH)
+    ap.add_argument("--seq-len", type=int, default=B.SEQ_LEN)
+    ap.add_argument("--steps", type=int, default=2)
+    args = ap.parse_args()
+
+    dev = torch.device("cuda")
+    torch.manual_seed(0)
+    wrapper, engine = B.build(args, dev)
+    from dolomite_engine_amd.optimization import LRScheduler
+
+    sched = LRScheduler(1e-5, 0, 0, None, 10**9, "constant", 0.1)
+
+    rec = defaultdict(list)  # short name -> [(ev_start, ev_end)]
+
+    def mk(nm):
+        def pre(mod, inp):
+            e = torch.cuda.Event(enable_timing=True)
+            e.record()
+            mod._t0 = e
+
+        def post(mod, inp, out):
+            e = torch.cuda.Event(enable_timing=True)
+            e.record()
+            rec[nm].append((mod._t0, e))
+
+        return pre, post
+
+    hooks = []
+    for name, mod in wrapper.model.named_modules():
+        if isinstance(mod, torch.nn.Linear):
+            short = ".".join(name.split(".")[-2:])  # attn.c_proj vs mlp.c_proj
+            pre, post = mk(short)
+            hooks.append(mod.register_forward_pre_hook(pre))
+            hooks.append(mod.register_forward_hook(post))
+
+    vocab = wrapper.model.config.vocab_size
+    B.run_steps(wrapper, engine, sched, 0, args.micro_batch, args.seq_len, vocab, 1)  # warmup, untimed
+    torch.cuda.synchronize()
+    rec.clear()
+    B.run_steps(wrapper, engine, sched, 0, args.micro_batch, args.seq_len, vocab, args.steps, step0=1)
+    torch.cuda.synchronize()
+
+    M = args.micro_batch * args.seq_len
+    cfg = wrapper.model.config
+    h, ff = cfg.n_embd, cfg.n_inner
+    D = h // cfg.n_head
+    qkv_n = h + 2 * D * (cfg.num_key_value_heads or 1) if (cfg.num_key_value_heads or 1) != cfg.n_head else 3 * h
+    flops = {
+        "attn.c_attn": 2 * M * qkv_n * h,
+        "attn.c_proj": 2 * M * h * h,
+        "mlp.c_fc": 2 * M * ff * h,
+        "mlp.c_proj": 2 * M * h * ff,
+    }
+    for nm, pairs in sorted(rec.items()):
+        tot = sum(a.elapsed_time(b) for a, b in pairs) / 1e3
+        avg = tot / len(pairs)
+        line = f"{nm:10s} calls={len(pairs):4d} avg={avg:7.3f}ms"
+        f = flops.get(nm)
+        if f:
+            line += f"  -> {f / 1e9 / avg:6.0f} TF"
+        print(line)
+
+
+if __name__ == "__main__":
+    main()
