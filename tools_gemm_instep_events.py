@@ -66,9 +66,9 @@ def main():
         "mlp.c_proj": 2 * M * h * ff,
     }
     for nm, pairs in sorted(rec.items()):
-        tot = sum(a.elapsed_time(b) for a, b in pairs) / 1e3
-        avg = tot / len(pairs)
-        line = f"{nm:10s} calls={len(pairs):4d} avg={avg:7.3f}ms"
+        ms = [a.elapsed_time(b) for a, b in pairs]
+        avg = sum(ms) / len(ms)
+        line = f"{nm:12s} calls={len(ms):4d} avg={avg:7.3f}ms min={min(ms):7.3f} max={max(ms):7.3f}"
         f = flops.get(nm)
         if f:
             line += f"  -> {f / 1e9 / avg:6.0f} TF"
