@@ -40,6 +40,11 @@ def main():
         return pre, post
 
     hooks = []
+    # whole-forward device interval (root module): arbitrates between the
+    # rocprof per-kernel durations and the per-linear event intervals
+    fpre, fpost = mk("WHOLE_FWD")
+    hooks.append(wrapper.model.register_forward_pre_hook(fpre))
+    hooks.append(wrapper.model.register_forward_hook(fpost))
     for name, mod in wrapper.model.named_modules():
         if isinstance(mod, torch.nn.Linear):
             short = ".".join(name.split(".")[-2:])  # attn.c_proj vs mlp.c_proj
