@@ -45,6 +45,34 @@ def main():
     fpre, fpost = mk("WHOLE_FWD")
     hooks.append(wrapper.model.register_forward_pre_hook(fpre))
     hooks.append(wrapper.model.register_forward_hook(fpost))
+
+    from dolomite_engine_amd.hf_models.modeling import Attention, GPTDolomiteBlock
+
+    def mkb(nm):
+        def bpre(mod, gout):
+            e = torch.cuda.Event(enable_timing=True)
+            e.record()
+            mod._bt0 = e
+
+        def bpost(mod, gin, gout):
+            e = torch.cuda.Event(enable_timing=True)
+            e.record()
+            rec[nm].append((mod._bt0, e))
+
+        return bpre, bpost
+
+    for name, mod in wrapper.model.named_modules():
+        if isinstance(mod, Attention):
+            pre, post = mk("ATTN_FWD")
+            hooks.append(mod.register_forward_pre_hook(pre))
+            hooks.append(mod.register_forward_hook(post))
+            bpre, bpost = mkb("ATTN_BWD")
+            hooks.append(mod.register_full_backward_pre_hook(bpre))
+            hooks.append(mod.register_full_backward_hook(bpost))
+        if isinstance(mod, GPTDolomiteBlock):
+            bpre, bpost = mkb("BLOCK_BWD")
+            hooks.append(mod.register_full_backward_pre_hook(bpre))
+            hooks.append(mod.register_full_backward_hook(bpost))
     for name, mod in wrapper.model.named_modules():
         if isinstance(mod, torch.nn.Linear):
             short = ".".join(name.split(".")[-2:])  # attn.c_proj vs mlp.c_proj
