@@ -99,10 +99,34 @@ def main():
             hooks.append(mod.register_forward_hook(post))
 
     vocab = wrapper.model.config.vocab_size
-    B.run_steps(wrapper, engine, sched, 0, args.micro_batch, args.seq_len, vocab, 1)  # warmup, untimed
+
+    def step(i):
+        # train_step with region brackets (same op order as train_utils.train_step)
+        def ev(nm):
+            e = torch.cuda.Event(enable_timing=True)
+            e.record()
+            return e
+
+        engine.zero_grad()
+        engine.set_sync(True)
+        batch = B.synth_batch(0, i, args.micro_batch, args.seq_len, vocab)
+        e0 = ev("")
+        loss = wrapper(batch)
+        e1 = ev("")
+        loss.backward()
+        e2 = ev("")
+        engine.step(lr=sched.get_lr(), grad_clip=1.0)
+        sched.step()
+        e3 = ev("")
+        rec["REGION_fwd"].append((e0, e1))
+        rec["REGION_bwd"].append((e1, e2))
+        rec["REGION_opt"].append((e2, e3))
+
+    step(0)  # warmup, untimed
     torch.cuda.synchronize()
     rec.clear()
-    B.run_steps(wrapper, engine, sched, 0, args.micro_batch, args.seq_len, vocab, args.steps, step0=1)
+    for i in range(args.steps):
+        step(1 + i)
     torch.cuda.synchronize()
 
     M = args.micro_batch * args.seq_len
