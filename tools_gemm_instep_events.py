@@ -36,6 +36,21 @@ def main():
             e = torch.cuda.Event(enable_timing=True)
             e.record()
             rec[nm].append((mod._t0, e))
+            # bracket this call's backward node (dgrad+wgrad for Linears)
+            if out.grad_fn is not None:
+                node, cell = out.grad_fn, [None]
+
+                def bpre(_g, cell=cell):
+                    cell[0] = torch.cuda.Event(enable_timing=True)
+                    cell[0].record()
+
+                def bpost(_gi, _go, cell=cell):
+                    e1 = torch.cuda.Event(enable_timing=True)
+                    e1.record()
+                    rec[nm + "_bwd"].append((cell[0], e1))
+
+                node.register_prehook(bpre)
+                node.register_hook(bpost)
 
         return pre, post
 
