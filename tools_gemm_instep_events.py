@@ -37,7 +37,7 @@ def main():
             e.record()
             rec[nm].append((mod._t0, e))
             # bracket this call's backward node (dgrad+wgrad for Linears)
-            if out.grad_fn is not None:
+            if getattr(out, "grad_fn", None) is not None:
                 node, cell = out.grad_fn, [None]
 
                 def bpre(_g, cell=cell):
