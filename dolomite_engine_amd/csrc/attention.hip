@@ -654,11 +654,9 @@ extern "C" int dolomite_fa_bwd_preprocess(dolomite_stream_t stream,
 
 // Backward is split FA2-style into two kernels (each recomputes P from
 // q/k/lse — cheaper than the atomics+barriers a fused version needs):
-//   fa_bwd_dkv_kernel : grid (kv-tile, seq, head-group); a head group folds
-//     merge (=2 when G is even) q-heads sharing one kv head, so dK/dV
-//     accumulate in registers over the q-tile loop AND across the folded
-//     heads, then store once into (T, H/merge, D) fp32 partials
-//     (G/merge contributors per kv strip for the finalize reduction).
+//   fa_bwd_dkv_kernel : grid (kv-tile, seq, q-head); dK/dV accumulate in
+//     registers over the q-tile loop, joined into fp32 buffers by one atomic
+//     pass per workgroup (G q-head contributors per kv strip).
 //   fa_bwd_dq_kernel  : grid (q-tile, seq, q-head); dQ accumulates in
 //     registers over the kv-tile loop — no atomics at all — and is stored
 //     once. K^T is the only LDS image (swizzled), staged per kv tile.
@@ -688,17 +686,9 @@ __global__ void __launch_bounds__(512, DPAD <= 96 ? 4 : 2) fa_bwd_dkv_kernel(
     constexpr int KCH = DPAD / 32;
     constexpr int DCH = DPAD / 16;
 
-    // grid z enumerates HEAD GROUPS: merge q-heads sharing one kv head fold
-    // into one workgroup (merge = H / gridDim.z, 2 when G is even). The K/V
-    // register fragments are staged once for the group and dK/dV accumulate
-    // across the folded heads in registers (dK = sum_h dK_h for shared K),
-    // halving the fp32 partial traffic, the finalize reduction depth and
-    // the per-head K/V restaging the per-q-head grid paid.
-    int tile_id = blockIdx.x, b = blockIdx.y, hg = blockIdx.z;
-    xcd_remap_tile_bh(tile_id, b, hg);
-    const int merge = H / gridDim.z;
-    const int GG = G / merge;
-    const int kvh = hg / GG;
+    int tile_id = blockIdx.x, b = blockIdx.y, h = blockIdx.z;
+    xcd_remap_tile_bh(tile_id, b, h);
+    const int kvh = h / G;
     // cu-derived values are wave-uniform: pin them to SGPRs so per-lane
     // address arithmetic built on them does not hold VGPR pairs across the
     // main loop (measured: the compiler otherwise spills pointer pairs)
@@ -778,6 +768,8 @@ __global__ void __launch_bounds__(512, DPAD <= 96 ? 4 : 2) fa_bwd_dkv_kernel(
 
     const int qt0 = ks / 64;
     const int nqt = (L + 63) / 64;
+    const int64_t q_hoff = (int64_t)(h / G) * q_gs + (int64_t)(h % G) * D;
+    const int64_t do_hoff = (int64_t)h * D;
 
     // MFMA pad columns [D, DPAD) zeroed ONCE: the fast staging path below
     // then only writes the real D columns per tile (guards and pad-fill
@@ -799,10 +791,6 @@ __global__ void __launch_bounds__(512, DPAD <= 96 ? 4 : 2) fa_bwd_dkv_kernel(
     // runtime division replaced by an exact magic multiply: pidx < 2048
     // and D/8 <= 16, so (pidx * (2^22/(D/8) + 1)) >> 22 is exact)
     const unsigned st_magic = (1u << 22) / (unsigned)(D / 8) + 1;
-    for (int mi = 0; mi < merge; ++mi) {
-    const int h = __builtin_amdgcn_readfirstlane(hg * merge + mi);
-    const int64_t q_hoff = (int64_t)kvh * q_gs + (int64_t)(h % G) * D;
-    const int64_t do_hoff = (int64_t)h * D;
     for (int qt = qt0; qt < nqt; ++qt) {
         const int qs = qt * 64;
         __syncthreads();  // previous iteration's image reads done
@@ -936,13 +924,11 @@ __global__ void __launch_bounds__(512, DPAD <= 96 ? 4 : 2) fa_bwd_dkv_kernel(
             __builtin_amdgcn_sched_barrier(0);
         }
     }
-    }  // mi (folded q-heads; dkr/dvr carry the running sum across them)
 
-    // Deterministic join: each (kv-tile, head-group) workgroup owns its keys'
-    // (t, hg) slot of the partial buffers (T, H/merge, D) exclusively, so
-    // dK/dV are plain stores (half the traffic of atomic RMW) and the in-
-    // register head fold is fixed-order; fa_grad_finalize reduces the G/merge
-    // group contributions per kv head in a FIXED order. The round-1
+    // Deterministic join: each (kv-tile, q-head) workgroup owns its keys'
+    // (t, h) slot of the per-head partial buffers exclusively, so dK/dV
+    // are plain stores (half the traffic of atomic RMW); fa_grad_finalize
+    // reduces the G q-head contributions in a FIXED order. The round-1
     // fp32 atomicAdd join was arrival-order nondeterministic, which made
     // bit-exact checkpoint resume a coin flip.
 #pragma unroll
@@ -952,7 +938,7 @@ __global__ void __launch_bounds__(512, DPAD <= 96 ? 4 : 2) fa_bwd_dkv_kernel(
             const int kpos = ks + wave * 16 + lg * 4 + r;
             const int d = dc * 16 + lr;
             if (kpos < kend && d < D) {
-                int64_t idx = ((int64_t)(s0 + kpos) * gridDim.z + hg) * D + d;
+                int64_t idx = ((int64_t)(s0 + kpos) * H + h) * D + d;
                 // non-temporal: the 1.3 GB of fp32 partials per launch must
                 // not evict the XCD-resident Q/dO slices (PMC: dkv read
                 // traffic 3.6 GB/launch vs ~1.4 GB ideal with plain stores)
@@ -1191,13 +1177,9 @@ static int launch_fa_bwd(hipStream_t stream, const __bf16* q, const __bf16* k, c
     constexpr int ST = 64 + 8;
     dim3 block(512);
     dim3 grid(max_tiles, batch, H);
-    // dkv folds pairs of q-heads sharing a kv head into one workgroup when G
-    // is even (partials become (T, H/2, D); finalize gets G/2 contributors).
-    // The host-side mirror of this rule lives in ops/functional.py.
-    dim3 grid_dkv(max_tiles, batch, (G % 2 == 0) ? H / 2 : H);
     constexpr int SQ = DPAD + 16;
     size_t shmem_dkv = (size_t)(64 * (128 + 4) * 2 + 64 * SQ * 2) * sizeof(__bf16);
-    hipLaunchKernelGGL((fa_bwd_dkv_kernel<DPAD>), grid_dkv, block, shmem_dkv, stream,
+    hipLaunchKernelGGL((fa_bwd_dkv_kernel<DPAD>), grid, block, shmem_dkv, stream,
                        q, k, v, dout, lse, delta, dk_acc, dv_acc, cu, H, Hkv, D, G,
                        q_ts, q_gs, k_ts, k_hs, v_ts, v_hs, do_ts, T, scale);
     int err = dol_last_error();

@@ -420,13 +420,10 @@ class VarlenAttention(torch.autograd.Function):
                 ),
                 "fa_bwd_preprocess",
             )
-            # (T, H/merge, D) per-head-group partials, written exclusively per
-            # workgroup (no zero-init needed, every valid slot is stored).
-            # merge mirrors the dkv launcher's fold rule: pairs of q-heads
-            # sharing a kv head accumulate in-register when G is even.
-            merge = 2 if lo.G % 2 == 0 else 1
-            dk_acc = torch.empty(T, lo.H // merge, lo.D, dtype=torch.float32, device=qkv.device)
-            dv_acc = torch.empty(T, lo.H // merge, lo.D, dtype=torch.float32, device=qkv.device)
+            # (T, H, D) per-q-head partials, written exclusively per
+            # workgroup (no zero-init needed, every valid slot is stored)
+            dk_acc = torch.empty(T, lo.H, lo.D, dtype=torch.float32, device=qkv.device)
+            dv_acc = torch.empty(T, lo.H, lo.D, dtype=torch.float32, device=qkv.device)
             dqkv = torch.empty_like(qkv)
             with hip.prof("fa_varlen_bwd"):
              hip.check(
@@ -444,7 +441,7 @@ class VarlenAttention(torch.autograd.Function):
             hip.check(
                 hip.lib().dolomite_fa_grad_finalize(
                     hip.stream(), hip.ptr(dk_acc), hip.ptr(dv_acc), hip.ptr(dqkv),
-                    T, lo.Hkv, lo.D, lo.G // merge, lo.row_len,
+                    T, lo.Hkv, lo.D, lo.G, lo.row_len,
                     lo.k_off, lo.kv_hstride, lo.v_off, hip.dt(qkv),
                 ),
                 "fa_grad_finalize",

@@ -128,11 +128,8 @@ int dolomite_fa_varlen_fwd(dolomite_stream_t stream,
  *  1. preprocess: delta[h,t] = rowsum(dO[t,h,:]*O[t,h,:]) fp32.
  *  2. main: grid over (kv-tile, seq, kv-head); recomputes P from q,k,lse;
  *     dk/dv accumulated in registers over the q-tile loop, then added to the
- *     pointers; dkv: one workgroup per (kv-tile, seq, HEAD-GROUP), where a
- *     head group folds merge q-heads sharing one kv head (merge = 2 when G
- *     is even, else 1); dk/dv accumulate in registers across the q-tile loop
- *     and the folded heads and store once, exclusively (no atomics), into
- *     (T, H/merge, D) fp32 partials;
+ *     pointers; dkv: one workgroup per (kv-tile, seq, Q-HEAD); dk/dv join
+ *     zero-initialized (T,Hkv,D) fp32 buffers with device-scope atomics;
  *     dq: one workgroup per (q-tile, seq, q-head), register-accumulated and
  *     stored bf16 directly into the packed dqkv q-slots.
  *  3. grad_finalize: casts the dk/dv fp32 accumulators into the packed dqkv.
@@ -152,10 +149,9 @@ int dolomite_fa_varlen_bwd(dolomite_stream_t stream,
                            int64_t v_tstride, int64_t v_hstride,
                            int64_t do_tstride,
                            float scale, int dtype);
-/* dk_acc/dv_acc are (T, H/merge, D) fp32 per-head-group partials written
- * exclusively (no atomics) by dolomite_fa_varlen_bwd (merge as above); pass
- * G/merge as this function's G so the finalize reduces each kv head's
- * G/merge group contributors in fixed order — deterministic gradients. */
+/* dk_acc/dv_acc are (T, H, D) fp32 PER-Q-HEAD partials written exclusively
+ * (no atomics) by dolomite_fa_varlen_bwd; the finalize reduces each kv
+ * head's G contributors in fixed order — deterministic gradients. */
 int dolomite_fa_grad_finalize(dolomite_stream_t stream,
                               const float* dk_acc, const float* dv_acc,
                               void* dqkv, int64_t T, int Hkv, int D, int G,

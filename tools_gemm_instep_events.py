@@ -117,7 +117,7 @@ def main():
 
     def step(i):
         # train_step with region brackets (same op order as train_utils.train_step)
-        def ev(nm):
+        def ev():
             e = torch.cuda.Event(enable_timing=True)
             e.record()
             return e
@@ -125,14 +125,14 @@ def main():
         engine.zero_grad()
         engine.set_sync(True)
         batch = B.synth_batch(0, i, args.micro_batch, args.seq_len, vocab)
-        e0 = ev("")
+        e0 = ev()
         loss = wrapper(batch)
-        e1 = ev("")
+        e1 = ev()
         loss.backward()
-        e2 = ev("")
+        e2 = ev()
         engine.step(lr=sched.get_lr(), grad_clip=1.0)
         sched.step()
-        e3 = ev("")
+        e3 = ev()
         rec["REGION_fwd"].append((e0, e1))
         rec["REGION_bwd"].append((e1, e2))
         rec["REGION_opt"].append((e2, e3))
