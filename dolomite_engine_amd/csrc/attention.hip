@@ -633,10 +633,62 @@ __global__ void __launch_bounds__(256) fa_bwd_preprocess_kernel(
     if (lane == 0) delta[(int64_t)h * T_total + t] = acc;
 }
 
+// Fast path for the contiguous bf16 layout (every named config): the
+// one-wave-per-(t,h) kernel above issues only two b32 loads per lane for an
+// 80-element row — far too shallow to cover HBM latency (measured 0.36 ms
+// vs 0.11 ms algorithmic). Here a block streams a TOK-token tile with b128
+// loads (8-element groups never cross a head boundary since D % 8 == 0),
+// parks one fp32 partial dot per group in LDS, then one thread per (t, h)
+// slot sums its D/8 contiguous partials in FIXED ascending order — the
+// reduction stays bit-deterministic for the resume tests.
+__global__ void __launch_bounds__(256) fa_bwd_preprocess_tiled(
+    const __bf16* __restrict__ o, const __bf16* __restrict__ dout, float* __restrict__ delta,
+    int64_t T_total, int H, int D, int TOK) {
+    extern __shared__ float part[];  // [TOK*H*D/8]
+    const int64_t hd = (int64_t)H * D;
+    const int64_t base = (int64_t)blockIdx.x * TOK * hd;
+    const int64_t total = T_total * hd;
+    const int np = (int)(TOK * hd / 8);
+#pragma unroll 4
+    for (int p = threadIdx.x; p < np; p += 256) {
+        int64_t flat = base + (int64_t)p * 8;
+        float acc = 0.f;
+        if (flat + 8 <= total) {
+            bf16x8 ov = *(const bf16x8*)(o + flat);
+            bf16x8 dv = *(const bf16x8*)(dout + flat);
+#pragma unroll
+            for (int e = 0; e < 8; ++e) acc += (float)ov[e] * (float)dv[e];
+        }
+        part[p] = acc;
+    }
+    __syncthreads();
+    const int slots = TOK * H;
+    const int dp8 = D / 8;
+    for (int s = threadIdx.x; s < slots; s += 256) {
+        float a = 0.f;
+        const float* ps = &part[s * dp8];
+        for (int i = 0; i < dp8; ++i) a += ps[i];
+        int tl = s / H, h = s - tl * H;
+        int64_t t = (int64_t)blockIdx.x * TOK + tl;
+        if (t < T_total) delta[(int64_t)h * T_total + t] = a;
+    }
+}
+
 extern "C" int dolomite_fa_bwd_preprocess(dolomite_stream_t stream,
                                           const void* o, const void* dout, float* delta,
                                           int64_t T, int H, int D,
                                           int64_t o_tstride, int64_t do_tstride, int dtype) {
+    const int64_t hd = (int64_t)H * D;
+    int TOK = 16;
+    while (TOK > 1 && TOK * hd * 4 / 8 > 65536) TOK /= 2;  // LDS partial budget
+    if (dtype == DOLOMITE_BF16 && D % 8 == 0 && o_tstride == hd && do_tstride == hd && T > 0 &&
+        TOK * hd * 4 / 8 <= 65536) {
+        dim3 grid((uint32_t)((T + TOK - 1) / TOK)), block(256);
+        size_t shmem = (size_t)(TOK * hd / 8) * sizeof(float);
+        hipLaunchKernelGGL((fa_bwd_preprocess_tiled), grid, block, shmem, (hipStream_t)stream,
+                           (const __bf16*)o, (const __bf16*)dout, delta, T, H, D, TOK);
+        return dol_last_error();
+    }
     int64_t rows = T * H;
     dim3 grid((uint32_t)((rows + 3) / 4)), block(256);
     if (dtype == DOLOMITE_BF16)

@@ -117,6 +117,19 @@ struct VecIO<uint16_t, 8> {
         for (int e = 0; e < 8; ++e) v[e] = f32_to_bf16(in[e]);
         *(us8_t*)p = v;
     }
+    // single-touch streams: bypass L2 allocation (and the write-allocate
+    // read-for-ownership a plain store of a full line still pays)
+    static __device__ __forceinline__ void load_nt(const uint16_t* p, float* out) {
+        us8_t v = __builtin_nontemporal_load((const us8_t*)p);
+#pragma unroll
+        for (int e = 0; e < 8; ++e) out[e] = bf16_to_f32((uint16_t)v[e]);
+    }
+    static __device__ __forceinline__ void store_nt(uint16_t* p, const float* in) {
+        us8_t v;
+#pragma unroll
+        for (int e = 0; e < 8; ++e) v[e] = f32_to_bf16(in[e]);
+        __builtin_nontemporal_store(v, (us8_t*)p);
+    }
 };
 
 template <>
@@ -132,6 +145,17 @@ struct VecIO<float, 4> {
         for (int e = 0; e < 4; ++e) v[e] = in[e];
         *(f4_t*)p = v;
     }
+    static __device__ __forceinline__ void load_nt(const float* p, float* out) {
+        f4_t v = __builtin_nontemporal_load((const f4_t*)p);
+#pragma unroll
+        for (int e = 0; e < 4; ++e) out[e] = v[e];
+    }
+    static __device__ __forceinline__ void store_nt(float* p, const float* in) {
+        f4_t v;
+#pragma unroll
+        for (int e = 0; e < 4; ++e) v[e] = in[e];
+        __builtin_nontemporal_store(v, (f4_t*)p);
+    }
 };
 
 template <>
@@ -145,6 +169,8 @@ struct VecIO<uint16_t, 2> {
         uint32_t v = (uint32_t)f32_to_bf16(in[0]) | ((uint32_t)f32_to_bf16(in[1]) << 16);
         *(uint32_t*)p = v;
     }
+    static __device__ __forceinline__ void load_nt(const uint16_t* p, float* out) { load(p, out); }
+    static __device__ __forceinline__ void store_nt(uint16_t* p, const float* in) { store(p, in); }
 };
 
 template <>
@@ -157,18 +183,24 @@ struct VecIO<float, 2> {
         p[0] = in[0];
         p[1] = in[1];
     }
+    static __device__ __forceinline__ void load_nt(const float* p, float* out) { load(p, out); }
+    static __device__ __forceinline__ void store_nt(float* p, const float* in) { store(p, in); }
 };
 
 template <>
 struct VecIO<uint16_t, 1> {
     static __device__ __forceinline__ void load(const uint16_t* p, float* out) { out[0] = bf16_to_f32(*p); }
     static __device__ __forceinline__ void store(uint16_t* p, const float* in) { *p = f32_to_bf16(in[0]); }
+    static __device__ __forceinline__ void load_nt(const uint16_t* p, float* out) { load(p, out); }
+    static __device__ __forceinline__ void store_nt(uint16_t* p, const float* in) { store(p, in); }
 };
 
 template <>
 struct VecIO<float, 1> {
     static __device__ __forceinline__ void load(const float* p, float* out) { out[0] = *p; }
     static __device__ __forceinline__ void store(float* p, const float* in) { *p = in[0]; }
+    static __device__ __forceinline__ void load_nt(const float* p, float* out) { load(p, out); }
+    static __device__ __forceinline__ void store_nt(float* p, const float* in) { store(p, in); }
 };
 
 template <>
@@ -181,4 +213,6 @@ struct VecIO<float, 8> {
         VecIO<float, 4>::store(p, in);
         VecIO<float, 4>::store(p + 4, in + 4);
     }
+    static __device__ __forceinline__ void load_nt(const float* p, float* out) { load(p, out); }
+    static __device__ __forceinline__ void store_nt(float* p, const float* in) { store(p, in); }
 };

@@ -206,8 +206,8 @@ __global__ void __launch_bounds__(256) norm_bwd_kernel(
         for (int i = 0; i < ITMAX; ++i) {
             int c0 = (i * 64 + lane) * V;
             if (c0 < H) {
-                VecIO<T, V>::load(sr + c0, sh[i]);
-                VecIO<T, V>::load(dyr + c0, dyv[i]);
+                VecIO<T, V>::load_nt(sr + c0, sh[i]);
+                VecIO<T, V>::load_nt(dyr + c0, dyv[i]);
 #pragma unroll
                 for (int kk = 0; kk < V; ++kk) {
                     float shat = (sh[i][kk] - mu) * r;
@@ -232,7 +232,7 @@ __global__ void __launch_bounds__(256) norm_bwd_kernel(
             if (c0 < H) {
                 float dxv[V];
                 float drv[V];
-                if (HAS_DRES) VecIO<T, V>::load(drr + c0, drv);
+                if (HAS_DRES) VecIO<T, V>::load_nt(drr + c0, drv);
 #pragma unroll
                 for (int kk = 0; kk < V; ++kk) {
                     if (IS_LN) {
@@ -251,7 +251,7 @@ __global__ void __launch_bounds__(256) norm_bwd_kernel(
                 if (HAS_DRES)
 #pragma unroll
                     for (int kk = 0; kk < V; ++kk) dxv[kk] += drv[kk];
-                VecIO<T, V>::store(dxr + c0, dxv);
+                VecIO<T, V>::store_nt(dxr + c0, dxv);
             }
         }
     }
