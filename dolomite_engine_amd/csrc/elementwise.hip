@@ -178,21 +178,19 @@ __global__ void __launch_bounds__(256) norm_bwd_kernel(
     const int64_t wslot = (int64_t)blockIdx.x * 4 + (threadIdx.x >> 6);
     const int64_t nslots = (int64_t)gridDim.x * 4;
 
-    float wv[ITMAX][V], dw_acc[ITMAX][V], db_acc[IS_LN ? ITMAX : 1][IS_LN ? V : 1];
+    // Register budget is the occupancy lever here (HBM-bound kernel, one
+    // wave per row): dw/db accumulate in PHASE 1 so neither the weight nor
+    // the raw dy has to stay live across the cross-lane reduce — only
+    // s_hat and dy*w survive to phase 2. ITMAX=5/V=8 drops ~40 VGPRs,
+    // buying a third wave per SIMD to cover the phase-boundary stalls.
+    float dw_acc[ITMAX][V], db_acc[IS_LN ? ITMAX : 1][IS_LN ? V : 1];
 #pragma unroll
-    for (int i = 0; i < ITMAX; ++i) {
-        int c0 = (i * 64 + lane) * V;
-        if (c0 < H)
-            VecIO<T, V>::load(w + c0, wv[i]);
-        else
-#pragma unroll
-            for (int kk = 0; kk < V; ++kk) wv[i][kk] = 0.f;
+    for (int i = 0; i < ITMAX; ++i)
 #pragma unroll
         for (int kk = 0; kk < V; ++kk) {
             dw_acc[i][kk] = 0.f;
             if (IS_LN) db_acc[i][kk] = 0.f;
         }
-    }
 
     for (int64_t row = wslot; row < T_rows; row += nslots) {
         const T* dyr = dy + row * (int64_t)H;
@@ -200,25 +198,38 @@ __global__ void __launch_bounds__(256) norm_bwd_kernel(
         float r = rstd[row];
         float mu = IS_LN ? mean[row] : 0.f;
 
-        float sh[ITMAX][V], dyv[ITMAX][V];
+        float sh[ITMAX][V], dyw[ITMAX][V];
         float d1 = 0.f, d2 = 0.f;
 #pragma unroll
         for (int i = 0; i < ITMAX; ++i) {
             int c0 = (i * 64 + lane) * V;
             if (c0 < H) {
+                float wv[V], dyv[V];
+                VecIO<T, V>::load(w + c0, wv);  // L2-resident, re-read per row
                 VecIO<T, V>::load_nt(sr + c0, sh[i]);
-                VecIO<T, V>::load_nt(dyr + c0, dyv[i]);
+                VecIO<T, V>::load_nt(dyr + c0, dyv);
 #pragma unroll
                 for (int kk = 0; kk < V; ++kk) {
                     float shat = (sh[i][kk] - mu) * r;
-                    float dyw = dyv[i][kk] * wv[i][kk];
+                    float dw_ = dyv[kk] * wv[kk];
                     sh[i][kk] = shat;
-                    d2 += dyw * shat;
-                    if (IS_LN) d1 += dyw;
+                    dyw[i][kk] = dw_;
+                    d2 += dw_ * shat;
+                    if (IS_LN) {
+                        d1 += dw_;
+                        dw_acc[i][kk] += dyv[kk] * shat;
+                        db_acc[i][kk] += dyv[kk];
+                    } else {
+                        // dw uses the CAST normalized value (reference casts
+                        // before the weight multiply)
+                        T tmp;
+                        store_from_f32(&tmp, shat);
+                        dw_acc[i][kk] += dyv[kk] * load_as_f32(&tmp);
+                    }
                 }
             } else {
 #pragma unroll
-                for (int kk = 0; kk < V; ++kk) { sh[i][kk] = 0.f; dyv[i][kk] = 0.f; }
+                for (int kk = 0; kk < V; ++kk) { sh[i][kk] = 0.f; dyw[i][kk] = 0.f; }
             }
         }
         float m2 = wave_reduce_sum(d2) / (float)H;
@@ -234,20 +245,9 @@ __global__ void __launch_bounds__(256) norm_bwd_kernel(
                 float drv[V];
                 if (HAS_DRES) VecIO<T, V>::load_nt(drr + c0, drv);
 #pragma unroll
-                for (int kk = 0; kk < V; ++kk) {
-                    if (IS_LN) {
-                        dxv[kk] = r * (dyv[i][kk] * wv[i][kk] - m1 - sh[i][kk] * m2);
-                        dw_acc[i][kk] += dyv[i][kk] * sh[i][kk];
-                        db_acc[i][kk] += dyv[i][kk];
-                    } else {
-                        dxv[kk] = r * (wv[i][kk] * dyv[i][kk] - sh[i][kk] * m2);
-                        // dw uses the CAST normalized value (reference casts
-                        // before the weight multiply)
-                        T tmp;
-                        store_from_f32(&tmp, sh[i][kk]);
-                        dw_acc[i][kk] += dyv[i][kk] * load_as_f32(&tmp);
-                    }
-                }
+                for (int kk = 0; kk < V; ++kk)
+                    dxv[kk] = IS_LN ? r * (dyw[i][kk] - m1 - sh[i][kk] * m2)
+                                    : r * (dyw[i][kk] - sh[i][kk] * m2);
                 if (HAS_DRES)
 #pragma unroll
                     for (int kk = 0; kk < V; ++kk) dxv[kk] += drv[kk];

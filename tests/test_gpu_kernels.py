@@ -171,6 +171,7 @@ def test_rope_packed_roundtrip_and_oracle(head_type, H, Hkv, D):
         ("mqa", 8, 1, 80, [4096]),        # one full-length 3B-shaped sequence
         ("mqa", 8, 1, 80, [7, 0, 9]),     # empty sequence in the packed batch
         ("gqa", 8, 2, 128, [65, 63]),     # edge tiles on both sides of 64
+        ("gqa", 6, 2, 64, [100, 60]),     # odd G=3 grouping
     ],
 )
 def test_varlen_attention_fwd_bwd_vs_oracle(head_type, H, Hkv, D, lens):
