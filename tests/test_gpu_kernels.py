@@ -218,6 +218,28 @@ def test_varlen_attention_deterministic_list_vs_tensor_packing():
     assert torch.equal(o1, o2)
 
 
+@pytest.mark.parametrize("head_type,H,Hkv,D", [("mqa", 8, 1, 80), ("gqa", 8, 2, 128)])
+def test_varlen_attention_backward_bitwise_deterministic(head_type, H, Hkv, D):
+    """Backward twice on the same inputs -> bitwise-identical dqkv. Guards
+    the whole deterministic-gradient chain (tiled preprocess LDS reduction,
+    exclusive per-q-head dk/dv partial stores, fixed-order finalize) that
+    bit-exact checkpoint resume depends on — the round-1 fp32 atomicAdd
+    join failed exactly this."""
+    g = torch.Generator().manual_seed(11)
+    lo = QKVLayout.make(H, Hkv, D, head_type)
+    cu = torch.tensor([0, 300, 428, 4524], dtype=torch.int32).cuda()
+    qkv0 = (torch.randn(4524, lo.row_len, generator=g) * 0.5).to(torch.bfloat16).cuda()
+    dout = (torch.randn(4524, H * D, generator=g) * 0.5).to(torch.bfloat16).cuda()
+    grads = []
+    for _ in range(2):
+        qkv = qkv0.clone().requires_grad_(True)
+        o = Fx.varlen_attention(qkv, cu, 4096, lo, 1.0 / math.sqrt(D))
+        o.backward(dout)
+        torch.cuda.synchronize()
+        grads.append(qkv.grad.clone())
+    assert torch.equal(grads[0], grads[1])
+
+
 @pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
 @pytest.mark.parametrize("TV", [(64, 512), (230, 49152)])
 def test_fused_cross_entropy(dtype, TV):
