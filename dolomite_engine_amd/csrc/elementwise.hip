@@ -803,9 +803,13 @@ __global__ void __launch_bounds__(256) adamw_kernel(
     float* __restrict__ master, uint16_t* __restrict__ param_out,
     const void* __restrict__ grad_v, float* __restrict__ m, float* __restrict__ v,
     int64_t n, float lr, float b1, float b2, float eps, float wd,
-    float bc1, float bc2) {
+    float bc1, float bc2, const float* __restrict__ gscale) {
     int64_t i0 = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) * 4;
     if (i0 >= n) return;
+    // fused grad-clip scale: one device scalar read replaces a whole
+    // read+write pass over the shard grads (reference clips then steps;
+    // scaling at the point of use is the same arithmetic in fp32)
+    const float gs = gscale ? *gscale : 1.f;
     if (i0 + 4 <= n) {  // vector fast path (16B loads/stores per stream)
         f4_t p4 = *(f4_t*)&master[i0];
         f4_t m4 = *(f4_t*)&m[i0];
@@ -818,6 +822,8 @@ __global__ void __launch_bounds__(256) adamw_kernel(
         } else {
             g4 = *(const f4_t*)&((const float*)grad_v)[i0];
         }
+#pragma unroll
+        for (int k = 0; k < 4; ++k) g4[k] *= gs;
         us4_t o4;
 #pragma unroll
         for (int k = 0; k < 4; ++k) {
@@ -841,6 +847,7 @@ __global__ void __launch_bounds__(256) adamw_kernel(
         int64_t i = i0 + k;
         if (i >= n) return;
         float g = GRAD_BF16 ? bf16_to_f32(((const uint16_t*)grad_v)[i]) : ((const float*)grad_v)[i];
+        g *= gs;
         float p = master[i];
         p *= (1.f - lr * wd);
         float mi = m[i] * b1 + (1.f - b1) * g;
@@ -859,7 +866,8 @@ extern "C" int dolomite_adamw_step(dolomite_stream_t stream,
                                    const void* grad, int grad_dtype,
                                    float* m, float* v,
                                    int64_t n, float lr, float beta1, float beta2,
-                                   float eps, float weight_decay, int step) {
+                                   float eps, float weight_decay, int step,
+                                   const float* gscale) {
     if (n == 0) return 0;
     float bc1 = 1.f - powf(beta1, (float)step);
     float bc2 = 1.f - powf(beta2, (float)step);
@@ -867,10 +875,94 @@ extern "C" int dolomite_adamw_step(dolomite_stream_t stream,
     dim3 grid((uint32_t)((nthreads + 255) / 256)), block(256);
     if (grad_dtype == DOLOMITE_BF16)
         hipLaunchKernelGGL((adamw_kernel<1>), grid, block, 0, (hipStream_t)stream,
-                           master, (uint16_t*)param_out_bf16, grad, m, v, n, lr, beta1, beta2, eps, weight_decay, bc1, bc2);
+                           master, (uint16_t*)param_out_bf16, grad, m, v, n, lr, beta1, beta2, eps, weight_decay, bc1, bc2, gscale);
     else
         hipLaunchKernelGGL((adamw_kernel<0>), grid, block, 0, (hipStream_t)stream,
-                           master, (uint16_t*)param_out_bf16, grad, m, v, n, lr, beta1, beta2, eps, weight_decay, bc1, bc2);
+                           master, (uint16_t*)param_out_bf16, grad, m, v, n, lr, beta1, beta2, eps, weight_decay, bc1, bc2, gscale);
+    return dol_last_error();
+}
+
+// ===========================================================================
+// Deterministic squared-sum (grad-norm input): fixed-order per-thread
+// strided accumulation + a fixed LDS tree per block -> one fp32 partial per
+// block, folded by dolomite_reduce_partials (fixed loop order). Reading the
+// bf16 flat grads directly halves the bytes of the fp32-cast-then-
+// pow(2).sum() path, and the fixed reduction order keeps grad-norm (and so
+// the clipped step) bit-deterministic for checkpoint-resume.
+// ===========================================================================
+
+#define SQSUM_BLOCKS 1024
+
+// 8 elements per thread for BOTH dtypes: the thread->element mapping and
+// per-thread accumulation order are then identical whether the grads are
+// the bf16 flat bucket (world-1 plain path) or the fp32 reduce-scatter
+// shard (collective path), so the two paths' grad norms agree BITWISE
+// (the world-1 RCCL-equivalence test asserts exactly that).
+template <typename T>
+__global__ void __launch_bounds__(256) sqsum_kernel(
+    const T* __restrict__ x, int64_t n, float* __restrict__ partials) {
+    __shared__ float red[256];
+    const int64_t stride = (int64_t)gridDim.x * 256 * 8;
+    float acc = 0.f;
+    for (int64_t i = ((int64_t)blockIdx.x * 256 + threadIdx.x) * 8; i < n; i += stride) {
+        float b[8];
+        if (i + 8 <= n) {
+            if constexpr (sizeof(T) == 2) {
+                VecIO<uint16_t, 8>::load_nt((const uint16_t*)x + i, b);
+            } else {
+                VecIO<float, 4>::load_nt((const float*)x + i, b);
+                VecIO<float, 4>::load_nt((const float*)x + i + 4, b + 4);
+            }
+        } else {
+#pragma unroll
+            for (int k = 0; k < 8; ++k) b[k] = (i + k < n) ? load_as_f32(x + i + k) : 0.f;
+        }
+#pragma unroll
+        for (int k = 0; k < 8; ++k) acc += b[k] * b[k];
+    }
+    red[threadIdx.x] = acc;
+    __syncthreads();
+#pragma unroll
+    for (int s_ = 128; s_ > 0; s_ >>= 1) {
+        if ((int)threadIdx.x < s_) red[threadIdx.x] += red[threadIdx.x + s_];
+        __syncthreads();
+    }
+    if (threadIdx.x == 0) partials[blockIdx.x] = red[0];
+}
+
+__global__ void __launch_bounds__(256) sqsum_fold_kernel(const float* __restrict__ partials,
+                                                         float* __restrict__ out) {
+    __shared__ float red[256];
+    float acc = 0.f;
+#pragma unroll
+    for (int k = 0; k < SQSUM_BLOCKS / 256; ++k) acc += partials[k * 256 + threadIdx.x];
+    red[threadIdx.x] = acc;
+    __syncthreads();
+#pragma unroll
+    for (int s_ = 128; s_ > 0; s_ >>= 1) {
+        if ((int)threadIdx.x < s_) red[threadIdx.x] += red[threadIdx.x + s_];
+        __syncthreads();
+    }
+    if (threadIdx.x == 0) out[0] = red[0];
+}
+
+extern "C" int dolomite_sqsum(dolomite_stream_t stream, const void* x, int64_t n,
+                              float* partials, float* out, int dtype) {
+    hipStream_t s = (hipStream_t)stream;
+    dim3 grid(SQSUM_BLOCKS), block(256);
+    if (n == 0) {
+        (void)hipMemsetAsync(out, 0, sizeof(float), s);
+        return dol_last_error();
+    }
+    if (dtype == DOLOMITE_BF16)
+        hipLaunchKernelGGL((sqsum_kernel<uint16_t>), grid, block, 0, s, (const uint16_t*)x, n, partials);
+    else
+        hipLaunchKernelGGL((sqsum_kernel<float>), grid, block, 0, s, (const float*)x, n, partials);
+    int err = dol_last_error();
+    if (err) return err;
+    // NOT dolomite_reduce_partials: its multi-split path joins with fp32
+    // atomicAdd (order-nondeterministic). One block, fixed tree.
+    hipLaunchKernelGGL((sqsum_fold_kernel), dim3(1), dim3(256), 0, s, partials, out);
     return dol_last_error();
 }
 

@@ -561,21 +561,27 @@ def fused_cross_entropy(logits, labels, ignore_index: int = -100):
 # ---------------------------------------------------------------------------
 
 
-def adamw_step_flat(master, grad, exp_avg, exp_avg_sq, step, lr, beta1, beta2, eps, weight_decay, param_out=None):
+def adamw_step_flat(master, grad, exp_avg, exp_avg_sq, step, lr, beta1, beta2, eps, weight_decay, param_out=None,
+                    grad_scale=None):
     """torch.optim.AdamW semantics on flat fp32 tensors, optional bf16
-    write-out of the updated params (the ZeRO-2 local shard step)."""
+    write-out of the updated params (the ZeRO-2 local shard step).
+    grad may be bf16 (the flat autograd bucket) — the kernel upcasts per
+    element. grad_scale: optional 0-dim DEVICE tensor (the grad-clip
+    coefficient) fused into the kernel's grad read."""
     if master.is_cuda:
         hip.check(
             hip.lib().dolomite_adamw_step(
                 hip.stream(), hip.ptr(master), hip.ptr(param_out), hip.ptr(grad),
                 hip.dt(grad), hip.ptr(exp_avg), hip.ptr(exp_avg_sq),
                 master.numel(), float(lr), float(beta1), float(beta2), float(eps),
-                float(weight_decay), int(step),
+                float(weight_decay), int(step), hip.ptr(grad_scale),
             ),
             "adamw_step",
         )
         return
     g = grad.float()
+    if grad_scale is not None:
+        g = g * float(grad_scale)
     master.mul_(1 - lr * weight_decay)
     exp_avg.mul_(beta1).add_(g, alpha=1 - beta1)
     exp_avg_sq.mul_(beta2).addcmul_(g, g, value=1 - beta2)

@@ -92,6 +92,9 @@ class ZeRO2Engine:
         # 8-GPU overlap path runs, instead of that branch being dead until
         # a multi-GPU node exists. Covered by tests/test_gpu_model.py.
         self.use_coll = self.world > 1 or (force_collectives and is_initialized())
+        self._sq_partials = None   # (1024,) fp32 sqsum scratch (lazy)
+        self._sq_out = None        # per-bucket squared-sum outputs (lazy)
+        self._clip_coef = None     # device scalar fused into the AdamW grad read
 
         params = [p for p in model.parameters() if p.requires_grad]
         assert params
@@ -192,7 +195,9 @@ class ZeRO2Engine:
 
     def _reduce_sync(self, b: _Bucket) -> None:
         if not self.use_coll:
-            b.shard_grad.copy_(b.flat_grad[b.shard_slice])
+            # single rank, no collectives: the optimizer reads the flat
+            # bf16/fp32 grad bucket directly (_grad_src) — the fp32 cast
+            # copy this branch used to do was a pure bandwidth pass
             return
         backend = dist.get_backend()
         if backend == "nccl":
@@ -213,18 +218,45 @@ class ZeRO2Engine:
             else:
                 self._reduce_sync(b)
 
+    def _grad_src(self, b: _Bucket) -> torch.Tensor:
+        """The tensor the optimizer step reads: the fp32 reduce-scatter
+        shard on the collective path, the flat param-dtype grad bucket
+        itself otherwise (the AdamW kernel upcasts per element)."""
+        return b.shard_grad if self.use_coll else b.flat_grad[b.shard_slice]
+
     def clip_grad_norm_(self, max_norm: float) -> torch.Tensor:
-        """Global grad-norm clip over shards (train_utils.py:99-103)."""
-        sq = torch.zeros((), dtype=torch.float32, device=self.device)
-        for b in self.buckets:
-            sq += b.shard_grad.pow(2).sum()
+        """Global grad-norm over shards (train_utils.py:99-103). The clip
+        COEFFICIENT is kept as a device scalar and fused into the AdamW
+        kernel's grad read (one scalar load instead of a read+write pass
+        over every shard grad). Norms use the fixed-order sqsum kernel so
+        the clipped step stays bit-deterministic across runs and across
+        the plain/collective world-1 paths."""
+        if self.on_gpu:
+            if self._sq_partials is None:
+                self._sq_partials = torch.empty(1024, dtype=torch.float32, device=self.device)
+                self._sq_out = torch.empty(len(self.buckets), dtype=torch.float32, device=self.device)
+            from .ops import hip  # noqa: PLC0415
+
+            for i, b in enumerate(self.buckets):
+                g = self._grad_src(b)
+                hip.check(
+                    hip.lib().dolomite_sqsum(
+                        hip.stream(), hip.ptr(g), g.numel(), hip.ptr(self._sq_partials),
+                        hip.ptr(self._sq_out, i), hip.dt(g),
+                    ),
+                    "sqsum",
+                )
+            sq = self._sq_out.sum()
+        else:
+            sq = torch.zeros((), dtype=torch.float32, device=self.device)
+            for b in self.buckets:
+                sq += self._grad_src(b).float().pow(2).sum()
         if self.world > 1:
             dist.all_reduce(sq, op=dist.ReduceOp.SUM)
         total_norm = sq.sqrt()
+        self._clip_coef = None
         if max_norm is not None:
-            clip_coef = (max_norm / (total_norm + 1e-6)).clamp(max=1.0)
-            for b in self.buckets:
-                b.shard_grad.mul_(clip_coef)
+            self._clip_coef = (max_norm / (total_norm + 1e-6)).clamp(max=1.0).to(torch.float32)
         return total_norm
 
     def step(self, lr: float | None = None, grad_clip: float | None = None) -> torch.Tensor:
@@ -232,14 +264,17 @@ class ZeRO2Engine:
         local shard -> all-gather updated params."""
         self._finish_reduces()
         grad_norm = self.clip_grad_norm_(grad_clip) if grad_clip is not None else None
+        if grad_clip is None:
+            self._clip_coef = None
         self.step_count += 1
         use_lr = self.lr if lr is None else lr
         for b in self.buckets:
             param_out = b.flat_param[b.shard_slice]
             adamw_step_flat(
-                b.master, b.shard_grad, b.exp_avg, b.exp_avg_sq, self.step_count,
+                b.master, self._grad_src(b), b.exp_avg, b.exp_avg_sq, self.step_count,
                 use_lr, self.beta1, self.beta2, self.eps, self.weight_decay,
                 param_out=param_out if b.flat_param.dtype != torch.float32 else None,
+                grad_scale=self._clip_coef,
             )
             if b.flat_param.dtype == torch.float32:
                 param_out.copy_(b.master)
@@ -273,7 +308,7 @@ class ZeRO2Engine:
         self._drain_wgrad()
         for b in self.buckets:
             b.flat_grad.zero_()
-            b.shard_grad.zero_()
+            # shard_grad needs no zeroing: the reduce overwrites every slot
 
     # ---- checkpointing (per-rank shards) ----------------------------------
 

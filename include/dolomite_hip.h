@@ -195,17 +195,28 @@ int dolomite_ce_bwd(dolomite_stream_t stream,
  * semantics: decoupled wd, bias-corrected moments.
  *   master/m/v: (n,) fp32.  grad: (n,) fp32 or bf16 (grad_dtype).
  *   param_out: (n,) bf16 copy of updated master (NULL to skip).
+ *   gscale: optional DEVICE scalar multiplied into every grad element
+ *           (the grad-clip coefficient, fused here so the clip costs one
+ *           scalar read instead of a read+write pass; NULL = 1.0).
  * ---------------------------------------------------------------------- */
 int dolomite_adamw_step(dolomite_stream_t stream,
                         float* master, void* param_out_bf16,
                         const void* grad, int grad_dtype,
                         float* m, float* v,
                         int64_t n, float lr, float beta1, float beta2,
-                        float eps, float weight_decay, int step);
+                        float eps, float weight_decay, int step,
+                        const float* gscale);
 
 /* Multiply a flat fp32 (or bf16) buffer by a scalar (grad clip apply). */
 int dolomite_scale_inplace(dolomite_stream_t stream, void* buf, int64_t n,
                            float scale, int dtype);
+
+/* Deterministic sum of squares (grad-norm input): out[0] = sum(x[i]^2) in
+ * fp32, fixed reduction order run-to-run (per-thread strided order + fixed
+ * LDS trees), so the clipped optimizer step stays bit-reproducible.
+ * partials: (1024,) fp32 scratch. x: (n,) fp32 or bf16. */
+int dolomite_sqsum(dolomite_stream_t stream, const void* x, int64_t n,
+                   float* partials, float* out, int dtype);
 
 /* ------------------------------------------------------------------------
  * Grouped expert GEMMs for the MoE family — replaces the reference's
