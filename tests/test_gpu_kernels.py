@@ -277,6 +277,56 @@ def test_adamw_matches_torch_on_gpu(golden_dir):
     torch.testing.assert_close(pout.cpu(), fx["p_final"].to(torch.bfloat16), rtol=0, atol=0)
 
 
+@pytest.mark.parametrize("dtype,n", [(torch.bfloat16, 1_000_003), (torch.float32, 257)])
+def test_sqsum_matches_torch_and_is_bitwise_stable(dtype, n):
+    """dolomite_sqsum (grad-norm input): matches torch fp32 within reduction
+    -order tolerance AND is bitwise identical across repeated runs (the
+    fixed-order property the clipped-step determinism relies on)."""
+    g = torch.Generator().manual_seed(12)
+    x = (torch.randn(n, generator=g) * 0.1).to(dtype).cuda()
+    partials = torch.empty(1024, dtype=torch.float32, device="cuda")
+    out = torch.empty(2, dtype=torch.float32, device="cuda")
+    for i in range(2):
+        hip.check(
+            hip.lib().dolomite_sqsum(
+                hip.stream(), hip.ptr(x), x.numel(), hip.ptr(partials), hip.ptr(out, i), hip.dt(x)
+            ),
+            "sqsum",
+        )
+    torch.cuda.synchronize()
+    assert torch.equal(out[0], out[1])  # fixed reduction order
+    ref = x.float().pow(2).sum()
+    torch.testing.assert_close(out[0], ref, rtol=1e-6, atol=1e-6)
+
+
+def test_adamw_fused_clip_matches_explicit_scale():
+    """grad_scale (device scalar fused into the AdamW kernel) must equal
+    pre-scaling the grads explicitly — exactly the old clip mul pass."""
+    g = torch.Generator().manual_seed(13)
+    n = 100_001
+    p0 = torch.randn(n, generator=g).float()
+    grad = (torch.randn(n, generator=g) * 0.5).to(torch.bfloat16).cuda()
+    coef = torch.tensor(0.3725, dtype=torch.float32).cuda()
+
+    def run(fused):
+        p = p0.clone().cuda()
+        m = torch.zeros_like(p)
+        v = torch.zeros_like(p)
+        pout = torch.empty_like(p, dtype=torch.bfloat16)
+        if fused:
+            Fx.adamw_step_flat(p, grad, m, v, 1, 1e-3, 0.9, 0.95, 1e-10, 0.1,
+                               param_out=pout, grad_scale=coef)
+        else:
+            scaled = (grad.float() * coef).to(torch.float32)
+            Fx.adamw_step_flat(p, scaled, m, v, 1, 1e-3, 0.9, 0.95, 1e-10, 0.1, param_out=pout)
+        torch.cuda.synchronize()
+        return p, m, v, pout
+
+    pa, ma, va, oa = run(True)
+    pb, mb, vb, ob = run(False)
+    assert torch.equal(pa, pb) and torch.equal(ma, mb) and torch.equal(va, vb) and torch.equal(oa, ob)
+
+
 class TestMoEGroupedGemm:
     """Grouped expert GEMM (csrc/moe_gemm.hip) vs the eager per-expert loop
     (reference moe/base.py:12-50 semantics) — ragged groups, empty groups,
