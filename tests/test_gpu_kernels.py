@@ -205,6 +205,52 @@ def test_varlen_attention_fwd_bwd_vs_oracle(head_type, H, Hkv, D, lens):
     torch.testing.assert_close(dv_g.float(), vc.grad, rtol=3e-2, atol=3e-2)
 
 
+@pytest.mark.parametrize("seed", range(6))
+def test_varlen_attention_random_sweep(seed):
+    """Seeded random head geometry + ragged packing through fwd+bwd vs the
+    fp32 oracle — shapes the fixed parametrization does not enumerate
+    (random D multiples of 8, random G, empty/singleton sequences)."""
+    import random as pyrandom
+
+    r = pyrandom.Random(1000 + seed)
+    Hkv = r.choice([1, 2, 4])
+    G = r.choice([1, 2, 3, 4])
+    H = Hkv * G
+    D = 8 * r.randint(2, 16)  # 16..128
+    n_seq = r.randint(1, 4)
+    lens = [r.choice([0, 1, r.randint(2, 300)]) for _ in range(n_seq)]
+    if sum(lens) == 0:
+        lens.append(37)
+    head_type = "mha" if Hkv == H else ("mqa" if Hkv == 1 else "gqa")
+    lo = QKVLayout.make(H, Hkv, D, head_type)
+    g = torch.Generator().manual_seed(seed)
+    cu = torch.tensor([0] + list(torch.tensor(lens).cumsum(0)), dtype=torch.int32)
+    T = int(cu[-1])
+    scale = 1.0 / math.sqrt(D)
+    qkv = (torch.randn(T, lo.row_len, generator=g) * 0.5).to(torch.bfloat16)
+
+    qkv_g = qkv.cuda().requires_grad_(True)
+    o = Fx.varlen_attention(qkv_g, cu.cuda(), max(max(lens), 1), lo, scale)
+    do = (torch.randn(T, H * D, generator=g) * 0.5).to(torch.bfloat16)
+    o.backward(do.cuda())
+    torch.cuda.synchronize()
+
+    q, k, v = lo.unpack_cpu(qkv)
+    qc = q.float().requires_grad_(True)
+    kc = k.float().requires_grad_(True)
+    vc = v.float().requires_grad_(True)
+    oc = oracle.attention_varlen_ref(qc, kc, vc, cu, scale)
+    oc.backward(do.float().reshape(T, H, D))
+
+    torch.testing.assert_close(
+        o.detach().cpu().float().reshape(T, H, D), oc.detach(), rtol=2e-2, atol=2e-2
+    )
+    dq_g, dk_g, dv_g = lo.unpack_cpu(qkv_g.grad.cpu())
+    torch.testing.assert_close(dq_g.float(), qc.grad, rtol=3e-2, atol=3e-2)
+    torch.testing.assert_close(dk_g.float(), kc.grad, rtol=3e-2, atol=3e-2)
+    torch.testing.assert_close(dv_g.float(), vc.grad, rtol=3e-2, atol=3e-2)
+
+
 def test_varlen_attention_deterministic_list_vs_tensor_packing():
     """Same packed input twice -> bitwise-identical output (the list-input
     path reuses the tensor path; reference requires bit-exactness there)."""
