@@ -1179,22 +1179,27 @@ __global__ void __launch_bounds__(512, 4) fa_bwd_dq_kernel(
             bf16x8 dsf1 = *(const bf16x8*)&dSw[lr * ST + 32 + lg * 8];
             lgkm_drain2x8(dsf0, dsf1);
             __builtin_amdgcn_sched_barrier(0);
-            bf16x4 klo[3], khi[3];
+            // depth-4 rotation: 3 fragments in flight per wait (the dq cap
+            // has ~16 VGPR of headroom the dkv cap does not)
+            bf16x4 klo[4], khi[4];
             tr16_issue<0>(aK0, aK1, klo[0], khi[0]);
             tr16_issue<((1 / DCH) * 32 * SQ + (1 % DCH) * 16) * 2>(aK0, aK1, klo[1], khi[1]);
+            tr16_issue<((2 / DCH) * 32 * SQ + (2 % DCH) * 16) * 2>(aK0, aK1, klo[2], khi[2]);
 #define DQ_STEP(i)                                                                                      \
     if constexpr ((i) < 2 * DCH) {                                                                      \
         constexpr int kc2_ = (i) / DCH, dc_ = (i) % DCH;                                                \
-        if constexpr ((i) + 2 < 2 * DCH) {                                                              \
-            constexpr int kn_ = ((i) + 2) / DCH, dn_ = ((i) + 2) % DCH;                                 \
-            tr16_issue<(kn_ * 32 * SQ + dn_ * 16) * 2>(aK0, aK1, klo[((i) + 2) % 3], khi[((i) + 2) % 3]); \
-            lgkm_wait2<4>(klo[(i) % 3], khi[(i) % 3]);                                                  \
+        if constexpr ((i) + 3 < 2 * DCH) {                                                              \
+            constexpr int kn_ = ((i) + 3) / DCH, dn_ = ((i) + 3) % DCH;                                 \
+            tr16_issue<(kn_ * 32 * SQ + dn_ * 16) * 2>(aK0, aK1, klo[((i) + 3) % 4], khi[((i) + 3) % 4]); \
+            lgkm_wait2<6>(klo[(i) % 4], khi[(i) % 4]);                                                  \
+        } else if constexpr ((i) + 2 < 2 * DCH) {                                                       \
+            lgkm_wait2<4>(klo[(i) % 4], khi[(i) % 4]);                                                  \
         } else if constexpr ((i) + 1 < 2 * DCH) {                                                       \
-            lgkm_wait2<2>(klo[(i) % 3], khi[(i) % 3]);                                                  \
+            lgkm_wait2<2>(klo[(i) % 4], khi[(i) % 4]);                                                  \
         } else {                                                                                        \
-            lgkm_wait2<0>(klo[(i) % 3], khi[(i) % 3]);                                                  \
+            lgkm_wait2<0>(klo[(i) % 4], khi[(i) % 4]);                                                  \
         }                                                                                               \
-        dq[dc_] = MFMA16(kc2_ ? dsf1 : dsf0, tr16_join8(klo[(i) % 3], khi[(i) % 3]), dq[dc_]);          \
+        dq[dc_] = MFMA16(kc2_ ? dsf1 : dsf0, tr16_join8(klo[(i) % 4], khi[(i) % 4]), dq[dc_]);          \
     }
             DQ_STEP(0) DQ_STEP(1) DQ_STEP(2) DQ_STEP(3)
             DQ_STEP(4) DQ_STEP(5) DQ_STEP(6) DQ_STEP(7)
