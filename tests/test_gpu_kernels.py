@@ -172,6 +172,8 @@ def test_rope_packed_roundtrip_and_oracle(head_type, H, Hkv, D):
         ("mqa", 8, 1, 80, [7, 0, 9]),     # empty sequence in the packed batch
         ("gqa", 8, 2, 128, [65, 63]),     # edge tiles on both sides of 64
         ("gqa", 6, 2, 64, [100, 60]),     # odd G=3 grouping
+        ("mqa", 4, 1, 96, [200, 56]),     # D == DPAD: pad-zero prologue skipped
+        ("gqa", 4, 2, 48, [77]),          # D=48 -> DPAD 64, 16-col pad band
     ],
 )
 def test_varlen_attention_fwd_bwd_vs_oracle(head_type, H, Hkv, D, lens):
