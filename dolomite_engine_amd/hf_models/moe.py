@@ -15,6 +15,7 @@ import torch.nn as nn
 import torch.nn.functional as F
 from transformers.modeling_outputs import MoeCausalLMOutputWithPast, MoeModelOutputWithPast
 
+from ..ops.functional import MoERowsCombine, MoERowsGather
 from .config import GPTDolomiteConfig
 from .modeling import (
     Attention,
@@ -130,8 +131,14 @@ class SparseMoE(nn.Module):
         _, index_sorted_experts = selected_flat.sort(0)
         batch_index = index_sorted_experts // self.top_k
         batch_gates = router_weights.flatten()[index_sorted_experts]
+        # inverse of the expert sort (inv[pair] = slot): lets the gather
+        # backward and the scatter-back run through the deterministic
+        # combine kernel instead of torch's atomic index_add
+        n_pairs = selected_flat.numel()
+        inv = torch.empty(n_pairs, dtype=torch.int32, device=hidden_states.device)
+        inv[index_sorted_experts] = torch.arange(n_pairs, dtype=torch.int32, device=hidden_states.device)
 
-        expert_inputs = hidden_states[batch_index]
+        expert_inputs = MoERowsGather.apply(hidden_states, batch_index, inv, self.top_k)
         h = self.c_fc(expert_inputs, num_tokens_per_expert)
         if self.is_glu:
             a, b = h.chunk(2, dim=-1)
@@ -140,8 +147,7 @@ class SparseMoE(nn.Module):
             h = self.act(h)
         h = self.c_proj(h, num_tokens_per_expert)
         h = h * batch_gates.unsqueeze(-1)
-        out = torch.zeros(total_q, self.hidden_size, dtype=h.dtype, device=h.device)
-        out = out.index_add(0, batch_index, h)
+        out = MoERowsCombine.apply(h, inv, batch_index, total_q, self.top_k)
         out = self.dropout(out.view(orig_shape))
         return out, router_logits
 

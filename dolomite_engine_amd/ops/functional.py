@@ -655,6 +655,58 @@ class GroupedExpertGemm(torch.autograd.Function):
         return dx, dw, db, None, None
 
 
+class MoERowsGather(torch.autograd.Function):
+    """expert_inputs = x[batch_index] with a deterministic, roofline-rate
+    backward: torch's indexing backward for this pattern (indexing_backward
+    / indexFuncLargeIndex) runs ~9x off the HBM roofline; the combine
+    kernel sums each token's top-k slot rows in FIXED order instead
+    (moe/base.py:108-127 semantics)."""
+
+    @staticmethod
+    def forward(ctx, x, batch_index, inv, k_top):
+        ctx.save_for_backward(batch_index, inv)
+        ctx.k_top = k_top
+        ctx.total_q = x.shape[0]
+        return x[batch_index]
+
+    @staticmethod
+    def backward(ctx, dexp):
+        batch_index, inv = ctx.saved_tensors
+        dx = _rows_combine(dexp, inv, batch_index, ctx.total_q, ctx.k_top)
+        return dx, None, None, None
+
+
+class MoERowsCombine(torch.autograd.Function):
+    """out = zeros(total_q, K).index_add(0, batch_index, h)
+    (moe/base.py:127) with the deterministic combine kernel on the forward
+    and a plain gather backward."""
+
+    @staticmethod
+    def forward(ctx, h, inv, batch_index, total_q, k_top):
+        ctx.save_for_backward(batch_index)
+        return _rows_combine(h, inv, batch_index, total_q, k_top)
+
+    @staticmethod
+    def backward(ctx, dout):
+        (batch_index,) = ctx.saved_tensors
+        return dout[batch_index], None, None, None, None
+
+
+def _rows_combine(h, inv, batch_index, total_q, k_top):
+    h = h.contiguous()
+    if h.is_cuda and h.dtype == torch.bfloat16 and h.shape[1] % 8 == 0:
+        out = torch.empty(total_q, h.shape[1], dtype=h.dtype, device=h.device)
+        hip.check(
+            hip.lib().dolomite_moe_rows_combine(
+                hip.stream(), hip.ptr(h), hip.ptr(inv), hip.ptr(out),
+                total_q, h.shape[1], k_top, hip.dt(h),
+            ),
+            "moe_rows_combine",
+        )
+        return out
+    return torch.zeros(total_q, h.shape[1], dtype=h.dtype, device=h.device).index_add(0, batch_index, h)
+
+
 def grouped_expert_gemm(x, weight, bias, num_tokens_per_expert):
     """Dispatch helper: HIP grouped kernel on CUDA bf16 with 8-aligned dims
     (one launch for all experts); returns None if unsupported so the caller

@@ -51,6 +51,7 @@ _SIGNATURES = {
     "dolomite_ce_bwd": ([_p, _p, _p, _p, _p, _p, _i64, _i64, _i64, _i32, _i32], _i32),
     "dolomite_adamw_step": ([_p, _p, _p, _p, _i32, _p, _p, _i64, _f32, _f32, _f32, _f32, _f32, _i32, _p], _i32),
     "dolomite_sqsum": ([_p, _p, _i64, _p, _p, _i32], _i32),
+    "dolomite_moe_rows_combine": ([_p, _p, _p, _p, _i64, _i32, _i32, _i32], _i32),
     "dolomite_scale_inplace": ([_p, _p, _i64, _f32, _i32], _i32),
     "dolomite_moe_gemm_fwd": ([_p, _p, _p, _p, _p, _p, _i32, _i32, _i32, _i32, _i32], _i32),
     "dolomite_moe_gemm_dgrad": ([_p, _p, _p, _p, _p, _i32, _i32, _i32, _i32, _i32], _i32),

@@ -126,13 +126,15 @@ int dolomite_fa_varlen_fwd(dolomite_stream_t stream,
 
 /* Backward, three passes (all launched by the host in order on `stream`):
  *  1. preprocess: delta[h,t] = rowsum(dO[t,h,:]*O[t,h,:]) fp32.
- *  2. main: grid over (kv-tile, seq, kv-head); recomputes P from q,k,lse;
- *     dk/dv accumulated in registers over the q-tile loop, then added to the
- *     pointers; dkv: one workgroup per (kv-tile, seq, Q-HEAD); dk/dv join
- *     zero-initialized (T,Hkv,D) fp32 buffers with device-scope atomics;
- *     dq: one workgroup per (q-tile, seq, q-head), register-accumulated and
- *     stored bf16 directly into the packed dqkv q-slots.
- *  3. grad_finalize: casts the dk/dv fp32 accumulators into the packed dqkv.
+ *  2. main: dkv — one workgroup per (kv-tile, seq, q-head); recomputes P
+ *     from q,k,lse; dk/dv accumulate in registers over the q-tile loop and
+ *     store EXCLUSIVELY (no atomics) into (T, H, D) fp32 per-q-head
+ *     partials. dq — one workgroup per (q-tile, seq, q-head),
+ *     register-accumulated and stored bf16 directly into the packed dqkv
+ *     q-slots.
+ *  3. grad_finalize: reduces each kv head's G partial contributions in
+ *     FIXED order (deterministic gradients — bit-exact resume depends on
+ *     it) and casts into the packed dqkv.
  */
 int dolomite_fa_bwd_preprocess(dolomite_stream_t stream,
                                const void* o, const void* dout, float* delta,
@@ -210,6 +212,15 @@ int dolomite_adamw_step(dolomite_stream_t stream,
 /* Multiply a flat fp32 (or bf16) buffer by a scalar (grad clip apply). */
 int dolomite_scale_inplace(dolomite_stream_t stream, void* buf, int64_t n,
                            float scale, int dtype);
+
+/* Deterministic MoE top-k row combine: out[t,:] = sum_j h[inv[t*k_top+j],:]
+ * (fixed j order). Replaces torch zeros+index_add for the expert scatter-back
+ * (moe/base.py:127) and the backward of the expert-row gather — both ~9x off
+ * the HBM roofline in eager torch. inv: (T*k_top,) int32 inverse of the
+ * expert sort (inv[pair] = slot). bf16, K % 8 == 0. */
+int dolomite_moe_rows_combine(dolomite_stream_t stream, const void* h,
+                              const int32_t* inv, void* out,
+                              int64_t T, int K, int k_top, int dtype);
 
 /* Deterministic sum of squares (grad-norm input): out[0] = sum(x[i]^2) in
  * fp32, fixed reduction order run-to-run (per-thread strided order + fixed

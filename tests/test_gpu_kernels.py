@@ -375,6 +375,35 @@ def test_adamw_fused_clip_matches_explicit_scale():
     assert torch.equal(pa, pb) and torch.equal(ma, mb) and torch.equal(va, vb) and torch.equal(oa, ob)
 
 
+def test_moe_rows_combine_matches_index_add():
+    """Deterministic top-k combine vs torch zeros+index_add (the reference
+    scatter-back, moe/base.py:127), plus bitwise run-to-run stability."""
+    from dolomite_engine_amd.ops.functional import _rows_combine
+
+    g = torch.Generator().manual_seed(14)
+    T, K, k_top, E = 1337, 1024, 2, 8
+    selected = torch.randint(0, E, (T, k_top), generator=g).flatten()
+    _, sorted_idx = selected.sort(0)
+    batch_index = (sorted_idx // k_top).cuda()
+    n = T * k_top
+    inv = torch.empty(n, dtype=torch.int32)
+    inv[sorted_idx] = torch.arange(n, dtype=torch.int32)
+    inv = inv.cuda()
+    h = (torch.randn(n, K, generator=g) * 0.5).to(torch.bfloat16).cuda()
+
+    out1 = _rows_combine(h, inv, batch_index, T, k_top)
+    out2 = _rows_combine(h, inv, batch_index, T, k_top)
+    torch.cuda.synchronize()
+    assert torch.equal(out1, out2)
+
+    ref = torch.zeros(T, K, dtype=h.dtype, device=h.device).index_add(0, batch_index, h)
+    # fp32-accumulated combine vs bf16 index_add: one rounding difference
+    torch.testing.assert_close(out1, ref, rtol=1e-2, atol=1e-2)
+    # and exactly equals the fp32 reference rounded once
+    ref32 = torch.zeros(T, K, dtype=torch.float32, device=h.device).index_add(0, batch_index.cuda(), h.float())
+    torch.testing.assert_close(out1.float(), ref32, rtol=4e-3, atol=4e-3)
+
+
 class TestMoEGroupedGemm:
     """Grouped expert GEMM (csrc/moe_gemm.hip) vs the eager per-expert loop
     (reference moe/base.py:12-50 semantics) — ragged groups, empty groups,
