@@ -146,7 +146,7 @@ def main():
 
     M = args.micro_batch * args.seq_len
     cfg = wrapper.model.config
-    h, ff = cfg.n_embd, cfg.n_inner
+    h, ff = cfg.n_embd, getattr(cfg, "n_inner", None) or 4 * cfg.n_embd
     D = h // cfg.n_head
     qkv_n = h + 2 * D * (cfg.num_key_value_heads or 1) if (cfg.num_key_value_heads or 1) != cfg.n_head else 3 * h
     flops = {
