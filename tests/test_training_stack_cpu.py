@@ -535,3 +535,37 @@ def test_zero2_gloo_world4_matches_single_process(tmp_path):
     sd = model.state_dict()
     for k, v in dist_result["state"].items():
         torch.testing.assert_close(sd[k], v, rtol=1e-5, atol=1e-6, msg=lambda m: f"{k}: {m}")
+
+
+def test_zero2_no_clip_matches_torch_adamw():
+    """engine.step(grad_clip=None) must equal torch AdamW with NO clipping —
+    exercises the fused-clip path's None branch (the coefficient must not
+    leak from a prior clipped step)."""
+    import torch
+
+    from dolomite_engine_amd.zero import ZeRO2Engine
+
+    model = _make_model(seed=3)
+    ref = _make_model(seed=3)
+    opt = torch.optim.AdamW(ref.parameters(), lr=1e-3, betas=(0.9, 0.95), eps=1e-10, weight_decay=0.1)
+    engine = ZeRO2Engine(model, lr=1e-3, betas=(0.9, 0.95), eps=1e-10, weight_decay=0.1, bucket_mb=1)
+
+    for step in range(3):
+        engine.zero_grad()
+        engine.set_sync(True)
+        loss = _wrapper_loss(model, {"text": _batches(0, step)})
+        loss.backward()
+        # first step clipped, later steps NOT: a stale clip coefficient
+        # from step 0 would corrupt steps 1-2
+        engine.step(lr=1e-3, grad_clip=1.0 if step == 0 else None)
+
+        opt.zero_grad()
+        loss_r = _wrapper_loss(ref, {"text": _batches(0, step)})
+        loss_r.backward()
+        if step == 0:
+            torch.nn.utils.clip_grad_norm_(ref.parameters(), 1.0)
+        opt.step()
+
+    sd, sr = model.state_dict(), ref.state_dict()
+    for k in sr:
+        torch.testing.assert_close(sd[k], sr[k], rtol=1e-5, atol=1e-6, msg=lambda m: f"{k}: {m}")
